@@ -1,0 +1,72 @@
+"""Pod/Trainer/Cluster model tests (mirror reference test_pod.py/test_cluster.py)."""
+import json
+
+from edl_amd.cluster.model import Cluster, Pod, Trainer, load_cluster, save_cluster
+from edl_amd.train.env import JobEnv
+
+
+def make_pod(pod_id, n_trainers=2):
+    p = Pod(pod_id=pod_id, addr="127.0.0.1")
+    for i in range(n_trainers):
+        p.trainers.append(Trainer(endpoint="127.0.0.1:%d" % (9000 + i), gpus=[str(i)],
+                                  rank_in_pod=i))
+    return p
+
+
+def test_pod_json_roundtrip():
+    p = make_pod("p1")
+    p.set_rank(3, 6)
+    q = Pod.from_json(p.to_json())
+    assert q == p
+    assert q.trainers[1].global_rank == 7
+
+
+def test_pod_from_env():
+    env = JobEnv(
+        {"nproc_per_node": 2},
+        env={"CUDA_VISIBLE_DEVICES": "", "PADDLE_TRAINER_PORTS": "7001,7002"},
+    )
+    pod = Pod.from_env(env)
+    assert len(pod.trainers) == 2
+    assert pod.trainers[0].endpoint.endswith(":7001")
+
+
+def test_cluster_rank_assignment_and_stage():
+    c = Cluster(pods=[make_pod("a"), make_pod("b"), make_pod("c")])
+    c.assign_ranks()
+    assert c.world_size() == 6
+    ranks = [t.global_rank for p in c.pods for t in p.trainers]
+    assert ranks == list(range(6))
+    s0 = c.stage
+    c.new_stage()
+    assert c.stage != s0
+    assert c.job_stage == 1
+
+
+def test_cluster_json_and_members():
+    c = Cluster(pods=[make_pod("a"), make_pod("b")])
+    c.assign_ranks()
+    d = Cluster.from_json(c.to_json())
+    assert d == c
+    assert d.same_members(c)
+    e = Cluster(pods=[make_pod("a")])
+    assert not e.same_members(c)
+    # stage survives serialization
+    assert json.loads(c.to_json())["stage"] == c.stage
+
+
+def test_save_load_cluster_guarded(coord_client):
+    c = Cluster(pods=[make_pod("a")])
+    c.assign_ranks()
+    # unguarded write
+    assert save_cluster(coord_client, c)
+    got = load_cluster(coord_client)
+    assert got == c
+    # guarded write only applies when the guard matches
+    coord_client.put("/test_job/rank/nodes/0", "a")
+    c.new_stage()
+    assert save_cluster(coord_client, c, leader_guard=("/test_job/rank/nodes/0", "a"))
+    assert load_cluster(coord_client).stage == c.stage
+    c2 = Cluster(pods=[make_pod("z")])
+    assert not save_cluster(coord_client, c2, leader_guard=("/test_job/rank/nodes/0", "zzz"))
+    assert load_cluster(coord_client).stage == c.stage

@@ -1,0 +1,122 @@
+"""Coordination store unit tests (substrate for everything else).
+
+Mirrors the etcd behaviors the reference relies on: leases+TTL expiry,
+put-if-absent CAS (set_server_not_exists), guarded transactions, watch."""
+import threading
+import time
+
+import pytest
+
+from edl_amd.coord.client import CoordClient
+from edl_amd.utils.errors import EdlStoreError
+
+
+def test_put_get_delete(coord_client):
+    c = coord_client
+    c.put("/a/b", "v1")
+    assert c.get("/a/b") == "v1"
+    c.put("/a/b", "v2")
+    assert c.get("/a/b") == "v2"
+    assert c.get("/missing") is None
+    assert c.delete("/a/b") == 1
+    assert c.get("/a/b") is None
+    assert c.delete("/a/b") == 0
+
+
+def test_range_and_delete_prefix(coord_client):
+    c = coord_client
+    for i in range(5):
+        c.put("/t/nodes/%d" % i, str(i))
+    c.put("/other", "x")
+    kvs = c.range("/t/nodes/")
+    assert [k for k, _ in kvs] == ["/t/nodes/%d" % i for i in range(5)]
+    assert c.delete_prefix("/t/") == 5
+    assert c.range("/t/") == []
+    assert c.get("/other") == "x"
+
+
+def test_lease_expiry(coord_client):
+    c = coord_client
+    lease = c.grant(0.6)
+    c.put("/lease/k", "v", lease)
+    assert c.get("/lease/k") == "v"
+    time.sleep(1.0)
+    assert c.get("/lease/k") is None
+    assert not c.keepalive(lease)
+
+
+def test_lease_keepalive(coord_client):
+    c = coord_client
+    lease = c.grant(0.8)
+    c.put("/ka/k", "v", lease)
+    for _ in range(4):
+        time.sleep(0.3)
+        assert c.keepalive(lease)
+    assert c.get("/ka/k") == "v"
+
+
+def test_cas_put_if_absent(coord_client):
+    c = coord_client
+    ok, val = c.put_if_absent("/rank/0", "pod_a")
+    assert ok and val == "pod_a"
+    ok, val = c.put_if_absent("/rank/0", "pod_b")
+    assert not ok and val == "pod_a"
+
+
+def test_cas_reacquire_after_lease_expiry(coord_client):
+    c = coord_client
+    lease = c.grant(0.5)
+    ok, _ = c.put_if_absent("/rank/0", "pod_a", lease)
+    assert ok
+    time.sleep(0.9)
+    ok, val = c.put_if_absent("/rank/0", "pod_b")
+    assert ok and val == "pod_b"
+
+
+def test_txn_guarded(coord_client):
+    c = coord_client
+    c.put("/rank/0", "leader_pod")
+    assert c.txn_if("/rank/0", "leader_pod", puts=[("/cluster", "{}")])
+    assert c.get("/cluster") == "{}"
+    assert not c.txn_if("/rank/0", "other_pod", puts=[("/cluster", "BAD")])
+    assert c.get("/cluster") == "{}"
+    # txn with deletes
+    assert c.txn_if("/rank/0", "leader_pod", dels=["/cluster"])
+    assert c.get("/cluster") is None
+
+
+def test_wait_wakes_on_change(coord_server, coord_client):
+    c = coord_client
+    rev = c.rev()
+    result = {}
+
+    def waiter():
+        w = CoordClient(coord_server.endpoint, "test_job")
+        result["changed"], result["rev"] = w.wait(rev, timeout=5.0)
+        w.close()
+
+    t = threading.Thread(target=waiter)
+    t.start()
+    time.sleep(0.2)
+    c.put("/wake", "1")
+    t.join(timeout=5)
+    assert result["changed"]
+
+
+def test_wait_timeout(coord_client):
+    c = coord_client
+    t0 = time.monotonic()
+    changed, _ = c.wait(c.rev(), timeout=0.4)
+    assert not changed
+    assert time.monotonic() - t0 < 3.0
+
+
+def test_unreachable_store_raises():
+    c = CoordClient("127.0.0.1:1", "job")
+    with pytest.raises(EdlStoreError):
+        c.get("/x")
+
+
+def test_table_key_layout(coord_client):
+    assert coord_client.table_key("resource", "p1") == "/test_job/resource/nodes/p1"
+    assert coord_client.table_key("resource") == "/test_job/resource/nodes/"

@@ -1,0 +1,173 @@
+"""End-to-end launcher tests: N agent processes sharing one coordination
+store on localhost — how the reference simulates multi-node
+(test_launch.sh:40-66).
+
+Covers: 2-agent success path + job flag; trainer failure -> FAILED;
+elastic scale-in (kill one agent, survivor stop-resumes at world 1);
+scale-out (second agent joins a running job, stage bumps to world 2)."""
+import json
+import os
+import signal
+import subprocess
+import sys
+import time
+
+import pytest
+
+from edl_amd.cluster.status import Status, load_job_status
+from edl_amd.coord.client import CoordClient
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+FAKE = os.path.join(REPO, "tests", "fake_trainer.py")
+
+
+def spawn_agent(store_ep, job_id, tmp_path, name, nodes_range="1:2", extra_env=None):
+    env = dict(os.environ)
+    env.update(
+        {
+            "PYTHONPATH": REPO + os.pathsep + env.get("PYTHONPATH", ""),
+            "EDL_LEASE_TTL": "2",
+            "EDL_DEMO_OUT": str(tmp_path / "demo_out.jsonl"),
+            "CUDA_VISIBLE_DEVICES": "",
+        }
+    )
+    env.update(extra_env or {})
+    logf = open(tmp_path / ("agent_%s.log" % name), "wb")
+    proc = subprocess.Popen(
+        [
+            sys.executable, "-m", "edl_amd.launch",
+            "--job_id", job_id,
+            "--store_endpoints", store_ep,
+            "--nodes_range", nodes_range,
+            "--nproc_per_node", "1",
+            "--log_dir", str(tmp_path / ("logs_" + name)),
+            FAKE,
+        ],
+        env=env,
+        stdout=logf,
+        stderr=subprocess.STDOUT,
+        cwd=REPO,
+        start_new_session=True,
+    )
+    proc._logf = logf
+    return proc
+
+
+def kill_tree(proc):
+    try:
+        os.killpg(os.getpgid(proc.pid), signal.SIGKILL)
+    except ProcessLookupError:
+        pass
+
+
+def read_runs(tmp_path):
+    p = tmp_path / "demo_out.jsonl"
+    if not p.exists():
+        return []
+    return [json.loads(line) for line in p.read_text().splitlines() if line.strip()]
+
+
+@pytest.fixture()
+def agent_reaper():
+    procs = []
+    yield procs
+    for p in procs:
+        kill_tree(p)
+        try:
+            p.wait(timeout=5)
+        except subprocess.TimeoutExpired:
+            pass
+        p._logf.close()
+
+
+def test_two_agents_success(coord_server, tmp_path, agent_reaper):
+    job = "job_ok"
+    a = spawn_agent(coord_server.endpoint, job, tmp_path, "a", nodes_range="2:2")
+    b = spawn_agent(coord_server.endpoint, job, tmp_path, "b", nodes_range="2:2")
+    agent_reaper.extend([a, b])
+    assert a.wait(timeout=60) == 0, (tmp_path / "agent_a.log").read_text()
+    assert b.wait(timeout=60) == 0, (tmp_path / "agent_b.log").read_text()
+    c = CoordClient(coord_server.endpoint, job)
+    assert load_job_status(c) == Status.SUCCEED
+    runs = read_runs(tmp_path)
+    assert sorted(r["rank"] for r in runs) == [0, 1]
+    assert all(r["world"] == 2 for r in runs)
+    c.close()
+
+
+def test_trainer_failure_marks_job_failed(coord_server, tmp_path, agent_reaper):
+    job = "job_fail"
+    a = spawn_agent(
+        coord_server.endpoint, job, tmp_path, "a", nodes_range="1:1",
+        extra_env={"EDL_DEMO_EXIT_CODE": "3"},
+    )
+    agent_reaper.append(a)
+    assert a.wait(timeout=60) != 0
+    c = CoordClient(coord_server.endpoint, job)
+    assert load_job_status(c) == Status.FAILED
+    c.close()
+
+
+def test_elastic_scale_in(coord_server, tmp_path, agent_reaper):
+    """Start 2 agents (trainers sleep), SIGKILL one; survivor must
+    stop-resume at world 1 and finish."""
+    job = "job_shrink"
+    a = spawn_agent(
+        coord_server.endpoint, job, tmp_path, "a",
+        extra_env={"EDL_DEMO_SLEEP": "4"},
+    )
+    b = spawn_agent(
+        coord_server.endpoint, job, tmp_path, "b",
+        extra_env={"EDL_DEMO_SLEEP": "4"},
+    )
+    agent_reaper.extend([a, b])
+    # wait until both trainers have started (world=2 recorded twice)
+    deadline = time.monotonic() + 30
+    while time.monotonic() < deadline:
+        if len([r for r in read_runs(tmp_path) if r["world"] == 2]) >= 2:
+            break
+        time.sleep(0.2)
+        assert a.poll() is None, (tmp_path / "agent_a.log").read_text()
+    else:
+        pytest.fail("both trainers did not start: %s" % read_runs(tmp_path))
+
+    kill_tree(b)
+    # survivor should re-cluster to world=1 (lease TTL 2 s) and finish
+    assert a.wait(timeout=90) == 0, (tmp_path / "agent_a.log").read_text()
+    runs = read_runs(tmp_path)
+    assert any(r["world"] == 1 for r in runs), runs
+    c = CoordClient(coord_server.endpoint, job)
+    assert load_job_status(c) == Status.SUCCEED
+    c.close()
+
+
+def test_elastic_scale_out(coord_server, tmp_path, agent_reaper):
+    """Start 1 agent with range 1:2 (trainer sleeps), add a second agent;
+    the generator must append it (stage bump) and both finish at world 2."""
+    job = "job_grow"
+    a = spawn_agent(
+        coord_server.endpoint, job, tmp_path, "a",
+        extra_env={"EDL_DEMO_SLEEP": "5"},
+    )
+    agent_reaper.append(a)
+    deadline = time.monotonic() + 30
+    while time.monotonic() < deadline:
+        if any(r["world"] == 1 for r in read_runs(tmp_path)):
+            break
+        time.sleep(0.2)
+        assert a.poll() is None, (tmp_path / "agent_a.log").read_text()
+    else:
+        pytest.fail("first trainer did not start")
+
+    b = spawn_agent(
+        coord_server.endpoint, job, tmp_path, "b",
+        extra_env={"EDL_DEMO_SLEEP": "5"},
+    )
+    agent_reaper.append(b)
+    assert a.wait(timeout=90) == 0, (tmp_path / "agent_a.log").read_text()
+    assert b.wait(timeout=90) == 0, (tmp_path / "agent_b.log").read_text()
+    runs = read_runs(tmp_path)
+    assert len([r for r in runs if r["world"] == 2]) >= 2, runs
+    c = CoordClient(coord_server.endpoint, job)
+    assert load_job_status(c) == Status.SUCCEED
+    c.close()
