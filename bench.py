@@ -1,0 +1,143 @@
+#!/usr/bin/env python3
+"""Driver benchmark contract: flagship ResNet50_vd data-parallel training
+step on N GPUs of one node (BASELINE.json metric: img/s whole node at
+total batch 256 == 32/GPU x 8; weak scaling keeps 32/GPU at every N).
+
+    python bench.py --gpus N --steps K --warmup W
+
+For N>1 the driver launches this under torch.distributed.run (one rank per
+GPU over RCCL); standalone invocation with --gpus>1 re-execs itself the
+same way. Rank 0 prints ONE JSON line. Timing: W untimed warmup steps,
+barrier+synchronize, time EXACTLY K steps, barrier+synchronize, MAX over
+ranks."""
+import argparse
+import json
+import os
+import sys
+import time
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=30)
+    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--batch_size", type=int, default=32, help="per-GPU batch")
+    p.add_argument("--model", default="resnet50_vd")
+    p.add_argument("--dtype", default="bf16")
+    p.add_argument("--bucket_mb", type=int, default=25)
+    p.add_argument("--graph_capture", type=int, default=None)
+    p.add_argument("--use_hip_ops", type=int, default=1)
+    return p.parse_args()
+
+
+def maybe_reexec(args):
+    """Standalone multi-GPU invocation -> torch.distributed.run."""
+    if args.gpus > 1 and "WORLD_SIZE" not in os.environ:
+        import subprocess
+
+        cmd = [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", "--nproc-per-node", str(args.gpus),
+            "--master-addr", "127.0.0.1", "--master-port", "29517",
+            os.path.abspath(__file__),
+        ] + sys.argv[1:]
+        os.execvp(cmd[0], cmd)
+
+
+def main():
+    args = parse_args()
+    maybe_reexec(args)
+
+    import torch
+
+    use_cuda = torch.cuda.is_available()
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+
+    sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+    from edl_amd.data.synthetic import SyntheticImageNet
+    from edl_amd.train import dist as edist
+    from edl_amd.train.engine import TrainerEngine
+
+    dtype = args.dtype if use_cuda else "fp32"
+    engine = TrainerEngine(
+        model=args.model,
+        per_device_batch=args.batch_size,
+        base_lr=0.1,
+        dtype=dtype,
+        channels_last=use_cuda,
+        bucket_mb=args.bucket_mb,
+        checkpoint_dir=None,
+        use_hip_ops=bool(args.use_hip_ops) and use_cuda,
+        graph_capture=None if args.graph_capture is None else bool(args.graph_capture),
+    ).setup()
+    engine.model.train()
+    engine.set_lr(engine.scaled_lr(0))
+    loader = SyntheticImageNet(
+        args.batch_size, engine.device, channels_last=use_cuda,
+        seed=1234 + rank,
+    )
+
+    # warmup (untimed); first step may trigger hipGraph capture
+    x, y = loader.next()
+    engine.maybe_capture(x, y)
+    for _ in range(max(1, args.warmup)):
+        x, y = loader.next()
+        engine.replay_step(x, y)
+
+    edist.barrier(engine.device)
+    if use_cuda:
+        torch.cuda.synchronize()
+    t0 = time.monotonic()
+    for _ in range(args.steps):
+        x, y = loader.next()
+        engine.replay_step(x, y)
+    if use_cuda:
+        torch.cuda.synchronize()
+    edist.barrier(engine.device)
+    elapsed = time.monotonic() - t0
+
+    # MAX over ranks (slowest rank defines job time)
+    if world > 1:
+        import torch.distributed as dist
+
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=engine.device if use_cuda else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    n_gpus = world if world > 1 else (1 if use_cuda else args.gpus)
+    total_imgs = args.steps * args.batch_size * world
+    img_per_s = total_imgs / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    if rank == 0:
+        result = {
+            "metric": "img/s",
+            "value": round(img_per_s, 1),
+            "unit": "img/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": round(img_per_s / 1828.0, 3),
+            "dtype": dtype,
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": args.batch_size * world,
+                "image_shape": "3x224x224",
+                "parallelism": "dp%d" % world,
+                "graph_capture": engine._graph is not None,
+            },
+        }
+        print(json.dumps(result))
+    edist.cleanup()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

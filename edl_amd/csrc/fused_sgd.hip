@@ -1,0 +1,57 @@
+// Fused momentum-SGD over a flat parameter bucket (fp32).
+//
+// One launch updates an entire bucket (params + grads + momentum live in
+// matching flat buffers — edl_amd/train/bucketed_ddp.py) and folds in the
+// data-parallel gradient average (grad_scale = 1/world), replacing the
+// reference's per-parameter Paddle Momentum update + separate scale pass
+// (reference example/collective/resnet50/train_with_fleet.py:106-111).
+//
+// Memory-bound: 3 reads + 2 writes of 4 B per element; float4-vectorized
+// (16 B/lane, guide Guideline 13), grid-stride.
+#include "common.h"
+
+extern "C" __global__ void fused_sgd_f32(
+    float* __restrict__ p, const float* __restrict__ g, float* __restrict__ m,
+    const float lr, const float mu, const float wd, const float scale,
+    const long long n) {
+  const long long n4 = n >> 2;
+  const float4* g4 = reinterpret_cast<const float4*>(g);
+  float4* p4 = reinterpret_cast<float4*>(p);
+  float4* m4 = reinterpret_cast<float4*>(m);
+  long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (; i < n4; i += stride) {
+    float4 pg = g4[i], pp = p4[i], pm = m4[i];
+    float d0 = fmaf(scale, pg.x, wd * pp.x);
+    float d1 = fmaf(scale, pg.y, wd * pp.y);
+    float d2 = fmaf(scale, pg.z, wd * pp.z);
+    float d3 = fmaf(scale, pg.w, wd * pp.w);
+    pm.x = fmaf(mu, pm.x, d0);
+    pm.y = fmaf(mu, pm.y, d1);
+    pm.z = fmaf(mu, pm.z, d2);
+    pm.w = fmaf(mu, pm.w, d3);
+    pp.x = fmaf(-lr, pm.x, pp.x);
+    pp.y = fmaf(-lr, pm.y, pp.y);
+    pp.z = fmaf(-lr, pm.z, pp.z);
+    pp.w = fmaf(-lr, pm.w, pp.w);
+    m4[i] = pm;
+    p4[i] = pp;
+  }
+  // tail (n % 4)
+  for (long long j = (n4 << 2) + (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       j < n; j += stride) {
+    float d = fmaf(scale, g[j], wd * p[j]);
+    float mv = fmaf(mu, m[j], d);
+    m[j] = mv;
+    p[j] = fmaf(-lr, mv, p[j]);
+  }
+}
+
+extern "C" void launch_fused_sgd_f32(float* p, const float* g, float* m,
+                                     float lr, float mu, float wd, float scale,
+                                     long long n, hipStream_t stream) {
+  const int block = 256;
+  const int grid = elementwise_grid((n + 3) / 4, block);
+  hipLaunchKernelGGL(fused_sgd_f32, dim3(grid), dim3(block), 0, stream,
+                     p, g, m, lr, mu, wd, scale, n);
+}

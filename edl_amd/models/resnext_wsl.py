@@ -1,0 +1,74 @@
+"""ResNeXt101_32x16d_wsl — the reference's distill TEACHER model
+(README.md:51-75: remote ResNeXt101_32x16d_wsl teachers served to
+ResNet50_vd students). Standard ResNeXt architecture (grouped 3×3,
+width = planes·(base_width/64)·groups), 7×7 stem.
+
+Served by edl_amd.distill.teacher_server; its forward is the target of the
+hand-written CDNA4 HIP conv/BN/ReLU path (BASELINE.json north star)."""
+import torch.nn as nn
+
+
+class ResNeXtBottleneck(nn.Module):
+    expansion = 4
+
+    def __init__(self, cin, planes, stride=1, groups=32, base_width=16):
+        super().__init__()
+        width = int(planes * (base_width / 64.0)) * groups
+        cout = planes * self.expansion
+        self.conv1 = nn.Conv2d(cin, width, 1, bias=False)
+        self.bn1 = nn.BatchNorm2d(width)
+        self.conv2 = nn.Conv2d(width, width, 3, stride=stride, padding=1,
+                               groups=groups, bias=False)
+        self.bn2 = nn.BatchNorm2d(width)
+        self.conv3 = nn.Conv2d(width, cout, 1, bias=False)
+        self.bn3 = nn.BatchNorm2d(cout)
+        self.relu = nn.ReLU(inplace=True)
+        self.downsample = None
+        if stride != 1 or cin != cout:
+            self.downsample = nn.Sequential(
+                nn.Conv2d(cin, cout, 1, stride=stride, bias=False),
+                nn.BatchNorm2d(cout),
+            )
+
+    def forward(self, x):
+        s = x if self.downsample is None else self.downsample(x)
+        y = self.relu(self.bn1(self.conv1(x)))
+        y = self.relu(self.bn2(self.conv2(y)))
+        y = self.bn3(self.conv3(y))
+        return self.relu(y + s)
+
+
+class ResNeXtWSL(nn.Module):
+    def __init__(self, depths=(3, 4, 23, 3), groups=32, base_width=16, num_classes=1000):
+        super().__init__()
+        self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
+        self.bn1 = nn.BatchNorm2d(64)
+        self.relu = nn.ReLU(inplace=True)
+        self.maxpool = nn.MaxPool2d(3, 2, padding=1)
+        cin = 64
+        stages = []
+        for bi, (p, d) in enumerate(zip([64, 128, 256, 512], depths)):
+            blocks = []
+            for i in range(d):
+                stride = 2 if i == 0 and bi != 0 else 1
+                blocks.append(ResNeXtBottleneck(cin, p, stride, groups, base_width))
+                cin = p * ResNeXtBottleneck.expansion
+            stages.append(nn.Sequential(*blocks))
+        self.stages = nn.Sequential(*stages)
+        self.avgpool = nn.AdaptiveAvgPool2d(1)
+        self.fc = nn.Linear(cin, num_classes)
+        for m in self.modules():
+            if isinstance(m, nn.Conv2d):
+                nn.init.kaiming_normal_(m.weight, mode="fan_out", nonlinearity="relu")
+            elif isinstance(m, nn.BatchNorm2d):
+                nn.init.ones_(m.weight)
+                nn.init.zeros_(m.bias)
+
+    def forward(self, x):
+        x = self.maxpool(self.relu(self.bn1(self.conv1(x))))
+        x = self.stages(x)
+        return self.fc(self.avgpool(x).flatten(1))
+
+
+def resnext101_32x16d_wsl(num_classes=1000):
+    return ResNeXtWSL((3, 4, 23, 3), groups=32, base_width=16, num_classes=num_classes)

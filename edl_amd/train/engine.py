@@ -1,0 +1,284 @@
+"""The MI355X training engine — what Paddle Fleet was to the reference.
+
+One process per GPU (RCCL over xGMI), bf16 autocast + channels_last NHWC,
+gradient-bucketed overlap all-reduce (bucketed_ddp), fused HIP ops where
+the extension is available (edl_amd.ops), optional whole-step hipGraph
+capture (launch-bound CNN steps replay as ONE graph), versioned async
+checkpointing, and elastic resume driven by the launcher's stop-resume.
+
+Replaces: fleet.distributed_optimizer + DistributedStrategy
+(train_with_fleet.py:367-381), FLAGS_sync_nccl_allreduce, fuse_all_reduce_ops,
+fleet save/load_check_point."""
+import os
+import time
+
+import torch
+import torch.nn.functional as F
+
+from ..models import build_model
+from ..utils.log import get_logger
+from . import dist as edist
+from .bucketed_ddp import BucketedAllReducer
+from .checkpoint import CheckpointManager
+from .env import TrainerEnv
+
+log = get_logger("edl.engine")
+
+
+def piecewise_lr(base_lr, epoch, boundaries=(30, 60, 90), decay=0.1, warmup_epochs=5,
+                 step_in_epoch=0.0):
+    """Reference LR policy (train_with_fleet.py piecewise_decay 30/60/90
+    x0.1) + linear warmup."""
+    e = epoch + step_in_epoch
+    if e < warmup_epochs:
+        return base_lr * (e + 1e-9) / warmup_epochs
+    mult = 1.0
+    for b in boundaries:
+        if epoch >= b:
+            mult *= decay
+    return base_lr * mult
+
+
+class TrainerEngine:
+    def __init__(
+        self,
+        model="resnet50_vd",
+        per_device_batch=32,
+        num_classes=1000,
+        base_lr=0.1,
+        momentum=0.9,
+        weight_decay=1e-4,
+        label_smoothing=0.1,
+        dtype="bf16",
+        channels_last=True,
+        bucket_mb=25,
+        checkpoint_dir=None,
+        use_hip_ops=True,
+        graph_capture=None,
+        kd_teacher=None,
+        kd_alpha=1.0,
+    ):
+        self.model_name = model
+        self.per_device_batch = per_device_batch
+        self.num_classes = num_classes
+        self.base_lr = base_lr
+        self.momentum = momentum
+        self.weight_decay = weight_decay
+        self.label_smoothing = label_smoothing
+        self.dtype = {"bf16": torch.bfloat16, "fp16": torch.float16,
+                      "fp32": torch.float32}[dtype]
+        self.channels_last = channels_last
+        self.bucket_mb = bucket_mb
+        self.checkpoint_dir = checkpoint_dir
+        self.use_hip_ops = use_hip_ops
+        self.graph_capture = graph_capture
+        self.kd_teacher = kd_teacher  # callable(images)->soft logits, or None
+        self.kd_alpha = kd_alpha
+
+        self.env = None
+        self.device = None
+        self.model = None
+        self.reducer = None
+        self.opt = None
+        self.ckpt = None
+        self.start_epoch = 0
+        self.global_step = 0
+        self._graph = None
+        self._static = {}
+
+    # ---- setup ----
+    def setup(self, env=None):
+        self.env, self.device = edist.init_from_env(env)
+        torch.manual_seed(42 + self.env.global_rank)
+        self.model = build_model(self.model_name, num_classes=self.num_classes)
+        self.model.to(self.device)
+        if self.channels_last and self.device.type == "cuda":
+            self.model.to(memory_format=torch.channels_last)
+        if self.use_hip_ops and self.device.type == "cuda":
+            from .. import ops
+
+            ops.swap_module_ops(self.model)
+
+        self.reducer = BucketedAllReducer(
+            self.model.parameters(), bucket_cap_mb=self.bucket_mb
+        )
+        self.opt = self._build_optimizer()
+        if self.checkpoint_dir:
+            self.ckpt = CheckpointManager(self.checkpoint_dir)
+            self._resume()
+        # everyone starts from rank-0's weights (fresh or resumed)
+        self.reducer.broadcast_params(src=0)
+        edist.barrier(self.device)
+        return self
+
+    def _build_optimizer(self):
+        from ..ops.sgd import FusedSGD
+
+        return FusedSGD(
+            self.model.parameters(), lr=self.base_lr, momentum=self.momentum,
+            weight_decay=self.weight_decay, grad_scale=self.reducer.grad_scale,
+            reducer=self.reducer,
+        )
+
+    @property
+    def world_size(self):
+        return edist.world_size()
+
+    @property
+    def global_batch(self):
+        return self.per_device_batch * self.world_size
+
+    def scaled_lr(self, epoch, step_in_epoch=0.0):
+        """Linear-scaling rule: base_lr is quoted at total batch 256
+        (reference train_parameters, models/resnet_vd.py) and rescales on
+        every elastic world change."""
+        lr = self.base_lr * self.global_batch / 256.0
+        return piecewise_lr(lr, epoch, step_in_epoch=step_in_epoch)
+
+    def set_lr(self, lr):
+        for g in self.opt.param_groups:
+            g["lr"] = lr
+
+    # ---- checkpoint ----
+    def _resume(self):
+        got = self.ckpt.load(map_location="cpu")
+        if got is None:
+            if self.env.is_rank0:
+                log.info("no checkpoint in %s; fresh start", self.checkpoint_dir)
+            return
+        model_state, opt_state, ts = got
+        self.model.load_state_dict(model_state)
+        if opt_state is not None:
+            try:
+                self.opt.load_state_dict(opt_state)
+            except (ValueError, KeyError) as e:
+                log.warning("optimizer state mismatch (%s); reset", e)
+        self.start_epoch = int(ts.get("epoch_no", -1)) + 1
+        self.global_step = int(ts.get("global_step", 0))
+        if self.env.is_rank0:
+            log.info(
+                "resumed from checkpoint v%s: next epoch %d, global_step %d (world=%d)",
+                ts.get("_version"), self.start_epoch, self.global_step, self.world_size,
+            )
+
+    def save_checkpoint(self, epoch, extra=None, blocking=False):
+        if self.ckpt is None or not self.env.is_rank0:
+            return None
+        ts = {"epoch_no": epoch, "global_step": self.global_step,
+              "world_size": self.world_size, "total_batch_size": self.global_batch}
+        ts.update(extra or {})
+        return self.ckpt.save(
+            self.model.state_dict(), ts, optimizer_state=self.opt.state_dict(),
+            blocking=blocking,
+        )
+
+    # ---- the step ----
+    def _loss(self, logits, labels, teacher_logits=None):
+        if teacher_logits is not None:
+            # KD soft-label cross-entropy (reference
+            # example/distill/resnet/train_with_fleet.py:254-259: student
+            # trains against teacher soft labels) — fused HIP kernel on GPU
+            from ..ops.functional import kd_soft_cross_entropy
+
+            kd = kd_soft_cross_entropy(logits, teacher_logits)
+            if self.kd_alpha >= 1.0:
+                return kd
+            ce = F.cross_entropy(logits.float(), labels,
+                                 label_smoothing=self.label_smoothing)
+            return self.kd_alpha * kd + (1 - self.kd_alpha) * ce
+        return F.cross_entropy(logits.float(), labels,
+                               label_smoothing=self.label_smoothing)
+
+    def train_step(self, images, labels, teacher_logits=None):
+        self.reducer.zero_grad()
+        with torch.autocast(device_type=self.device.type, dtype=self.dtype,
+                            enabled=self.dtype != torch.float32):
+            logits = self.model(images)
+        loss = self._loss(logits, labels, teacher_logits)
+        loss.backward()
+        self.reducer.finalize()
+        if not getattr(self.opt, "handles_grad_scale", False):
+            scale = self.reducer.grad_scale
+            if scale != 1.0:
+                for b in self.reducer._buckets:
+                    b.buffer.mul_(scale)
+        self.opt.step()
+        self.global_step += 1
+        return loss
+
+    # ---- hipGraph capture ----
+    def maybe_capture(self, images, labels):
+        """Capture the full train step (fwd+bwd+allreduce+opt) as ONE hip
+        graph. CNN steps at bs 32/GPU are launch-bound; replaying a single
+        graph removes per-kernel launch gaps (guide §launches-baseline).
+        Returns True if capture succeeded."""
+        if self.device.type != "cuda" or self._graph is not None:
+            return False
+        want = self.graph_capture
+        if want is None:
+            want = os.environ.get("EDL_GRAPH_CAPTURE", "1") == "1"
+        if not want:
+            return False
+        try:
+            self._static["x"] = images.clone()
+            self._static["y"] = labels.clone()
+            torch.cuda.synchronize()
+            # warmup on a side stream (required before capture)
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                for _ in range(3):
+                    self.train_step(self._static["x"], self._static["y"])
+            torch.cuda.current_stream().wait_stream(s)
+            torch.cuda.synchronize()
+            g = torch.cuda.CUDAGraph()
+            step_before = self.global_step
+            with torch.cuda.graph(g):
+                self._static["loss"] = self.train_step(self._static["x"], self._static["y"])
+            self.global_step = step_before  # capture itself is not a step
+            self._graph = g
+            log.info("captured train step as hipGraph")
+            return True
+        except Exception as e:  # noqa: BLE001 - fall back to eager
+            log.warning("hipGraph capture failed (%s); running eager", e)
+            self._graph = None
+            self._static = {}
+            return False
+
+    def replay_step(self, images, labels):
+        """Run one training step; uses the captured graph when available."""
+        if self._graph is not None:
+            self._static["x"].copy_(images, non_blocking=True)
+            self._static["y"].copy_(labels, non_blocking=True)
+            self._graph.replay()
+            self.global_step += 1
+            return self._static["loss"]
+        return self.train_step(images, labels)
+
+    # ---- epoch driver ----
+    def train_epoch(self, epoch, loader, steps, log_every=50, on_step=None):
+        self.model.train()
+        t0 = time.monotonic()
+        imgs_done = 0
+        loss = None
+        for it in range(steps):
+            self.set_lr(self.scaled_lr(epoch, it / max(1, steps)))
+            x, y = loader.next()
+            loss = self.replay_step(x, y)
+            imgs_done += self.global_batch
+            if on_step:
+                on_step(epoch, it)
+            if log_every and (it + 1) % log_every == 0 and self.env.is_rank0:
+                if self.device.type == "cuda":
+                    torch.cuda.synchronize()
+                dt = time.monotonic() - t0
+                log.info(
+                    "epoch %d step %d/%d loss=%.4f %.1f img/s (world=%d)",
+                    epoch, it + 1, steps, loss.item(), imgs_done / dt, self.world_size,
+                )
+        if self.device.type == "cuda":
+            torch.cuda.synchronize()
+        dt = time.monotonic() - t0
+        return {"epoch": epoch, "steps": steps, "time_s": dt,
+                "img_per_s": imgs_done / dt,
+                "loss": float(loss.item()) if loss is not None else None}

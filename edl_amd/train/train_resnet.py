@@ -1,0 +1,118 @@
+"""Flagship collective trainer — the `train_with_fleet.py` equivalent.
+
+Spawned by the edlrun launcher (one process per GPU); reads the env
+contract (TrainerEnv), builds the RCCL world, trains ResNet50_vd (or any
+model in the zoo) on synthetic ImageNet-shaped data with elastic
+stop-resume: per-epoch (and optional per-N-steps) checkpoints, LR rescaled
+by the linear-scaling rule on every world change, train status reported to
+the coordination store so the generator can veto near-end scaling.
+
+CLI parity subset of reference example/collective/resnet50/train_with_fleet.py.
+"""
+import argparse
+import os
+import sys
+import time
+
+import torch
+
+from ..cluster.status import TrainStatus, save_train_status
+from ..coord.client import CoordClient
+from ..data.synthetic import SyntheticImageNet
+from ..utils.log import get_logger
+from . import dist as edist
+from .engine import TrainerEngine
+from .env import TrainerEnv
+
+log = get_logger("edl.train")
+
+
+def parse_args(argv=None):
+    p = argparse.ArgumentParser("edl_amd resnet trainer")
+    p.add_argument("--model", default="resnet50_vd")
+    p.add_argument("--batch_size", type=int, default=32, help="per-GPU batch")
+    p.add_argument("--num_epochs", type=int, default=2)
+    p.add_argument("--steps_per_epoch", type=int, default=100)
+    p.add_argument("--lr", type=float, default=0.1)
+    p.add_argument("--momentum", type=float, default=0.9)
+    p.add_argument("--weight_decay", type=float, default=1e-4)
+    p.add_argument("--checkpoint", default=None, help="checkpoint dir (resume + save)")
+    p.add_argument("--checkpoint_steps", type=int, default=0,
+                   help="also checkpoint every N steps (0 = per-epoch only)")
+    p.add_argument("--data_dir", default=None, help="unused (synthetic data)")
+    p.add_argument("--use_hip_ops", type=int, default=1)
+    p.add_argument("--graph_capture", type=int, default=None)
+    p.add_argument("--bucket_mb", type=int, default=25)
+    p.add_argument("--dtype", default="bf16", choices=["bf16", "fp16", "fp32"])
+    p.add_argument("--profile", action="store_true")
+    return p.parse_args(argv)
+
+
+def main(argv=None):
+    args = parse_args(argv)
+    tenv = TrainerEnv()
+    ckpt_dir = args.checkpoint or tenv.checkpoint_dir
+
+    engine = TrainerEngine(
+        model=args.model,
+        per_device_batch=args.batch_size,
+        base_lr=args.lr,
+        momentum=args.momentum,
+        weight_decay=args.weight_decay,
+        dtype=args.dtype if torch.cuda.is_available() else "fp32",
+        channels_last=torch.cuda.is_available(),
+        bucket_mb=args.bucket_mb,
+        checkpoint_dir=ckpt_dir,
+        use_hip_ops=bool(args.use_hip_ops) and torch.cuda.is_available(),
+        graph_capture=None if args.graph_capture is None else bool(args.graph_capture),
+    ).setup(tenv)
+
+    # status reporting back to the control plane (reference TrainStatus flow,
+    # utils/train_status.py; generator reads it to veto near-end scale-out)
+    store = None
+    if tenv.store_endpoints:
+        try:
+            store = CoordClient(tenv.store_endpoints, tenv.job_id)
+        except Exception as e:  # noqa: BLE001
+            log.warning("no coordination store (%s); status reporting off", e)
+    pod_id = os.environ.get("EDL_POD_ID", "pod")
+
+    def report(status):
+        if store is not None and tenv.is_rank0:
+            try:
+                save_train_status(store, pod_id, status)
+            except Exception:  # noqa: BLE001
+                pass
+
+    report(TrainStatus.RUNNING)
+    loader = SyntheticImageNet(
+        args.batch_size, engine.device,
+        channels_last=engine.channels_last and engine.device.type == "cuda",
+        seed=1234 + engine.env.global_rank,
+    )
+
+    def on_step(epoch, it):
+        if args.checkpoint_steps and (engine.global_step % args.checkpoint_steps == 0):
+            engine.save_checkpoint(epoch - 1, extra={"mid_epoch": True})
+
+    t_start = time.monotonic()
+    for epoch in range(engine.start_epoch, args.num_epochs):
+        if epoch == args.num_epochs - 1:
+            report(TrainStatus.NEARTHEEND)
+        stats = engine.train_epoch(epoch, loader, args.steps_per_epoch, on_step=on_step)
+        if engine.env.is_rank0:
+            log.info("epoch %d done: %.1f img/s (world=%d, global_batch=%d)",
+                     epoch, stats["img_per_s"], engine.world_size, engine.global_batch)
+        engine.save_checkpoint(epoch)
+    if engine.ckpt is not None:
+        engine.ckpt.wait()
+    report(TrainStatus.SUCCEED)
+    edist.barrier(engine.device)
+    if engine.env.is_rank0:
+        log.info("training done in %.1fs", time.monotonic() - t_start)
+    edist.cleanup()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())
