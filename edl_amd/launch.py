@@ -80,8 +80,10 @@ class Launcher:
             self._client, self._pod.pod_id,
             min_nodes=self._env.min_nodes, max_nodes=self._env.max_nodes,
         )
+        retry = float(os.environ.get("EDL_LEADER_RETRY", "3"))
         self._elector = LeaderElector(
-            self._client, self._pod.pod_id, on_elected=lambda: self._generator.start()
+            self._client, self._pod.pod_id, on_elected=lambda: self._generator.start(),
+            retry_interval=retry,
         ).start()
 
     def launch(self):

@@ -27,6 +27,7 @@ def spawn_agent(store_ep, job_id, tmp_path, name, nodes_range="1:2", extra_env=N
         {
             "PYTHONPATH": REPO + os.pathsep + env.get("PYTHONPATH", ""),
             "EDL_LEASE_TTL": "2",
+            "EDL_LEADER_RETRY": "0.5",
             "EDL_DEMO_OUT": str(tmp_path / "demo_out.jsonl"),
             "CUDA_VISIBLE_DEVICES": "",
         }
@@ -112,13 +113,16 @@ def test_elastic_scale_in(coord_server, tmp_path, agent_reaper):
     """Start 2 agents (trainers sleep), SIGKILL one; survivor must
     stop-resume at world 1 and finish."""
     job = "job_shrink"
+    # sleep must exceed worst-case failure detection (lease TTL 2 s +
+    # election retry + generator period), or the survivor finishes at
+    # world 2 before any resize when the KILLED agent was the leader
     a = spawn_agent(
         coord_server.endpoint, job, tmp_path, "a",
-        extra_env={"EDL_DEMO_SLEEP": "4"},
+        extra_env={"EDL_DEMO_SLEEP": "10"},
     )
     b = spawn_agent(
         coord_server.endpoint, job, tmp_path, "b",
-        extra_env={"EDL_DEMO_SLEEP": "4"},
+        extra_env={"EDL_DEMO_SLEEP": "10"},
     )
     agent_reaper.extend([a, b])
     # wait until both trainers have started (world=2 recorded twice)
