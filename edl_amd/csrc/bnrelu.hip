@@ -1,0 +1,313 @@
+// Fused BatchNorm(+residual Add)+ReLU, NHWC bf16, training fwd + bwd.
+//
+// Replaces the reference's fuse_bn_act_ops / cudnn_batchnorm_spatial_persistent
+// knobs (reference train_with_fleet.py:374, train_pretrain.sh:20) — and, on
+// MI355X, torch-autocast's fp32 BatchNorm path, which pays TWO full-tensor
+// casts (bf16->fp32->bf16: the SubTensorOpWithCast* kernels = ~13% of step
+// time in rocprof) plus 5 MIOpen kernels per BN. Here: bf16 reads with fp32
+// math, 3 lean kernels fwd / 3 bwd, ReLU and the bottleneck's residual add
+// folded in, dgamma/dbeta written straight into fp32 grad bucket views.
+//
+// Layout: x is NHWC [M rows, C channels], C % 8 == 0 (all ResNet/ResNeXt
+// widths). Each thread owns 8 consecutive channels (one ushort4x2 = 16 B
+// load, guide G13); a block's 256 threads cover 2048 channels' worth of a
+// row-group; blocks grid-stride over rows and atomically fold per-channel
+// fp32 partials (distinct addresses — L2-rate, not contended).
+#include "common.h"
+
+using bf16 = __hip_bfloat16;
+
+struct F8 {
+  float v[8];
+};
+
+__device__ __forceinline__ F8 load8(const bf16* p) {
+  // 16-byte vector load of 8 bf16
+  const uint4 raw = *reinterpret_cast<const uint4*>(p);
+  F8 o;
+  const ushort* u = reinterpret_cast<const ushort*>(&raw);
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    union { unsigned u32; float f; } c;
+    c.u32 = ((unsigned)u[i]) << 16;
+    o.v[i] = c.f;
+  }
+  return o;
+}
+
+__device__ __forceinline__ void store8(bf16* p, const F8& x) {
+  uint4 raw;
+  ushort* u = reinterpret_cast<ushort*>(&raw);
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    u[i] = (ushort)(__hip_bfloat16_raw(__float2bfloat16(x.v[i])).x);
+  }
+  *reinterpret_cast<uint4*>(p) = raw;
+}
+
+// ---------------- forward ----------------
+
+// K1: per-channel sum and sum-of-squares partials (training stats).
+extern "C" __global__ void bn_stats_kernel(
+    const bf16* __restrict__ x, float* __restrict__ sums,  // [2, C]
+    const long long M, const int C) {
+  const int c8 = C >> 3;  // channel-octet count
+  const int tpr = c8;     // threads per row-slice (each owns 8 channels)
+  const int tid = blockIdx.x * blockDim.x + threadIdx.x;
+  const int lane_c = tid % tpr;          // which channel octet
+  const long long row0 = tid / tpr;      // starting row
+  const long long rstride = ((long long)gridDim.x * blockDim.x) / tpr;
+  const int c0 = lane_c * 8;
+
+  float s[8] = {0}, q[8] = {0};
+  for (long long r = row0; r < M; r += rstride) {
+    F8 v = load8(x + r * C + c0);
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      s[i] += v.v[i];
+      q[i] = fmaf(v.v[i], v.v[i], q[i]);
+    }
+  }
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    atomicAdd(&sums[c0 + i], s[i]);
+    atomicAdd(&sums[C + c0 + i], q[i]);
+  }
+}
+
+// K2: finalize mean/invstd, update running stats. One thread per channel.
+extern "C" __global__ void bn_finalize_kernel(
+    const float* __restrict__ sums, const float* __restrict__ gamma,
+    const float* __restrict__ beta, float* __restrict__ mean_out,
+    float* __restrict__ invstd_out, float* __restrict__ scale_out,
+    float* __restrict__ shift_out, float* __restrict__ running_mean,
+    float* __restrict__ running_var, const float momentum, const float eps,
+    const long long M, const int C) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  const float inv_m = 1.0f / (float)M;
+  const float mean = sums[c] * inv_m;
+  const float var = fmaxf(sums[C + c] * inv_m - mean * mean, 0.0f);
+  const float invstd = rsqrtf(var + eps);
+  mean_out[c] = mean;
+  invstd_out[c] = invstd;
+  const float g = gamma[c];
+  scale_out[c] = g * invstd;
+  shift_out[c] = beta[c] - g * invstd * mean;
+  if (running_mean != nullptr) {
+    // unbiased variance for the running estimate (torch semantics)
+    const float ub = (M > 1) ? var * (float)M / (float)(M - 1) : var;
+    running_mean[c] = fmaf(momentum, mean - running_mean[c], running_mean[c]);
+    running_var[c] = fmaf(momentum, ub - running_var[c], running_var[c]);
+  }
+}
+
+// K3: y = relu?(x*scale + shift [+ res]); NHWC vectorized by channel octets.
+template <bool RELU, bool ADD>
+__global__ void bn_apply_kernel(
+    const bf16* __restrict__ x, const bf16* __restrict__ res,
+    bf16* __restrict__ y, const float* __restrict__ scale,
+    const float* __restrict__ shift, const long long M, const int C) {
+  const int c8 = C >> 3;
+  const long long total = M * c8;
+  long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (; i < total; i += stride) {
+    const int oct = (int)(i % c8);
+    const long long eoff = (i / c8) * C + oct * 8;
+    F8 v = load8(x + eoff);
+    const int c0 = oct * 8;
+#pragma unroll
+    for (int k = 0; k < 8; ++k) v.v[k] = fmaf(v.v[k], scale[c0 + k], shift[c0 + k]);
+    if (ADD) {
+      F8 rv = load8(res + eoff);
+#pragma unroll
+      for (int k = 0; k < 8; ++k) v.v[k] += rv.v[k];
+    }
+    if (RELU) {
+#pragma unroll
+      for (int k = 0; k < 8; ++k) v.v[k] = fmaxf(v.v[k], 0.0f);
+    }
+    store8(y + eoff, v);
+  }
+}
+
+// ---------------- backward ----------------
+
+// B1: per-channel reductions s1 = sum(dy_eff), s2 = sum(dy_eff * xhat),
+// where dy_eff applies the fused ReLU mask (y > 0).
+template <bool RELU>
+__global__ void bn_bwd_reduce_kernel(
+    const bf16* __restrict__ dy, const bf16* __restrict__ y,
+    const bf16* __restrict__ x, const float* __restrict__ mean,
+    const float* __restrict__ invstd, float* __restrict__ sums,  // [2, C]
+    const long long M, const int C) {
+  const int c8 = C >> 3;
+  const int tpr = c8;
+  const int tid = blockIdx.x * blockDim.x + threadIdx.x;
+  const int lane_c = tid % tpr;
+  const long long row0 = tid / tpr;
+  const long long rstride = ((long long)gridDim.x * blockDim.x) / tpr;
+  const int c0 = lane_c * 8;
+
+  float mu[8], is[8];
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    mu[i] = mean[c0 + i];
+    is[i] = invstd[c0 + i];
+  }
+  float s1[8] = {0}, s2[8] = {0};
+  for (long long r = row0; r < M; r += rstride) {
+    const long long eoff = r * C + c0;
+    F8 g = load8(dy + eoff);
+    F8 xv = load8(x + eoff);
+    if (RELU) {
+      F8 yv = load8(y + eoff);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) g.v[i] = yv.v[i] > 0.0f ? g.v[i] : 0.0f;
+    }
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      s1[i] += g.v[i];
+      s2[i] = fmaf(g.v[i], (xv.v[i] - mu[i]) * is[i], s2[i]);
+    }
+  }
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    atomicAdd(&sums[c0 + i], s1[i]);
+    atomicAdd(&sums[C + c0 + i], s2[i]);
+  }
+}
+
+// (no separate dgamma/dbeta kernel: the bwd-reduce workspace IS [dbeta; dgamma]
+// — sums[0:C] = s1 = dbeta, sums[C:2C] = s2 = dgamma; Python returns views.)
+
+// B3: dx = scale * (dy_eff - s1/M - xhat * s2/M); optional dres = dy_eff.
+template <bool RELU, bool ADD, bool TRAINING>
+__global__ void bn_bwd_dx_kernel(
+    const bf16* __restrict__ dy, const bf16* __restrict__ y,
+    const bf16* __restrict__ x, const float* __restrict__ mean,
+    const float* __restrict__ invstd, const float* __restrict__ gamma,
+    const float* __restrict__ sums, bf16* __restrict__ dx,
+    bf16* __restrict__ dres, const long long M, const int C) {
+  const int c8 = C >> 3;
+  const long long total = M * c8;
+  const float inv_m = 1.0f / (float)M;
+  long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (; i < total; i += stride) {
+    const int oct = (int)(i % c8);
+    const int c0 = oct * 8;
+    const long long eoff = (i / c8) * C + c0;
+    F8 g = load8(dy + eoff);
+    if (RELU) {
+      F8 yv = load8(y + eoff);
+#pragma unroll
+      for (int k = 0; k < 8; ++k) g.v[k] = yv.v[k] > 0.0f ? g.v[k] : 0.0f;
+    }
+    if (ADD) store8(dres + eoff, g);
+    F8 xv = load8(x + eoff);
+    F8 o;
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      const int c = c0 + k;
+      const float is = invstd[c];
+      if (TRAINING) {
+        const float xhat = (xv.v[k] - mean[c]) * is;
+        o.v[k] = gamma[c] * is *
+                 (g.v[k] - sums[c] * inv_m - xhat * sums[C + c] * inv_m);
+      } else {
+        o.v[k] = gamma[c] * is * g.v[k];
+      }
+    }
+    store8(dx + eoff, o);
+  }
+}
+
+// ---------------- launchers ----------------
+
+static inline int stats_grid(long long M, int C) {
+  // enough thread-rows to cover the tensor a few times over; cap atomics
+  const int c8 = C >> 3;
+  long long rows_per_block = 256 / c8 > 0 ? 256 / c8 : 1;
+  long long want = (M + rows_per_block - 1) / rows_per_block;
+  long long cap = 1024;
+  long long g = want < cap ? want : cap;
+  return (int)(g > 0 ? g : 1);
+}
+
+extern "C" void launch_bn_stats(const void* x, float* sums, long long M, int C,
+                                hipStream_t s) {
+  hipLaunchKernelGGL(bn_stats_kernel, dim3(stats_grid(M, C)), dim3(256), 0, s,
+                     (const bf16*)x, sums, M, C);
+}
+
+extern "C" void launch_bn_finalize(const float* sums, const float* gamma,
+                                   const float* beta, float* mean, float* invstd,
+                                   float* scale, float* shift, float* rmean,
+                                   float* rvar, float momentum, float eps,
+                                   long long M, int C, hipStream_t s) {
+  hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 255) / 256), dim3(256), 0, s,
+                     sums, gamma, beta, mean, invstd, scale, shift, rmean, rvar,
+                     momentum, eps, M, C);
+}
+
+extern "C" void launch_bn_apply(const void* x, const void* res, void* y,
+                                const float* scale, const float* shift,
+                                long long M, int C, bool relu, bool add,
+                                hipStream_t s) {
+  const long long total = M * (C >> 3);
+  const int grid = elementwise_grid(total, 256);
+  if (relu && add)
+    hipLaunchKernelGGL((bn_apply_kernel<true, true>), dim3(grid), dim3(256), 0, s,
+                       (const bf16*)x, (const bf16*)res, (bf16*)y, scale, shift, M, C);
+  else if (relu)
+    hipLaunchKernelGGL((bn_apply_kernel<true, false>), dim3(grid), dim3(256), 0, s,
+                       (const bf16*)x, nullptr, (bf16*)y, scale, shift, M, C);
+  else if (add)
+    hipLaunchKernelGGL((bn_apply_kernel<false, true>), dim3(grid), dim3(256), 0, s,
+                       (const bf16*)x, (const bf16*)res, (bf16*)y, scale, shift, M, C);
+  else
+    hipLaunchKernelGGL((bn_apply_kernel<false, false>), dim3(grid), dim3(256), 0, s,
+                       (const bf16*)x, nullptr, (bf16*)y, scale, shift, M, C);
+}
+
+extern "C" void launch_bn_bwd_reduce(const void* dy, const void* y, const void* x,
+                                     const float* mean, const float* invstd,
+                                     float* sums, long long M, int C, bool relu,
+                                     hipStream_t s) {
+  const int grid = stats_grid(M, C);
+  if (relu)
+    hipLaunchKernelGGL((bn_bwd_reduce_kernel<true>), dim3(grid), dim3(256), 0, s,
+                       (const bf16*)dy, (const bf16*)y, (const bf16*)x, mean, invstd,
+                       sums, M, C);
+  else
+    hipLaunchKernelGGL((bn_bwd_reduce_kernel<false>), dim3(grid), dim3(256), 0, s,
+                       (const bf16*)dy, nullptr, (const bf16*)x, mean, invstd,
+                       sums, M, C);
+}
+
+extern "C" void launch_bn_bwd_dx(const void* dy, const void* y, const void* x,
+                                 const float* mean, const float* invstd,
+                                 const float* gamma, const float* sums, void* dx,
+                                 void* dres, long long M, int C, bool relu,
+                                 bool add, bool training, hipStream_t s) {
+  const long long total = M * (C >> 3);
+  const int grid = elementwise_grid(total, 256);
+#define CASE(R, A, T)                                                          \
+  hipLaunchKernelGGL((bn_bwd_dx_kernel<R, A, T>), dim3(grid), dim3(256), 0, s, \
+                     (const bf16*)dy, (const bf16*)y, (const bf16*)x, mean,    \
+                     invstd, gamma, sums, (bf16*)dx, (bf16*)dres, M, C)
+  if (training) {
+    if (relu && add) CASE(true, true, true);
+    else if (relu) CASE(true, false, true);
+    else if (add) CASE(false, true, true);
+    else CASE(false, false, true);
+  } else {
+    if (relu && add) CASE(true, true, false);
+    else if (relu) CASE(true, false, false);
+    else if (add) CASE(false, true, false);
+    else CASE(false, false, false);
+  }
+#undef CASE
+}

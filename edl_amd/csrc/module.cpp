@@ -16,6 +16,20 @@ extern "C" void launch_kd_ce_fwd_bf16(const void*, const void*, float*, int, int
                                       hipStream_t);
 extern "C" void launch_kd_ce_bwd_bf16(const void*, const void*, void*, float, int,
                                       int, hipStream_t);
+extern "C" void launch_bn_stats(const void*, float*, long long, int, hipStream_t);
+extern "C" void launch_bn_finalize(const float*, const float*, const float*, float*,
+                                   float*, float*, float*, float*, float*, float,
+                                   float, long long, int, hipStream_t);
+extern "C" void launch_bn_apply(const void*, const void*, void*, const float*,
+                                const float*, long long, int, bool, bool,
+                                hipStream_t);
+extern "C" void launch_bn_bwd_reduce(const void*, const void*, const void*,
+                                     const float*, const float*, float*, long long,
+                                     int, bool, hipStream_t);
+extern "C" void launch_bn_bwd_dx(const void*, const void*, const void*, const float*,
+                                 const float*, const float*, const float*, void*,
+                                 void*, long long, int, bool, bool, bool,
+                                 hipStream_t);
 
 namespace {
 
@@ -74,6 +88,85 @@ torch::Tensor kd_ce_backward(torch::Tensor s, torch::Tensor t, double gout_over_
   return ds;
 }
 
+// x: NHWC bf16 viewed as [M, C] contiguous (channels_last 4-D collapses to
+// this). gamma/beta/running stats: fp32 [C].
+void check_bn_inputs(const torch::Tensor& x, int64_t C) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16,
+              "bn: bf16 GPU tensor required");
+  TORCH_CHECK(C % 8 == 0, "bn: C % 8 == 0 required, got ", C);
+  TORCH_CHECK(C <= 2048, "bn: C <= 2048 supported");
+}
+
+std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor gamma,
+                                        torch::Tensor beta, torch::Tensor rmean,
+                                        torch::Tensor rvar, double momentum,
+                                        double eps,
+                                        c10::optional<torch::Tensor> res,
+                                        bool relu) {
+  // x: [M, C] contiguous view of an NHWC tensor (Python side reshapes)
+  const int64_t C = gamma.numel();
+  TORCH_CHECK(x.dim() == 2 && x.size(1) == C && x.is_contiguous(),
+              "bn: x must be a contiguous [M, C] view");
+  check_bn_inputs(x, C);
+  const long long M = x.numel() / C;
+  auto opts = gamma.options().dtype(torch::kFloat32);
+  auto sums = torch::zeros({2, C}, opts);
+  auto mean = torch::empty({C}, opts);
+  auto invstd = torch::empty({C}, opts);
+  auto scale = torch::empty({C}, opts);
+  auto shift = torch::empty({C}, opts);
+  auto y = torch::empty_like(x);
+  auto s = cur_stream();
+  launch_bn_stats(x.data_ptr(), sums.data_ptr<float>(), M, (int)C, s);
+  launch_bn_finalize(sums.data_ptr<float>(), gamma.data_ptr<float>(),
+                     beta.data_ptr<float>(), mean.data_ptr<float>(),
+                     invstd.data_ptr<float>(), scale.data_ptr<float>(),
+                     shift.data_ptr<float>(),
+                     rmean.defined() ? rmean.data_ptr<float>() : nullptr,
+                     rvar.defined() ? rvar.data_ptr<float>() : nullptr,
+                     (float)momentum, (float)eps, M, (int)C, s);
+  launch_bn_apply(x.data_ptr(), res.has_value() ? res->data_ptr() : nullptr,
+                  y.data_ptr(), scale.data_ptr<float>(), shift.data_ptr<float>(),
+                  M, (int)C, relu, res.has_value(), s);
+  return {y, mean, invstd};
+}
+
+torch::Tensor bn_fwd_eval(torch::Tensor x, torch::Tensor scale, torch::Tensor shift,
+                          c10::optional<torch::Tensor> res, bool relu) {
+  const int64_t C = scale.numel();
+  check_bn_inputs(x, C);
+  const long long M = x.numel() / C;
+  auto y = torch::empty_like(x);
+  launch_bn_apply(x.data_ptr(), res.has_value() ? res->data_ptr() : nullptr,
+                  y.data_ptr(), scale.data_ptr<float>(), shift.data_ptr<float>(),
+                  M, (int)C, relu, res.has_value(), cur_stream());
+  return y;
+}
+
+std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y, torch::Tensor x,
+                                  torch::Tensor mean, torch::Tensor invstd,
+                                  torch::Tensor gamma, bool relu, bool add,
+                                  bool training) {
+  const int64_t C = gamma.numel();
+  check_bn_inputs(x, C);
+  const long long M = x.numel() / C;
+  auto sums = torch::zeros({2, C}, gamma.options().dtype(torch::kFloat32));
+  auto dx = torch::empty_like(x);
+  auto dres = add ? torch::empty_like(x) : torch::Tensor();
+  auto s = cur_stream();
+  auto dyc = dy.is_contiguous() ? dy : dy.contiguous();
+  launch_bn_bwd_reduce(dyc.data_ptr(), relu ? y.data_ptr() : nullptr, x.data_ptr(),
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       sums.data_ptr<float>(), M, (int)C, relu, s);
+  launch_bn_bwd_dx(dyc.data_ptr(), relu ? y.data_ptr() : nullptr, x.data_ptr(),
+                   mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                   gamma.data_ptr<float>(), sums.data_ptr<float>(), dx.data_ptr(),
+                   add ? dres.data_ptr() : nullptr, M, (int)C, relu, add, training,
+                   s);
+  // dbeta = sums[0], dgamma = sums[1] (views of the reduce workspace)
+  return {dx, sums[1], sums[0], dres};
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -81,5 +174,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused flat momentum-SGD update (p,g,m flat fp32; folds grad_scale)");
   m.def("kd_ce_forward", &kd_ce_forward, "KD soft-label CE forward -> per-row loss");
   m.def("kd_ce_backward", &kd_ce_backward, "KD soft-label CE backward -> dlogits");
+  m.def("bn_fwd_train", &bn_fwd_train,
+        "fused NHWC bf16 BN(+add)+ReLU train fwd -> (y, mean, invstd)");
+  m.def("bn_fwd_eval", &bn_fwd_eval, "fused NHWC bf16 BN(+add)+ReLU eval fwd");
+  m.def("bn_bwd", &bn_bwd,
+        "fused BN(+add)+ReLU bwd -> (dx, dgamma, dbeta, dres?)");
   m.attr("_arch") = "gfx950";
 }

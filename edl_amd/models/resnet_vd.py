@@ -6,23 +6,49 @@ shortcut = 2×2 avg-pool + 1×1 conv on stride-2 stages, bottleneck with
 stride on the 3×3), re-implemented natively in PyTorch — this is the "bag
 of tricks" ResNet-D of He et al., public architecture.
 
-MI355X notes: intended to run channels_last (NHWC) + bf16 autocast; the
-hot fused ops (BN+ReLU epilogues, KD loss, SGD) are swapped in by
-edl_amd.ops at the engine level, keeping this definition plain torch so it
-also runs on CPU for tests.
-"""
+MI355X-first structure: every BN is an edl_amd.ops.bnrelu fused module —
+BN+ReLU after convs, BN+Add+ReLU at block tails — so on GPU the whole
+non-conv side of the network runs as hand-written NHWC bf16 CDNA4 kernels
+(plain-torch fallback keeps CPU tests runnable)."""
 import torch
 import torch.nn as nn
 
+from ..ops.bnrelu import BNAddReLU2d, BNReLU2d
 
-def _conv_bn(cin, cout, k, stride=1, act=True):
-    layers = [
-        nn.Conv2d(cin, cout, k, stride=stride, padding=(k - 1) // 2, bias=False),
-        nn.BatchNorm2d(cout),
-    ]
-    if act:
-        layers.append(nn.ReLU(inplace=True))
-    return nn.Sequential(*layers)
+
+class ConvBN(nn.Module):
+    def __init__(self, cin, cout, k, stride=1, act=True):
+        super().__init__()
+        self.conv = nn.Conv2d(cin, cout, k, stride=stride, padding=(k - 1) // 2,
+                              bias=False)
+        self.bn = BNReLU2d(cout, act=act)
+
+    def forward(self, x):
+        return self.bn(self.conv(x))
+
+
+class VdShortcut(nn.Module):
+    """Identity, or (vd) avgpool+1x1conv+BN projection."""
+
+    def __init__(self, cin, cout, stride, if_first):
+        super().__init__()
+        self.pool = None
+        self.proj = None
+        if cin != cout or stride != 1:
+            if stride != 1 and not if_first:
+                # the "vd" trick: downsample by avg-pool, then 1x1 stride 1
+                self.pool = nn.AvgPool2d(2, 2, ceil_mode=True)
+                conv_stride = 1
+            else:
+                conv_stride = stride
+            self.proj = ConvBN(cin, cout, 1, stride=conv_stride, act=False)
+
+    def forward(self, x):
+        if self.pool is not None:
+            x = self.pool(x)
+        if self.proj is not None:
+            x = self.proj(x)
+        return x
 
 
 class BottleneckVd(nn.Module):
@@ -31,24 +57,16 @@ class BottleneckVd(nn.Module):
     def __init__(self, cin, planes, stride=1, if_first=False):
         super().__init__()
         cout = planes * self.expansion
-        self.conv0 = _conv_bn(cin, planes, 1)
-        self.conv1 = _conv_bn(planes, planes, 3, stride=stride)
-        self.conv2 = _conv_bn(planes, cout, 1, act=False)
-        self.shortcut = None
-        if cin != cout or stride != 1:
-            sc = []
-            if stride != 1 and not if_first:
-                # the "vd" trick: downsample by avg-pool, then 1x1 stride 1
-                sc.append(nn.AvgPool2d(2, 2, ceil_mode=True))
-            sc.append(nn.Conv2d(cin, cout, 1, stride=1 if not if_first else stride, bias=False))
-            sc.append(nn.BatchNorm2d(cout))
-            self.shortcut = nn.Sequential(*sc)
-        self.relu = nn.ReLU(inplace=True)
+        self.conv0 = ConvBN(cin, planes, 1)
+        self.conv1 = ConvBN(planes, planes, 3, stride=stride)
+        self.conv2 = nn.Conv2d(planes, cout, 1, bias=False)
+        self.bn_add = BNAddReLU2d(cout)  # relu(bn(conv2) + shortcut), fused
+        self.shortcut = VdShortcut(cin, cout, stride, if_first)
 
     def forward(self, x):
-        s = x if self.shortcut is None else self.shortcut(x)
+        s = self.shortcut(x)
         y = self.conv2(self.conv1(self.conv0(x)))
-        return self.relu(y + s)
+        return self.bn_add(y, s)
 
 
 class BasicBlockVd(nn.Module):
@@ -57,21 +75,14 @@ class BasicBlockVd(nn.Module):
     def __init__(self, cin, planes, stride=1, if_first=False):
         super().__init__()
         cout = planes * self.expansion
-        self.conv0 = _conv_bn(cin, planes, 3, stride=stride)
-        self.conv1 = _conv_bn(planes, cout, 3, act=False)
-        self.shortcut = None
-        if cin != cout or stride != 1:
-            sc = []
-            if stride != 1 and not if_first:
-                sc.append(nn.AvgPool2d(2, 2, ceil_mode=True))
-            sc.append(nn.Conv2d(cin, cout, 1, stride=1 if not if_first else stride, bias=False))
-            sc.append(nn.BatchNorm2d(cout))
-            self.shortcut = nn.Sequential(*sc)
-        self.relu = nn.ReLU(inplace=True)
+        self.conv0 = ConvBN(cin, planes, 3, stride=stride)
+        self.conv1 = nn.Conv2d(planes, cout, 3, padding=1, bias=False)
+        self.bn_add = BNAddReLU2d(cout)
+        self.shortcut = VdShortcut(cin, cout, stride, if_first)
 
     def forward(self, x):
-        s = x if self.shortcut is None else self.shortcut(x)
-        return self.relu(self.conv1(self.conv0(x)) + s)
+        s = self.shortcut(x)
+        return self.bn_add(self.conv1(self.conv0(x)), s)
 
 
 _DEPTHS = {
@@ -89,9 +100,9 @@ class ResNetVd(nn.Module):
         super().__init__()
         block, depths = _DEPTHS[layers]
         self.stem = nn.Sequential(
-            _conv_bn(3, 32, 3, stride=2),
-            _conv_bn(32, 32, 3),
-            _conv_bn(32, 64, 3),
+            ConvBN(3, 32, 3, stride=2),
+            ConvBN(32, 32, 3),
+            ConvBN(32, 64, 3),
             nn.MaxPool2d(3, 2, padding=1),
         )
         planes = [64, 128, 256, 512]
@@ -113,9 +124,6 @@ class ResNetVd(nn.Module):
         for m in self.modules():
             if isinstance(m, nn.Conv2d):
                 nn.init.kaiming_normal_(m.weight, mode="fan_out", nonlinearity="relu")
-            elif isinstance(m, nn.BatchNorm2d):
-                nn.init.ones_(m.weight)
-                nn.init.zeros_(m.bias)
 
     def forward(self, x):
         x = self.stages(self.stem(x))

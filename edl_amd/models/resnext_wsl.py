@@ -3,9 +3,11 @@
 ResNet50_vd students). Standard ResNeXt architecture (grouped 3×3,
 width = planes·(base_width/64)·groups), 7×7 stem.
 
-Served by edl_amd.distill.teacher_server; its forward is the target of the
-hand-written CDNA4 HIP conv/BN/ReLU path (BASELINE.json north star)."""
+Served by edl_amd.distill.teacher_server; BN(+Add)+ReLU run as the fused
+CDNA4 kernels (edl_amd.ops.bnrelu) on GPU."""
 import torch.nn as nn
+
+from ..ops.bnrelu import BNAddReLU2d, BNReLU2d
 
 
 class ResNeXtBottleneck(nn.Module):
@@ -16,34 +18,31 @@ class ResNeXtBottleneck(nn.Module):
         width = int(planes * (base_width / 64.0)) * groups
         cout = planes * self.expansion
         self.conv1 = nn.Conv2d(cin, width, 1, bias=False)
-        self.bn1 = nn.BatchNorm2d(width)
+        self.bn1 = BNReLU2d(width)
         self.conv2 = nn.Conv2d(width, width, 3, stride=stride, padding=1,
                                groups=groups, bias=False)
-        self.bn2 = nn.BatchNorm2d(width)
+        self.bn2 = BNReLU2d(width)
         self.conv3 = nn.Conv2d(width, cout, 1, bias=False)
-        self.bn3 = nn.BatchNorm2d(cout)
-        self.relu = nn.ReLU(inplace=True)
+        self.bn_add = BNAddReLU2d(cout)
         self.downsample = None
         if stride != 1 or cin != cout:
             self.downsample = nn.Sequential(
                 nn.Conv2d(cin, cout, 1, stride=stride, bias=False),
-                nn.BatchNorm2d(cout),
+                BNReLU2d(cout, act=False),
             )
 
     def forward(self, x):
         s = x if self.downsample is None else self.downsample(x)
-        y = self.relu(self.bn1(self.conv1(x)))
-        y = self.relu(self.bn2(self.conv2(y)))
-        y = self.bn3(self.conv3(y))
-        return self.relu(y + s)
+        y = self.bn1(self.conv1(x))
+        y = self.bn2(self.conv2(y))
+        return self.bn_add(self.conv3(y), s)
 
 
 class ResNeXtWSL(nn.Module):
     def __init__(self, depths=(3, 4, 23, 3), groups=32, base_width=16, num_classes=1000):
         super().__init__()
         self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
-        self.bn1 = nn.BatchNorm2d(64)
-        self.relu = nn.ReLU(inplace=True)
+        self.bn1 = BNReLU2d(64)
         self.maxpool = nn.MaxPool2d(3, 2, padding=1)
         cin = 64
         stages = []
@@ -60,12 +59,9 @@ class ResNeXtWSL(nn.Module):
         for m in self.modules():
             if isinstance(m, nn.Conv2d):
                 nn.init.kaiming_normal_(m.weight, mode="fan_out", nonlinearity="relu")
-            elif isinstance(m, nn.BatchNorm2d):
-                nn.init.ones_(m.weight)
-                nn.init.zeros_(m.bias)
 
     def forward(self, x):
-        x = self.maxpool(self.relu(self.bn1(self.conv1(x))))
+        x = self.maxpool(self.bn1(self.conv1(x)))
         x = self.stages(x)
         return self.fc(self.avgpool(x).flatten(1))
 

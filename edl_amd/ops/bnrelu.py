@@ -1,0 +1,122 @@
+"""Fused BatchNorm(+Add)+ReLU modules backed by the CDNA4 kernels.
+
+BNReLU2d / BNAddReLU2d replace (BatchNorm2d, ReLU) and the bottleneck tail
+(BatchNorm2d, +residual, ReLU). On GPU with bf16 channels_last input the
+HIP kernels run (fp32 math, no autocast cast traffic, no
+num_batches_tracked kernel); anywhere else a plain torch reference runs —
+the same math the numerics tests compare against.
+
+State-dict layout matches nn.BatchNorm2d (weight/bias/running_mean/
+running_var/num_batches_tracked), so checkpoints interchange."""
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from . import available, ext
+
+
+class _FusedBN(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x2d, gamma, beta, rmean, rvar, momentum, eps, res2d, relu):
+        y, mean, invstd = ext().bn_fwd_train(
+            x2d, gamma, beta, rmean, rvar, momentum, eps, res2d, relu
+        )
+        ctx.save_for_backward(x2d, y, mean, invstd, gamma)
+        ctx.relu = relu
+        ctx.has_res = res2d is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x2d, y, mean, invstd, gamma = ctx.saved_tensors
+        dx, dgamma, dbeta, dres = ext().bn_bwd(
+            dy, y, x2d, mean, invstd, gamma, ctx.relu, ctx.has_res, True
+        )
+        return (dx, dgamma, dbeta, None, None, None, None,
+                dres if ctx.has_res else None, None)
+
+
+def _to_2d(t):
+    """NCHW-channels_last tensor -> [M, C] contiguous view (zero-copy)."""
+    n, c, h, w = t.shape
+    return t.permute(0, 2, 3, 1).reshape(n * h * w, c)
+
+
+def _from_2d(t2d, shape):
+    n, c, h, w = shape
+    return t2d.view(n, h, w, c).permute(0, 3, 1, 2)
+
+
+class BNReLU2d(nn.Module):
+    """BatchNorm2d (+ optional fused ReLU). act=False gives plain BN."""
+
+    def __init__(self, num_features, eps=1e-5, momentum=0.1, act=True):
+        super().__init__()
+        self.num_features = num_features
+        self.eps = eps
+        self.momentum = momentum
+        self.act = act
+        self.weight = nn.Parameter(torch.ones(num_features))
+        self.bias = nn.Parameter(torch.zeros(num_features))
+        self.register_buffer("running_mean", torch.zeros(num_features))
+        self.register_buffer("running_var", torch.ones(num_features))
+        self.register_buffer("num_batches_tracked", torch.tensor(0, dtype=torch.long))
+
+    def _use_fused(self, x, res=None):
+        if not (x.is_cuda and x.dtype == torch.bfloat16 and available()):
+            return False
+        if self.num_features % 8 != 0 or self.num_features > 2048:
+            return False
+        if not x.is_contiguous(memory_format=torch.channels_last):
+            return False
+        if res is not None and (
+            res.dtype != torch.bfloat16
+            or not res.is_contiguous(memory_format=torch.channels_last)
+        ):
+            return False
+        return True
+
+    def _fallback(self, x, res=None):
+        y = F.batch_norm(
+            x.float(), self.running_mean, self.running_var, self.weight, self.bias,
+            self.training, self.momentum, self.eps,
+        )
+        if res is not None:
+            y = y + res.float()
+        if self.act:
+            y = F.relu(y)
+        return y.to(x.dtype)
+
+    def _fused(self, x, res=None):
+        x2d = _to_2d(x)
+        res2d = _to_2d(res) if res is not None else None
+        if self.training:
+            y2d = _FusedBN.apply(
+                x2d, self.weight, self.bias, self.running_mean, self.running_var,
+                self.momentum, self.eps, res2d, self.act,
+            )
+        else:
+            invstd = torch.rsqrt(self.running_var + self.eps)
+            scale = self.weight * invstd
+            shift = self.bias - self.running_mean * scale
+            y2d = ext().bn_fwd_eval(x2d, scale, shift, res2d, self.act)
+        return _from_2d(y2d, x.shape)
+
+    def forward(self, x, res=None):
+        if self._use_fused(x, res):
+            if self.training or not torch.is_grad_enabled():
+                return self._fused(x, res)
+        return self._fallback(x, res)
+
+    def extra_repr(self):
+        return "%d, act=%s" % (self.num_features, self.act)
+
+
+class BNAddReLU2d(BNReLU2d):
+    """The bottleneck tail: relu(bn(x) + residual) as one fused op."""
+
+    def __init__(self, num_features, eps=1e-5, momentum=0.1):
+        super().__init__(num_features, eps=eps, momentum=momentum, act=True)
+
+    def forward(self, x, res):  # res is mandatory here
+        return super().forward(x, res)
