@@ -47,12 +47,14 @@ __device__ __forceinline__ void store8(bf16* p, const F8& x) {
 
 // ---------------- forward ----------------
 
-// K1: per-channel sum and sum-of-squares partials (training stats).
-// Per-block LDS reduction first; ONE global atomicAdd per channel per
-// block (per-THREAD global atomics on 2C words measured 2.3 ms/dispatch —
-// ~1000x this kernel's memory time — from same-word serialization).
+// K1: per-channel sum and sum-of-squares PARTIALS (training stats).
+// Two-stage: each block LDS-reduces its rows and STORES its 2C partials to
+// partial[block][2C]; the finalize kernel reduces over blocks. No global
+// atomics at all (single-stage atomics measured 17-40 us/dispatch fixed
+// cost: same-word serialization at ~88 adds/us for small C, grid x 2C
+// traffic for large C).
 extern "C" __global__ void bn_stats_kernel(
-    const bf16* __restrict__ x, float* __restrict__ sums,  // [2, C]
+    const bf16* __restrict__ x, float* __restrict__ partial,  // [grid, 2C]
     const long long M, const int C) {
   __shared__ float lsum[2 * 2048];
   const int c8 = C >> 3;  // channel-octet count
@@ -81,14 +83,16 @@ extern "C" __global__ void bn_stats_kernel(
     atomicAdd(&lsum[C + c0 + i], q[i]);
   }
   __syncthreads();
-  for (int i = threadIdx.x; i < 2 * C; i += blockDim.x) {
-    atomicAdd(&sums[i], lsum[i]);
-  }
+  float* out = partial + (long long)blockIdx.x * 2 * C;
+  for (int i = threadIdx.x; i < 2 * C; i += blockDim.x) out[i] = lsum[i];
 }
 
-// K2: finalize mean/invstd, update running stats. One thread per channel.
+// K2: reduce partials over blocks, finalize mean/invstd, update running
+// stats. One thread per channel; thread c's reads of partial[b][c] are
+// coalesced across the warp for each fixed b.
 extern "C" __global__ void bn_finalize_kernel(
-    const float* __restrict__ sums, const float* __restrict__ gamma,
+    const float* __restrict__ partial, const int nblocks,
+    const float* __restrict__ gamma,
     const float* __restrict__ beta, float* __restrict__ mean_out,
     float* __restrict__ invstd_out, float* __restrict__ scale_out,
     float* __restrict__ shift_out, float* __restrict__ running_mean,
@@ -96,9 +100,14 @@ extern "C" __global__ void bn_finalize_kernel(
     const long long M, const int C) {
   const int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
+  float s = 0.0f, q = 0.0f;
+  for (int b = 0; b < nblocks; ++b) {
+    s += partial[(long long)b * 2 * C + c];
+    q += partial[(long long)b * 2 * C + C + c];
+  }
   const float inv_m = 1.0f / (float)M;
-  const float mean = sums[c] * inv_m;
-  const float var = fmaxf(sums[C + c] * inv_m - mean * mean, 0.0f);
+  const float mean = s * inv_m;
+  const float var = fmaxf(q * inv_m - mean * mean, 0.0f);
   const float invstd = rsqrtf(var + eps);
   mean_out[c] = mean;
   invstd_out[c] = invstd;
@@ -193,9 +202,19 @@ __global__ void bn_bwd_reduce_kernel(
     atomicAdd(&lsum[C + c0 + i], s2[i]);
   }
   __syncthreads();
-  for (int i = threadIdx.x; i < 2 * C; i += blockDim.x) {
-    atomicAdd(&sums[i], lsum[i]);
-  }
+  float* out = sums + (long long)blockIdx.x * 2 * C;  // [grid, 2C] partials
+  for (int i = threadIdx.x; i < 2 * C; i += blockDim.x) out[i] = lsum[i];
+}
+
+// B1b: reduce bwd partials over blocks -> sums[2C] (= [dbeta; dgamma]).
+extern "C" __global__ void bn_bwd_finalize_kernel(
+    const float* __restrict__ partial, const int nblocks,
+    float* __restrict__ sums, const int C) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= 2 * C) return;
+  float s = 0.0f;
+  for (int b = 0; b < nblocks; ++b) s += partial[(long long)b * 2 * C + i];
+  sums[i] = s;
 }
 
 // (no separate dgamma/dbeta kernel: the bwd-reduce workspace IS [dbeta; dgamma]
@@ -245,30 +264,40 @@ __global__ void bn_bwd_dx_kernel(
 
 // ---------------- launchers ----------------
 
-static inline int stats_grid(long long M, int C) {
-  // enough thread-rows to cover the tensor a few times over; cap atomics
+extern "C" int bn_stats_grid(long long M, int C) {
+  // cover the tensor with enough blocks for bandwidth, but keep the
+  // partial buffer (grid x 2C fp32) around <= 1 MiB
   const int c8 = C >> 3;
   long long rows_per_block = 256 / c8 > 0 ? 256 / c8 : 1;
   long long want = (M + rows_per_block - 1) / rows_per_block;
-  long long cap = 1024;
+  long long cap = 262144 / (2 * (long long)C);  // 1 MiB of partials
+  if (cap > 1024) cap = 1024;
+  if (cap < 8) cap = 8;
   long long g = want < cap ? want : cap;
   return (int)(g > 0 ? g : 1);
 }
 
-extern "C" void launch_bn_stats(const void* x, float* sums, long long M, int C,
-                                hipStream_t s) {
-  hipLaunchKernelGGL(bn_stats_kernel, dim3(stats_grid(M, C)), dim3(256), 0, s,
-                     (const bf16*)x, sums, M, C);
+extern "C" void launch_bn_stats(const void* x, float* partial, int grid,
+                                long long M, int C, hipStream_t s) {
+  hipLaunchKernelGGL(bn_stats_kernel, dim3(grid), dim3(256), 0, s,
+                     (const bf16*)x, partial, M, C);
 }
 
-extern "C" void launch_bn_finalize(const float* sums, const float* gamma,
+extern "C" void launch_bn_finalize(const float* partial, int nblocks,
+                                   const float* gamma,
                                    const float* beta, float* mean, float* invstd,
                                    float* scale, float* shift, float* rmean,
                                    float* rvar, float momentum, float eps,
                                    long long M, int C, hipStream_t s) {
   hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 255) / 256), dim3(256), 0, s,
-                     sums, gamma, beta, mean, invstd, scale, shift, rmean, rvar,
-                     momentum, eps, M, C);
+                     partial, nblocks, gamma, beta, mean, invstd, scale, shift,
+                     rmean, rvar, momentum, eps, M, C);
+}
+
+extern "C" void launch_bn_bwd_finalize(const float* partial, int nblocks,
+                                       float* sums, int C, hipStream_t s) {
+  hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((2 * C + 255) / 256), dim3(256),
+                     0, s, partial, nblocks, sums, C);
 }
 
 extern "C" void launch_bn_apply(const void* x, const void* res, void* y,
@@ -293,17 +322,16 @@ extern "C" void launch_bn_apply(const void* x, const void* res, void* y,
 
 extern "C" void launch_bn_bwd_reduce(const void* dy, const void* y, const void* x,
                                      const float* mean, const float* invstd,
-                                     float* sums, long long M, int C, bool relu,
-                                     hipStream_t s) {
-  const int grid = stats_grid(M, C);
+                                     float* partial, int grid, long long M, int C,
+                                     bool relu, hipStream_t s) {
   if (relu)
     hipLaunchKernelGGL((bn_bwd_reduce_kernel<true>), dim3(grid), dim3(256), 0, s,
                        (const bf16*)dy, (const bf16*)y, (const bf16*)x, mean, invstd,
-                       sums, M, C);
+                       partial, M, C);
   else
     hipLaunchKernelGGL((bn_bwd_reduce_kernel<false>), dim3(grid), dim3(256), 0, s,
                        (const bf16*)dy, nullptr, (const bf16*)x, mean, invstd,
-                       sums, M, C);
+                       partial, M, C);
 }
 
 extern "C" void launch_bn_bwd_dx(const void* dy, const void* y, const void* x,

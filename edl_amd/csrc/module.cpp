@@ -16,16 +16,19 @@ extern "C" void launch_kd_ce_fwd_bf16(const void*, const void*, float*, int, int
                                       hipStream_t);
 extern "C" void launch_kd_ce_bwd_bf16(const void*, const void*, void*, float, int,
                                       int, hipStream_t);
-extern "C" void launch_bn_stats(const void*, float*, long long, int, hipStream_t);
-extern "C" void launch_bn_finalize(const float*, const float*, const float*, float*,
-                                   float*, float*, float*, float*, float*, float,
-                                   float, long long, int, hipStream_t);
+extern "C" int bn_stats_grid(long long, int);
+extern "C" void launch_bn_stats(const void*, float*, int, long long, int,
+                                hipStream_t);
+extern "C" void launch_bn_finalize(const float*, int, const float*, const float*,
+                                   float*, float*, float*, float*, float*, float*,
+                                   float, float, long long, int, hipStream_t);
 extern "C" void launch_bn_apply(const void*, const void*, void*, const float*,
                                 const float*, long long, int, bool, bool,
                                 hipStream_t);
 extern "C" void launch_bn_bwd_reduce(const void*, const void*, const void*,
-                                     const float*, const float*, float*, long long,
-                                     int, bool, hipStream_t);
+                                     const float*, const float*, float*, int,
+                                     long long, int, bool, hipStream_t);
+extern "C" void launch_bn_bwd_finalize(const float*, int, float*, int, hipStream_t);
 extern "C" void launch_bn_bwd_dx(const void*, const void*, const void*, const float*,
                                  const float*, const float*, const float*, void*,
                                  void*, long long, int, bool, bool, bool,
@@ -110,15 +113,16 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor gamma,
   check_bn_inputs(x, C);
   const long long M = x.numel() / C;
   auto opts = gamma.options().dtype(torch::kFloat32);
-  auto sums = torch::zeros({2, C}, opts);
+  const int grid = bn_stats_grid(M, (int)C);
+  auto partial = torch::empty({grid, 2 * C}, opts);
   auto mean = torch::empty({C}, opts);
   auto invstd = torch::empty({C}, opts);
   auto scale = torch::empty({C}, opts);
   auto shift = torch::empty({C}, opts);
   auto y = torch::empty_like(x);
   auto s = cur_stream();
-  launch_bn_stats(x.data_ptr(), sums.data_ptr<float>(), M, (int)C, s);
-  launch_bn_finalize(sums.data_ptr<float>(), gamma.data_ptr<float>(),
+  launch_bn_stats(x.data_ptr(), partial.data_ptr<float>(), grid, M, (int)C, s);
+  launch_bn_finalize(partial.data_ptr<float>(), grid, gamma.data_ptr<float>(),
                      beta.data_ptr<float>(), mean.data_ptr<float>(),
                      invstd.data_ptr<float>(), scale.data_ptr<float>(),
                      shift.data_ptr<float>(),
@@ -150,14 +154,19 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y, torch::Tens
   const int64_t C = gamma.numel();
   check_bn_inputs(x, C);
   const long long M = x.numel() / C;
-  auto sums = torch::zeros({2, C}, gamma.options().dtype(torch::kFloat32));
+  auto opts = gamma.options().dtype(torch::kFloat32);
+  const int grid = bn_stats_grid(M, (int)C);
+  auto partial = torch::empty({grid, 2 * C}, opts);
+  auto sums = torch::empty({2, C}, opts);
   auto dx = torch::empty_like(x);
   auto dres = add ? torch::empty_like(x) : torch::Tensor();
   auto s = cur_stream();
   auto dyc = dy.is_contiguous() ? dy : dy.contiguous();
   launch_bn_bwd_reduce(dyc.data_ptr(), relu ? y.data_ptr() : nullptr, x.data_ptr(),
                        mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                       sums.data_ptr<float>(), M, (int)C, relu, s);
+                       partial.data_ptr<float>(), grid, M, (int)C, relu, s);
+  launch_bn_bwd_finalize(partial.data_ptr<float>(), grid, sums.data_ptr<float>(),
+                         (int)C, s);
   launch_bn_bwd_dx(dyc.data_ptr(), relu ? y.data_ptr() : nullptr, x.data_ptr(),
                    mean.data_ptr<float>(), invstd.data_ptr<float>(),
                    gamma.data_ptr<float>(), sums.data_ptr<float>(), dx.data_ptr(),

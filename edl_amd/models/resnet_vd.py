@@ -14,12 +14,13 @@ import torch
 import torch.nn as nn
 
 from ..ops.bnrelu import BNAddReLU2d, BNReLU2d
+from ..ops.conv import Conv2dFast
 
 
 class ConvBN(nn.Module):
     def __init__(self, cin, cout, k, stride=1, act=True):
         super().__init__()
-        self.conv = nn.Conv2d(cin, cout, k, stride=stride, padding=(k - 1) // 2,
+        self.conv = Conv2dFast(cin, cout, k, stride=stride, padding=(k - 1) // 2,
                               bias=False)
         self.bn = BNReLU2d(cout, act=act)
 
@@ -59,7 +60,7 @@ class BottleneckVd(nn.Module):
         cout = planes * self.expansion
         self.conv0 = ConvBN(cin, planes, 1)
         self.conv1 = ConvBN(planes, planes, 3, stride=stride)
-        self.conv2 = nn.Conv2d(planes, cout, 1, bias=False)
+        self.conv2 = Conv2dFast(planes, cout, 1, bias=False)
         self.bn_add = BNAddReLU2d(cout)  # relu(bn(conv2) + shortcut), fused
         self.shortcut = VdShortcut(cin, cout, stride, if_first)
 
@@ -76,7 +77,7 @@ class BasicBlockVd(nn.Module):
         super().__init__()
         cout = planes * self.expansion
         self.conv0 = ConvBN(cin, planes, 3, stride=stride)
-        self.conv1 = nn.Conv2d(planes, cout, 3, padding=1, bias=False)
+        self.conv1 = Conv2dFast(planes, cout, 3, padding=1, bias=False)
         self.bn_add = BNAddReLU2d(cout)
         self.shortcut = VdShortcut(cin, cout, stride, if_first)
 

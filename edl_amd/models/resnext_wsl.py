@@ -8,6 +8,7 @@ CDNA4 kernels (edl_amd.ops.bnrelu) on GPU."""
 import torch.nn as nn
 
 from ..ops.bnrelu import BNAddReLU2d, BNReLU2d
+from ..ops.conv import Conv2dFast
 
 
 class ResNeXtBottleneck(nn.Module):
@@ -17,17 +18,17 @@ class ResNeXtBottleneck(nn.Module):
         super().__init__()
         width = int(planes * (base_width / 64.0)) * groups
         cout = planes * self.expansion
-        self.conv1 = nn.Conv2d(cin, width, 1, bias=False)
+        self.conv1 = Conv2dFast(cin, width, 1, bias=False)
         self.bn1 = BNReLU2d(width)
-        self.conv2 = nn.Conv2d(width, width, 3, stride=stride, padding=1,
+        self.conv2 = Conv2dFast(width, width, 3, stride=stride, padding=1,
                                groups=groups, bias=False)
         self.bn2 = BNReLU2d(width)
-        self.conv3 = nn.Conv2d(width, cout, 1, bias=False)
+        self.conv3 = Conv2dFast(width, cout, 1, bias=False)
         self.bn_add = BNAddReLU2d(cout)
         self.downsample = None
         if stride != 1 or cin != cout:
             self.downsample = nn.Sequential(
-                nn.Conv2d(cin, cout, 1, stride=stride, bias=False),
+                Conv2dFast(cin, cout, 1, stride=stride, bias=False),
                 BNReLU2d(cout, act=False),
             )
 
@@ -41,7 +42,7 @@ class ResNeXtBottleneck(nn.Module):
 class ResNeXtWSL(nn.Module):
     def __init__(self, depths=(3, 4, 23, 3), groups=32, base_width=16, num_classes=1000):
         super().__init__()
-        self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
+        self.conv1 = Conv2dFast(3, 64, 7, stride=2, padding=3, bias=False)
         self.bn1 = BNReLU2d(64)
         self.maxpool = nn.MaxPool2d(3, 2, padding=1)
         cin = 64
