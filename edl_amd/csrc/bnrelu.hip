@@ -100,10 +100,27 @@ extern "C" __global__ void bn_finalize_kernel(
     const long long M, const int C) {
   const int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
+  // 8 independent accumulators: a single s+= chain leaves every load
+  // latency-exposed (measured 125 us at nblocks=1024 vs ~5 us unrolled)
+  float sa[8] = {0}, qa[8] = {0};
+  const long long st = 2 * C;
+  int b = 0;
+  for (; b + 8 <= nblocks; b += 8) {
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      sa[u] += partial[(long long)(b + u) * st + c];
+      qa[u] += partial[(long long)(b + u) * st + C + c];
+    }
+  }
+  for (; b < nblocks; ++b) {
+    sa[0] += partial[(long long)b * st + c];
+    qa[0] += partial[(long long)b * st + C + c];
+  }
   float s = 0.0f, q = 0.0f;
-  for (int b = 0; b < nblocks; ++b) {
-    s += partial[(long long)b * 2 * C + c];
-    q += partial[(long long)b * 2 * C + C + c];
+#pragma unroll
+  for (int u = 0; u < 8; ++u) {
+    s += sa[u];
+    q += qa[u];
   }
   const float inv_m = 1.0f / (float)M;
   const float mean = s * inv_m;
@@ -212,9 +229,16 @@ extern "C" __global__ void bn_bwd_finalize_kernel(
     float* __restrict__ sums, const int C) {
   const int i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= 2 * C) return;
-  float s = 0.0f;
-  for (int b = 0; b < nblocks; ++b) s += partial[(long long)b * 2 * C + i];
-  sums[i] = s;
+  const long long st = 2 * C;
+  float sa[8] = {0};
+  int b = 0;
+  for (; b + 8 <= nblocks; b += 8) {
+#pragma unroll
+    for (int u = 0; u < 8; ++u) sa[u] += partial[(long long)(b + u) * st + i];
+  }
+  for (; b < nblocks; ++b) sa[0] += partial[(long long)b * st + i];
+  sums[i] = ((sa[0] + sa[1]) + (sa[2] + sa[3])) +
+            ((sa[4] + sa[5]) + (sa[6] + sa[7]));
 }
 
 // (no separate dgamma/dbeta kernel: the bwd-reduce workspace IS [dbeta; dgamma]
@@ -270,8 +294,8 @@ extern "C" int bn_stats_grid(long long M, int C) {
   const int c8 = C >> 3;
   long long rows_per_block = 256 / c8 > 0 ? 256 / c8 : 1;
   long long want = (M + rows_per_block - 1) / rows_per_block;
-  long long cap = 262144 / (2 * (long long)C);  // 1 MiB of partials
-  if (cap > 1024) cap = 1024;
+  long long cap = 131072 / (2 * (long long)C);  // 512 KiB of partials
+  if (cap > 512) cap = 512;
   if (cap < 8) cap = 8;
   long long g = want < cap ? want : cap;
   return (int)(g > 0 ? g : 1);

@@ -14,12 +14,20 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 
+import os
+
+# 1x1 path selector for A/B measurement: "matmul" (hipBLASLt/rocBLAS via
+# torch.matmul) or "miopen" (F.conv2d).
+_CONV1X1 = os.environ.get("EDL_CONV1X1", "matmul")
+
+
 class Conv2dFast(nn.Conv2d):
     """nn.Conv2d drop-in; 1x1/group-1 convs on CUDA go through matmul."""
 
     def forward(self, x):
         if (
-            x.is_cuda
+            _CONV1X1 == "matmul"
+            and x.is_cuda
             and self.kernel_size == (1, 1)
             and self.groups == 1
             and self.padding == (0, 0)
