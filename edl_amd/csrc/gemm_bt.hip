@@ -1,0 +1,167 @@
+// Hand-written bf16 MFMA GEMM, B-transposed layout:
+//     C[M, N] = A[M, K] @ B[N, K]^T        (all row-major, bf16 in/out)
+//
+// This is the conv1x1 shape on NHWC: fwd  y = x2d @ W^T   (W: [Cout, Cin])
+//                                    dgrad dx = dy2d @ (W^T)^T via a [Cin,
+//                                    Cout] transposed-weight copy — the same
+//                                    kernel.  MIOpen's igemm ran these at
+//                                    ~60 TF and hipBLASLt's heuristic picks
+//                                    were worse (rocprof gpurun_out/prof3/4);
+//                                    dense bf16 peak is 2.5 PF.
+//
+// Structure (cdna_hip_programming.md §5 "step-3" class):
+//   * 256 threads = 4 waves; per-wave 64x64 output; v_mfma_f32_16x16x32_bf16
+//   * BK = 64; A/B tiles staged to LDS by global_load_lds_dwordx4 (16 B),
+//     double-buffered; one __syncthreads per K-tile (its implicit vmcnt(0)
+//     drains the in-flight glds of the next tile — the 2-phase recipe)
+//   * LDS XOR swizzle (byte slot ^= row&7) on the glds SOURCE address and
+//     the ds_read address (rule 21: linear dest + inverse-swizzled source)
+//     -> conflict-free ds_read_b128 fragments
+//   * row-clamped A loads for the M tail (C-write is row-guarded)
+//   * two tile shapes: 128x128 (N % 128 == 0) and 256x64
+#include "common.h"
+
+using bf16 = __hip_bfloat16;
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+extern __shared__ __attribute__((aligned(16))) char smem[];
+
+template <int BM, int BN, int WAVES_M, int WAVES_N>
+__global__ __launch_bounds__(256, 2) void gemm_bt_kernel(
+    const bf16* __restrict__ A, const bf16* __restrict__ B,
+    bf16* __restrict__ C, const int M, const int N, const int K) {
+  constexpr int BK = 64;
+  constexpr int A_BYTES = BM * BK * 2;  // row stride 128 B
+  constexpr int B_BYTES = BN * BK * 2;
+  // LDS: [A0 | B0 | A1 | B1]
+  char* lds = smem;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+
+  // tile coordinates (grid.x = tiles_m * tiles_n, n fastest)
+  const int tiles_n = N / BN;
+  const int tile_m = blockIdx.x / tiles_n;
+  const int tile_n = blockIdx.x % tiles_n;
+  const int m0 = tile_m * BM;
+  const int n0 = tile_n * BN;
+
+  const int wm = (wave / WAVES_N) * 64;  // wave's sub-tile origin
+  const int wn = (wave % WAVES_N) * 64;
+
+  f32x4 acc[4][4] = {};
+
+  // ---- staging: each wave stages its share of A and B chunks ----
+  // chunk = 1 KiB = 8 rows x 8 slots(16 B); lane l -> row l/8, slot l%8
+  constexpr int A_CHUNKS = A_BYTES / 1024;
+  constexpr int B_CHUNKS = B_BYTES / 1024;
+  auto stage = [&](int buf, int kt) {
+    const long long k0 = (long long)kt * BK;
+    char* abase = lds + buf * (A_BYTES + B_BYTES);
+    char* bbase = abase + A_BYTES;
+#pragma unroll
+    for (int i = 0; i < A_CHUNKS / 4; ++i) {
+      const int ch = wave * (A_CHUNKS / 4) + i;
+      const int r = ch * 8 + (lane >> 3);
+      const int slot = lane & 7;
+      const int gslot = slot ^ (r & 7);
+      long long grow = m0 + r;
+      if (grow >= M) grow = M - 1;  // clamped dup row; C-write is guarded
+      const bf16* src = A + grow * K + k0 + gslot * 8;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)(abase + ch * 1024), 16, 0, 0);
+    }
+#pragma unroll
+    for (int i = 0; i < B_CHUNKS / 4; ++i) {
+      const int ch = wave * (B_CHUNKS / 4) + i;
+      const int r = ch * 8 + (lane >> 3);
+      const int slot = lane & 7;
+      const int gslot = slot ^ (r & 7);
+      const bf16* src = B + (long long)(n0 + r) * K + k0 + gslot * 8;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)(bbase + ch * 1024), 16, 0, 0);
+    }
+  };
+
+  // ---- fragment reads (swizzled ds_read_b128) ----
+  auto read_a = [&](int buf, int mf, int kk) -> bf16x8 {
+    const char* abase = lds + buf * (A_BYTES + B_BYTES);
+    const int r = wm + mf * 16 + (lane & 15);
+    const int c = kk * 4 + (lane >> 4);
+    return *(const __attribute__((address_space(3))) bf16x8*)(
+        (const __attribute__((address_space(3))) char*)(abase) + r * 128 +
+        ((c ^ (r & 7)) << 4));
+  };
+  auto read_b = [&](int buf, int nf, int kk) -> bf16x8 {
+    const char* bbase = lds + buf * (A_BYTES + B_BYTES) + A_BYTES;
+    const int r = wn + nf * 16 + (lane & 15);
+    const int c = kk * 4 + (lane >> 4);
+    return *(const __attribute__((address_space(3))) bf16x8*)(
+        (const __attribute__((address_space(3))) char*)(bbase) + r * 128 +
+        ((c ^ (r & 7)) << 4));
+  };
+
+  const int KT = K / BK;
+  stage(0, 0);
+  __syncthreads();
+
+  for (int kt = 0; kt < KT; ++kt) {
+    const int cur = kt & 1;
+    if (kt + 1 < KT) stage(cur ^ 1, kt + 1);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      bf16x8 a[4], b[4];
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf) a[mf] = read_a(cur, mf, kk);
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) b[nf] = read_b(cur, nf, kk);
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < 4; ++nf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[mf], b[nf], acc[mf][nf], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: C[m][n] bf16; frag C/D map col=lane&15,
+  // row=(lane>>4)*4+reg ----
+  const int cn = lane & 15;
+  const int r4 = (lane >> 4) * 4;
+#pragma unroll
+  for (int mf = 0; mf < 4; ++mf) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int m = m0 + wm + mf * 16 + r4 + reg;
+      if (m < M) {
+        bf16* crow = C + (long long)m * N + n0 + wn + cn;
+#pragma unroll
+        for (int nf = 0; nf < 4; ++nf) {
+          crow[nf * 16] = __float2bfloat16(acc[mf][nf][reg]);
+        }
+      }
+    }
+  }
+}
+
+extern "C" void launch_gemm_bt(const void* A, const void* B, void* C, int M,
+                               int N, int K, hipStream_t s) {
+  if (N % 128 == 0) {
+    constexpr int BM = 128, BN = 128;
+    const int grid = ((M + BM - 1) / BM) * (N / BN);
+    const int lds_bytes = 2 * (BM * 64 * 2 + BN * 64 * 2);
+    hipLaunchKernelGGL((gemm_bt_kernel<BM, BN, 2, 2>), dim3(grid), dim3(256),
+                       lds_bytes, s, (const bf16*)A, (const bf16*)B, (bf16*)C,
+                       M, N, K);
+  } else {  // N % 64 == 0
+    constexpr int BM = 256, BN = 64;
+    const int grid = ((M + BM - 1) / BM) * (N / BN);
+    const int lds_bytes = 2 * (BM * 64 * 2 + BN * 64 * 2);
+    hipLaunchKernelGGL((gemm_bt_kernel<BM, BN, 4, 1>), dim3(grid), dim3(256),
+                       lds_bytes, s, (const bf16*)A, (const bf16*)B, (bf16*)C,
+                       M, N, K);
+  }
+}

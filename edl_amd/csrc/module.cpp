@@ -29,6 +29,8 @@ extern "C" void launch_bn_bwd_reduce(const void*, const void*, const void*,
                                      const float*, const float*, float*, int,
                                      long long, int, bool, hipStream_t);
 extern "C" void launch_bn_bwd_finalize(const float*, int, float*, int, hipStream_t);
+extern "C" void launch_gemm_bt(const void*, const void*, void*, int, int, int,
+                               hipStream_t);
 extern "C" void launch_bn_bwd_dx(const void*, const void*, const void*, const float*,
                                  const float*, const float*, const float*, void*,
                                  void*, long long, int, bool, bool, bool,
@@ -176,6 +178,24 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y, torch::Tens
   return {dx, sums[1], sums[0], dres};
 }
 
+torch::Tensor gemm_bt(torch::Tensor a, torch::Tensor b) {
+  // C[M, N] = A[M, K] @ B[N, K]^T, bf16 row-major
+  TORCH_CHECK(a.is_cuda() && b.is_cuda(), "gemm_bt: GPU tensors required");
+  TORCH_CHECK(a.scalar_type() == torch::kBFloat16 &&
+                  b.scalar_type() == torch::kBFloat16,
+              "gemm_bt: bf16 only");
+  TORCH_CHECK(a.dim() == 2 && b.dim() == 2 && a.size(1) == b.size(1),
+              "gemm_bt: [M,K] x [N,K]");
+  auto ac = a.contiguous();
+  auto bc = b.contiguous();
+  const int M = (int)a.size(0), K = (int)a.size(1), N = (int)b.size(0);
+  TORCH_CHECK(K % 64 == 0 && N % 64 == 0, "gemm_bt: K,N % 64 == 0");
+  auto c = torch::empty({M, N}, a.options());
+  launch_gemm_bt(ac.data_ptr(), bc.data_ptr(), c.data_ptr(), M, N, K,
+                 cur_stream());
+  return c;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -188,5 +208,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_fwd_eval", &bn_fwd_eval, "fused NHWC bf16 BN(+add)+ReLU eval fwd");
   m.def("bn_bwd", &bn_bwd,
         "fused BN(+add)+ReLU bwd -> (dx, dgamma, dbeta, dres?)");
+  m.def("gemm_bt", &gemm_bt, "bf16 MFMA GEMM: C[M,N] = A[M,K] @ B[N,K]^T");
   m.attr("_arch") = "gfx950";
 }

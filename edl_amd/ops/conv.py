@@ -16,17 +16,38 @@ import torch.nn.functional as F
 
 import os
 
-# 1x1 path selector for A/B measurement: "matmul" (hipBLASLt/rocBLAS via
-# torch.matmul) or "miopen" (F.conv2d).
-_CONV1X1 = os.environ.get("EDL_CONV1X1", "matmul")
+from . import available, ext
+
+# 1x1 path selector: "hip" (our MFMA gemm_bt kernel), "matmul"
+# (hipBLASLt/rocBLAS via torch.matmul), "miopen" (F.conv2d).
+_CONV1X1 = os.environ.get("EDL_CONV1X1", "hip")
+
+
+class _Conv1x1Hip(torch.autograd.Function):
+    """y2d = x2d @ W^T via the gemm_bt MFMA kernel; dgrad reuses the same
+    kernel on a transposed-weight copy; wgrad (TN shape) goes through
+    torch.matmul until the TN kernel lands."""
+
+    @staticmethod
+    def forward(ctx, x2d, w):  # w: [Cout, Cin] bf16
+        ctx.save_for_backward(x2d, w)
+        return ext().gemm_bt(x2d, w)
+
+    @staticmethod
+    def backward(ctx, dy2d):
+        x2d, w = ctx.saved_tensors
+        dy2d = dy2d.contiguous()
+        dx = ext().gemm_bt(dy2d, w.t().contiguous())
+        dw = dy2d.t() @ x2d  # TN: [Cout, M] @ [M, Cin]
+        return dx, dw
 
 
 class Conv2dFast(nn.Conv2d):
-    """nn.Conv2d drop-in; 1x1/group-1 convs on CUDA go through matmul."""
+    """nn.Conv2d drop-in; 1x1/group-1 convs on CUDA bypass MIOpen."""
 
     def forward(self, x):
         if (
-            _CONV1X1 == "matmul"
+            _CONV1X1 != "miopen"
             and x.is_cuda
             and self.kernel_size == (1, 1)
             and self.groups == 1
@@ -42,7 +63,18 @@ class Conv2dFast(nn.Conv2d):
                 x = x.contiguous(memory_format=torch.channels_last)
             x2d = x.permute(0, 2, 3, 1).reshape(n * h * w, c)
             wt = self.weight.view(self.out_channels, c)
-            y2d = x2d @ wt.t()
+            if (
+                _CONV1X1 == "hip"
+                and available()
+                and x2d.dtype == torch.bfloat16
+                and c % 64 == 0
+                and self.out_channels % 64 == 0
+            ):
+                if wt.dtype != torch.bfloat16:
+                    wt = wt.to(torch.bfloat16)
+                y2d = _Conv1x1Hip.apply(x2d, wt)
+            else:
+                y2d = x2d @ wt.t()
             return (
                 y2d.view(n, h, w, self.out_channels).permute(0, 3, 1, 2)
             )
