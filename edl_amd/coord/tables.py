@@ -14,6 +14,9 @@ ETCD_STATE = "state"                # train State (checkpoint metadata)
 ETCD_BARRIER = "barrier"            # barrier arrivals per stage
 ETCD_DIST_READER = "dist_reader"    # distributed reader leader endpoint
 ETCD_SERVICE = "service"            # distill teacher registry (per service name)
+ETCD_SERVICE_CLIENTS = "service_clients"  # distill student registrations
+ETCD_SERVICE_ASSIGN = "service_assign"    # balance output: client -> teachers
+ETCD_BALANCE = "balance"            # discovery-server self-registry (__balance__)
 
 import os
 
