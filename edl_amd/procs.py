@@ -48,7 +48,7 @@ class TrainerProcs:
             log_path = os.path.join(self._job_env.log_dir, "workerlog.%d" % t.rank_in_pod)
             f = open(log_path, "ab", buffering=0)
             cmd = self._cmd
-            if cmd and cmd[0].endswith(".py"):
+            if cmd and (cmd[0].endswith(".py") or cmd[0] == "-m"):
                 cmd = [sys.executable, "-u"] + cmd
             proc = subprocess.Popen(
                 cmd, env=env, stdout=f, stderr=subprocess.STDOUT, start_new_session=True

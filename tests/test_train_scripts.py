@@ -1,0 +1,66 @@
+"""End-to-end trainer-script tests under the real launcher on CPU:
+fit_a_line (BASELINE config 1: collective on gloo world 2) and ctr."""
+import os
+import subprocess
+import sys
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def run_edlrun(coord_server, tmp_path, script_args, n_agents=2, timeout=180):
+    procs = []
+    env = dict(os.environ)
+    env.update({
+        "PYTHONPATH": REPO + os.pathsep + env.get("PYTHONPATH", ""),
+        "EDL_LEASE_TTL": "2",
+        "EDL_LEADER_RETRY": "0.5",
+        "CUDA_VISIBLE_DEVICES": "",
+    })
+    for i in range(n_agents):
+        logf = open(tmp_path / ("agent%d.log" % i), "wb")
+        p = subprocess.Popen(
+            [sys.executable, "-m", "edl_amd.launch",
+             "--job_id", "script_job", "--store_endpoints", coord_server.endpoint,
+             "--nodes_range", "%d:%d" % (n_agents, n_agents),
+             "--nproc_per_node", "1",
+             "--log_dir", str(tmp_path / ("logs%d" % i)), "--"] + script_args,
+            env=env, stdout=logf, stderr=subprocess.STDOUT, cwd=REPO,
+            start_new_session=True)
+        p._logf = logf
+        procs.append(p)
+    try:
+        for p in procs:
+            assert p.wait(timeout=timeout) == 0, \
+                (tmp_path / "agent0.log").read_text() + \
+                (tmp_path / ("agent%d.log" % (n_agents - 1))).read_text()
+    finally:
+        for p in procs:
+            if p.poll() is None:
+                p.kill()
+            p._logf.close()
+
+
+@pytest.mark.parametrize("model", ["fit_a_line", "ctr"])
+def test_simple_trainer_world2(coord_server, tmp_path, model):
+    run_edlrun(
+        coord_server, tmp_path,
+        ["-m", "edl_amd.train.train_simple", "--model", model,
+         "--num_epochs", "1", "--steps_per_epoch", "20",
+         "--checkpoint", str(tmp_path / "ck")],
+    )
+    assert (tmp_path / "ck" / "checkpoint.0").is_dir()
+
+
+def test_resnet_trainer_world2_tiny(coord_server, tmp_path):
+    """The flagship trainer end-to-end under the launcher on CPU (tiny)."""
+    run_edlrun(
+        coord_server, tmp_path,
+        ["-m", "edl_amd.train.train_resnet", "--model", "resnet18_vd",
+         "--batch_size", "2", "--num_epochs", "1", "--steps_per_epoch", "2",
+         "--checkpoint", str(tmp_path / "ck"), "--use_hip_ops", "0",
+         "--dtype", "fp32"],
+        timeout=420,
+    )
+    assert (tmp_path / "ck" / "checkpoint.0").is_dir()
