@@ -1,0 +1,92 @@
+"""Distill throughput benchmark on one node.
+
+Configs (reference README.md:84-85):
+  --shared : teacher + student on the same GPU(s)      (ref: 656 img/s)
+  default  : service distill — teachers on their own process/GPU slice
+             (ref EDL service distill: 1514 img/s whole node)
+
+Single-GPU form (gpurun): both processes share cuda:0.
+
+    python tools/bench_distill.py --steps 20 --batch_size 32
+Prints one JSON line with img/s (whole job).
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+sys.path.insert(0, REPO)
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--batch_size", type=int, default=32)
+    ap.add_argument("--teacher_batch_size", type=int, default=16)
+    ap.add_argument("--require_num", type=int, default=1)
+    ap.add_argument("--teacher_model", default="resnext101_32x16d_wsl")
+    args = ap.parse_args()
+
+    import torch
+
+    from edl_amd.distill.reader import DistillReader
+    from edl_amd.distill.teacher_server import TeacherServer, TeacherService
+    from edl_amd.train.engine import TrainerEngine
+
+    svc = TeacherService(args.teacher_model, 1000)
+    srv = TeacherServer(svc, "127.0.0.1", 0).start()
+    engine = TrainerEngine(model="resnet50_vd", per_device_batch=args.batch_size,
+                           base_lr=0.01, use_hip_ops=torch.cuda.is_available(),
+                           graph_capture=False, kd_alpha=1.0).setup()
+    engine.model.train()
+    total = args.warmup + args.steps
+    rng = np.random.RandomState(0)
+    batches = [(rng.randn(args.batch_size, 3, 224, 224).astype(np.float32),
+                rng.randint(0, 1000, (args.batch_size,)).astype(np.int64))
+               for _ in range(4)]
+
+    def batch_gen():
+        for i in range(total):
+            yield batches[i % len(batches)]
+
+    dr = DistillReader(["img", "label"], ["logits"],
+                       teacher_batch_size=args.teacher_batch_size,
+                       require_num=args.require_num)
+    dr.set_batch_generator(batch_gen)
+    dr.set_fixed_teacher(["127.0.0.1:%d" % srv.port])
+
+    n = 0
+    t0 = None
+    for img, label, logits in dr():
+        x = torch.from_numpy(img).to(engine.device)
+        if engine.device.type == "cuda":
+            x = x.to(torch.bfloat16).contiguous(memory_format=torch.channels_last)
+        y = torch.from_numpy(label).to(engine.device)
+        t = torch.from_numpy(np.ascontiguousarray(logits)).to(engine.device)
+        engine.train_step(x, y, teacher_logits=t)
+        n += 1
+        if n == args.warmup:
+            if engine.device.type == "cuda":
+                torch.cuda.synchronize()
+            t0 = time.monotonic()
+    if engine.device.type == "cuda":
+        torch.cuda.synchronize()
+    dt = time.monotonic() - t0
+    timed = n - args.warmup
+    print(json.dumps({
+        "metric": "img/s", "value": round(timed * args.batch_size / dt, 1),
+        "unit": "img/s", "mode": "distill_shared_1gpu",
+        "steps": timed, "ms_per_step": round(dt / timed * 1e3, 2),
+        "teacher": args.teacher_model, "dtype": "bf16", "data": "synthetic",
+        "vs_baseline_shared_656": round(timed * args.batch_size / dt / 656.0, 3),
+    }))
+    srv.stop()
+
+
+if __name__ == "__main__":
+    main()
