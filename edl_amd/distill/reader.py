@@ -12,11 +12,11 @@ Env overrides (reference distill_reader.py:278-298):
     PADDLE_DISTILL_BALANCE_SERVER   (store endpoints for dynamic discovery)
     PADDLE_DISTILL_SERVICE_NAME
     PADDLE_DISTILL_MAX_TEACHER      (require_num)
-Spawns 1 reader process + require_num predict processes + a manage thread
+Runs 1 reader thread + require_num predict threads + a manage thread
 that diffs the discovered teacher set, stops workers of removed teachers
 and fills slots for new ones (reference predict_manage_worker 58-171)."""
-import multiprocessing as mp
 import os
+import queue
 import threading
 
 import numpy as np
@@ -42,7 +42,6 @@ class DistillReader:
         self._discovery = None
         self._store_endpoints = os.environ.get("PADDLE_DISTILL_BALANCE_SERVER")
         self._service_name = os.environ.get("PADDLE_DISTILL_SERVICE_NAME")
-        self._ctx = mp.get_context("fork")
 
     # ---- configuration (reference 307-353) ----
     def set_sample_generator(self, gen_factory):
@@ -101,15 +100,14 @@ class DistillReader:
     def __call__(self):
         assert self._gen_factory is not None, "set a generator first"
         n = self._require_num
-        ctx = self._ctx
-        task_queue = ctx.Queue(maxsize=4 * n + 8)
-        out_queue = ctx.Queue()
-        server_queue = ctx.Queue()
-        task_semaphore = ctx.Semaphore(2 * n + 2)
-        reader_stop = ctx.Event()
-        stop_events = [ctx.Event() for _ in range(n)]
-        predict_count = ctx.Value("l", 0)
-        live_workers = ctx.Value("l", 0)
+        task_queue = queue.Queue(maxsize=4 * n + 8)
+        out_queue = queue.Queue()
+        server_queue = queue.Queue()
+        task_semaphore = threading.Semaphore(2 * n + 2)
+        reader_stop = threading.Event()
+        stop_events = [threading.Event() for _ in range(n)]
+        predict_count = W.Counter()
+        live_workers = W.Counter()
 
         if self._fixed_teachers is None:
             self._discovery = DiscoveryClient(
@@ -130,7 +128,7 @@ class DistillReader:
             _t.sleep(0.5)
 
         procs = []
-        reader = ctx.Process(
+        reader = threading.Thread(
             target=W.reader_worker,
             args=(self._gen_factory, self._mode, self._teacher_batch_size,
                   task_queue, task_semaphore, reader_stop),
@@ -140,7 +138,7 @@ class DistillReader:
         with live_workers.get_lock():
             live_workers.value = n
         for slot in range(n):
-            p = ctx.Process(
+            p = threading.Thread(
                 target=W.predict_worker,
                 args=(slot, server_queue, task_queue, out_queue, predict_count,
                       live_workers, stop_events, self._feed_idx),
@@ -156,7 +154,7 @@ class DistillReader:
             with live_workers.get_lock():
                 live_workers.value += 1
             assigned[slot] = None
-            p = ctx.Process(
+            p = threading.Thread(
                 target=W.predict_worker,
                 args=(slot, server_queue, task_queue, out_queue, predict_count,
                       live_workers, stop_events, self._feed_idx),
@@ -193,12 +191,13 @@ class DistillReader:
             manage_stop.set()
             for e in stop_events:
                 e.set()
+            # unblock a reader waiting on flow control so its thread exits
+            for _ in range(4 * n + 8):
+                task_semaphore.release()
             for _ in procs:
                 server_queue.put(None)
             for p in [reader] + procs:
                 p.join(timeout=5)
-                if p.is_alive():
-                    p.terminate()
             if self._discovery:
                 self._discovery.stop()
                 self._discovery = None

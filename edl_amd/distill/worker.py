@@ -16,8 +16,8 @@ Processes and protocol:
 The NOP teacher (reference _TestNopPaddlePredictServer, 324-333) lets the
 whole 1+N-process pipeline run in CPU tests: predictions == the feed.
 """
-import multiprocessing as mp
 import queue as pyqueue
+import threading
 import time
 
 import numpy as np
@@ -28,6 +28,19 @@ from .timeline import timeline
 log = get_logger("edl.distill.worker")
 
 _NOP_PREDICT_TEST = False  # set True in tests (reference distill_worker.py:36-43)
+
+
+class Counter:
+    """Shared counter with a lock (the pipeline runs on THREADS: forking
+    after CUDA/HIP init hangs the child runtime — observed on MI355X — and
+    the workers are RPC/IO-bound anyway, so threads are the right tool)."""
+
+    def __init__(self, v=0):
+        self.value = v
+        self._lock = threading.Lock()
+
+    def get_lock(self):
+        return self._lock
 
 
 class _PoisonPill:
