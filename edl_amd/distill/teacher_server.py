@@ -43,10 +43,13 @@ class TeacherService:
     def predict(self, images):
         """images: numpy [B, 3, H, W] float32 -> logits numpy [B, classes]."""
         x = torch.from_numpy(np.ascontiguousarray(images)).to(self.device)
-        if self.use_bf16:
-            x = x.to(torch.bfloat16).contiguous(memory_format=torch.channels_last)
         with self._lock:
-            y = self.model(x)
+            if self.use_bf16:
+                x = x.contiguous(memory_format=torch.channels_last)
+                with torch.autocast("cuda", torch.bfloat16):
+                    y = self.model(x)
+            else:
+                y = self.model(x)
         return y.float().cpu().numpy()
 
 

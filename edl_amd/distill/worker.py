@@ -182,7 +182,7 @@ def _predict_loop(slot, item, client, task_queue, out_queue, predict_count,
             raise
 
 
-def fetch_ordered(out_queue, task_semaphore, on_worker_exit=None, timeout=300.0):
+def fetch_ordered(out_queue, task_semaphore, on_worker_exit=None, timeout=120.0):
     """Generator of (samples, predictions) in task_id order. Terminates
     after the pill AND all preceding tasks have been yielded."""
     store = {}
