@@ -26,7 +26,10 @@ def kd_soft_cross_entropy(student_logits, teacher_logits):
     reference distill example (soft_label cross_entropy,
     example/distill/resnet/train_with_fleet.py:254-259)."""
     if student_logits.is_cuda and available():
-        return _KDSoftCE.apply(student_logits, teacher_logits.detach())
+        t = teacher_logits.detach()
+        if t.dtype != student_logits.dtype:
+            t = t.to(student_logits.dtype)
+        return _KDSoftCE.apply(student_logits, t)
     # torch reference (CPU tests / numerics baseline)
     logp = torch.log_softmax(student_logits.float(), dim=1)
     soft = torch.softmax(teacher_logits.detach().float(), dim=1)
