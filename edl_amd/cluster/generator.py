@@ -62,6 +62,18 @@ class ClusterGenerator:
         statuses = load_pods_status(self._client)
         current = load_cluster(self._client)
 
+        # external scale-in requests (cluster/scale.py): drop named pods
+        from .scale import clear_scale_request, read_scale_request
+
+        req = read_scale_request(self._client)
+        removed_by_request = set()
+        if req:
+            removed_by_request = set(req.get("remove_pods") or [])
+            if removed_by_request:
+                for pid in removed_by_request:
+                    resource.pop(pid, None)
+                clear_scale_request(self._client)
+
         if current is None:
             return self._first_boot(resource)
 

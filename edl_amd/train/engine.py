@@ -57,6 +57,7 @@ class TrainerEngine:
         graph_capture=None,
         kd_teacher=None,
         kd_alpha=1.0,
+        recompute=False,
     ):
         self.model_name = model
         self.per_device_batch = per_device_batch
@@ -74,6 +75,8 @@ class TrainerEngine:
         self.graph_capture = graph_capture
         self.kd_teacher = kd_teacher  # callable(images)->soft logits, or None
         self.kd_alpha = kd_alpha
+        self.recompute = recompute  # activation checkpointing (reference
+        # dist_strategy.forward_recompute, train_with_fleet.py:322-325)
 
         self.env = None
         self.device = None
@@ -98,6 +101,22 @@ class TrainerEngine:
             from .. import ops
 
             ops.swap_module_ops(self.model)
+        if self.recompute and hasattr(self.model, "stages"):
+            import torch.utils.checkpoint as ckpt_mod
+
+            class _Recompute(torch.nn.Module):
+                def __init__(self, inner):
+                    super().__init__()
+                    self.inner = inner
+
+                def forward(self, x):
+                    if self.training and torch.is_grad_enabled():
+                        return ckpt_mod.checkpoint(self.inner, x, use_reentrant=False)
+                    return self.inner(x)
+
+            self.model.stages = torch.nn.Sequential(
+                *[_Recompute(s) for s in self.model.stages]
+            )
 
         self.reducer = BucketedAllReducer(
             self.model.parameters(), bucket_cap_mb=self.bucket_mb

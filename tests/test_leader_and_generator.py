@@ -169,3 +169,32 @@ def test_barrier_scale_in_raises(coord_client):
     with pytest.raises(EdlPodIDNotExistError):
         barrier(coord_client, "not_a_member", timeout=5)
     reg.stop()
+
+
+def test_scale_in_request_drops_pod(coord_server, coord_client):
+    """External scale-in API (reference PodServer.ScaleIn): the generator
+    drops the named pod on its next pass."""
+    from edl_amd.cluster.scale import read_scale_request, request_scale
+    from edl_amd.coord.client import CoordClient
+
+    pod_a, pod_b = make_pod("a"), make_pod("b")
+    reg_a = ResourceRegister(coord_client, pod_a).start()
+    cb = CoordClient(coord_server.endpoint, "test_job")
+    reg_b = ResourceRegister(cb, pod_b).start()
+    save_pod_status(coord_client, "a", Status.INITIAL)
+    save_pod_status(cb, "b", Status.INITIAL)
+    coord_client.put(coord_client.table_key(tables.ETCD_POD_RANK, "0"), "a")
+    gen = ClusterGenerator(coord_client, "a", min_nodes=1, period=0.1)
+    c = gen.generate_once()
+    assert c.pod_ids() == ["a", "b"]
+
+    save_pod_status(cb, "b", Status.RUNNING)  # running pods are not re-added
+    request_scale(coord_client, remove_pods=["b"])
+    c2 = gen.generate_once()
+    assert c2.pod_ids() == ["a"]
+    assert read_scale_request(coord_client) is None  # consumed
+    c3 = gen.generate_once()
+    assert c3.pod_ids() == ["a"]  # stays out (status RUNNING, not INITIAL)
+    reg_a.stop()
+    reg_b.stop()
+    cb.close()
