@@ -1,0 +1,211 @@
+// Implicit-GEMM 3x3 convolution (stride 1 or 2, same-pad) on PADDED NHWC
+// bf16 input — the ResNet bottleneck 3x3s (MIOpen's igemm ran them at a
+// few % of MFMA peak).
+//
+//   y[M, Cout] = sum_{s=(dy,dx)} A_s[M, Cin] @ B_s[Cout, Cin]^T
+//
+// where A_s row m reads xp[n, h*stride+dy, w*stride+dx, :] of the padded
+// image xp [N, Hp, Wp, Cin] (halo pre-zeroed by pad_nhwc — no border
+// predicates in the hot loop), and B = W3 [Cout, 9*Cin] (s-major repacked
+// weights) is EXACTLY the gemm_bt B operand with K = 9*Cin.
+//
+// Same engine as gemm_bt: 4 waves, 64x64 per wave, BK=64,
+// global_load_lds 16B staging (XOR source+read swizzle), double-buffered
+// LDS, mfma_f32_16x16x32_bf16. The only delta: each thread PRECOMPUTES
+// the padded-row byte address for the A rows it stages (fixed across the
+// K loop); the k-tile adds shift_off[s] + 128*cb.
+#include "common.h"
+
+using bf16 = __hip_bfloat16;
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+extern __shared__ __attribute__((aligned(16))) char smem[];
+
+// zero-halo pad: xp[n, h+1, w+1, c] = x[n, h, w, c]; halo = 0.
+extern "C" __global__ void pad_nhwc_kernel(
+    const bf16* __restrict__ x, bf16* __restrict__ xp, const int HW_in,
+    const int H, const int W, const int Hp, const int Wp, const int C) {
+  const int c8 = C >> 3;
+  const long long total = (long long)gridDim.y * Hp * Wp * c8;  // per image n
+  // grid: x = flat over Hp*Wp*c8 (strided), y = image index
+  const int n = blockIdx.y;
+  long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  const long long img_elems = (long long)Hp * Wp * c8;
+  for (; i < img_elems; i += stride) {
+    const int oct = (int)(i % c8);
+    const long long pix = i / c8;
+    const int wp = (int)(pix % Wp);
+    const int hp = (int)(pix / Wp);
+    bf16* dst = xp + ((long long)n * Hp * Wp + pix) * C + oct * 8;
+    const int h = hp - 1, w = wp - 1;
+    if (h >= 0 && h < H && w >= 0 && w < W) {
+      const bf16* src = x + (((long long)n * H + h) * W + w) * C + oct * 8;
+      *reinterpret_cast<uint4*>(dst) = *reinterpret_cast<const uint4*>(src);
+    } else {
+      *reinterpret_cast<uint4*>(dst) = uint4{0, 0, 0, 0};
+    }
+  }
+  (void)HW_in;
+  (void)total;
+}
+
+template <int BM, int BN, int WAVES_M, int WAVES_N>
+__global__ __launch_bounds__(256, 2) void conv3x3_kernel(
+    const bf16* __restrict__ XP, const bf16* __restrict__ B,
+    bf16* __restrict__ C_out, const int M, const int N, const int Cin,
+    const int HW_out, const int W_out, const int Hp, const int Wp,
+    const int stride_hw) {
+  constexpr int BK = 64;
+  constexpr int A_BYTES = BM * BK * 2;
+  constexpr int B_BYTES = BN * BK * 2;
+  const int K = 9 * Cin;          // gemm K
+  const int cb_per_s = Cin >> 6;  // 64-wide channel blocks per shift
+  char* lds = smem;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+
+  const int tiles_n = N / BN;
+  const int m0 = (blockIdx.x / tiles_n) * BM;
+  const int n0 = (blockIdx.x % tiles_n) * BN;
+  const int wm = (wave / WAVES_N) * 64;
+  const int wn = (wave % WAVES_N) * 64;
+
+  // ---- per-thread A-row address table (fixed across the K loop) ----
+  constexpr int A_CHUNKS = A_BYTES / 1024;
+  constexpr int B_CHUNKS = B_BYTES / 1024;
+  long long arow[A_CHUNKS / 4];
+#pragma unroll
+  for (int i = 0; i < A_CHUNKS / 4; ++i) {
+    const int ch = wave * (A_CHUNKS / 4) + i;
+    const int r = ch * 8 + (lane >> 3);
+    long long m = m0 + r;
+    if (m >= M) m = M - 1;  // clamped dup row; C-write guarded
+    const int n_img = (int)(m / HW_out);
+    const int rem = (int)(m % HW_out);
+    const int h = rem / W_out;
+    const int w = rem % W_out;
+    arow[i] = (((long long)n_img * Hp + h * stride_hw) * Wp + w * stride_hw) *
+              Cin;  // element offset of (dy=0, dx=0)
+  }
+  int shift_elems[9];
+#pragma unroll
+  for (int s = 0; s < 9; ++s) {
+    shift_elems[s] = ((s / 3) * Wp + (s % 3)) * Cin;
+  }
+
+  auto stage = [&](int buf, int kt) {
+    const int s = kt / cb_per_s;
+    const int cb = kt % cb_per_s;
+    char* abase = lds + buf * (A_BYTES + B_BYTES);
+    char* bbase = abase + A_BYTES;
+#pragma unroll
+    for (int i = 0; i < A_CHUNKS / 4; ++i) {
+      const int ch = wave * (A_CHUNKS / 4) + i;
+      const int r = ch * 8 + (lane >> 3);
+      const int gslot = (lane & 7) ^ (r & 7);
+      const bf16* src = XP + arow[i] + shift_elems[s] + cb * 64 + gslot * 8;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)(abase + ch * 1024), 16, 0, 0);
+    }
+    const long long k0 = (long long)kt * BK;
+#pragma unroll
+    for (int i = 0; i < B_CHUNKS / 4; ++i) {
+      const int ch = wave * (B_CHUNKS / 4) + i;
+      const int r = ch * 8 + (lane >> 3);
+      const int gslot = (lane & 7) ^ (r & 7);
+      const bf16* src = B + (long long)(n0 + r) * K + k0 + gslot * 8;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)(bbase + ch * 1024), 16, 0, 0);
+    }
+  };
+
+  auto read_a = [&](int buf, int mf, int kk) -> bf16x8 {
+    const char* abase = lds + buf * (A_BYTES + B_BYTES);
+    const int r = wm + mf * 16 + (lane & 15);
+    const int c = kk * 4 + (lane >> 4);
+    return *(const __attribute__((address_space(3))) bf16x8*)(
+        (const __attribute__((address_space(3))) char*)(abase) + r * 128 +
+        ((c ^ (r & 7)) << 4));
+  };
+  auto read_b = [&](int buf, int nf, int kk) -> bf16x8 {
+    const char* bbase = lds + buf * (A_BYTES + B_BYTES) + A_BYTES;
+    const int r = wn + nf * 16 + (lane & 15);
+    const int c = kk * 4 + (lane >> 4);
+    return *(const __attribute__((address_space(3))) bf16x8*)(
+        (const __attribute__((address_space(3))) char*)(bbase) + r * 128 +
+        ((c ^ (r & 7)) << 4));
+  };
+
+  f32x4 acc[4][4] = {};
+  const int KT = K / BK;
+  stage(0, 0);
+  __syncthreads();
+  for (int kt = 0; kt < KT; ++kt) {
+    const int cur = kt & 1;
+    if (kt + 1 < KT) stage(cur ^ 1, kt + 1);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      bf16x8 a[4], b[4];
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf) a[mf] = read_a(cur, mf, kk);
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) b[nf] = read_b(cur, nf, kk);
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < 4; ++nf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[mf], b[nf], acc[mf][nf], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  const int cn = lane & 15;
+  const int r4 = (lane >> 4) * 4;
+#pragma unroll
+  for (int mf = 0; mf < 4; ++mf) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int m = m0 + wm + mf * 16 + r4 + reg;
+      if (m < M) {
+        bf16* crow = C_out + (long long)m * N + n0 + wn + cn;
+#pragma unroll
+        for (int nf = 0; nf < 4; ++nf)
+          crow[nf * 16] = __float2bfloat16(acc[mf][nf][reg]);
+      }
+    }
+  }
+}
+
+extern "C" void launch_pad_nhwc(const void* x, void* xp, int Nimg, int H, int W,
+                                int Hp, int Wp, int C, hipStream_t s) {
+  const long long per_img = (long long)Hp * Wp * (C >> 3);
+  int gx = (int)((per_img + 255) / 256);
+  if (gx > 1024) gx = 1024;
+  hipLaunchKernelGGL(pad_nhwc_kernel, dim3(gx, Nimg), dim3(256), 0, s,
+                     (const bf16*)x, (bf16*)xp, H * W, H, W, Hp, Wp, C);
+}
+
+extern "C" void launch_conv3x3(const void* xp, const void* w3, void* y, int M,
+                               int Cout, int Cin, int HW_out, int W_out, int Hp,
+                               int Wp, int stride, hipStream_t s) {
+  if (Cout % 128 == 0) {
+    constexpr int BM = 128, BN = 128;
+    const int grid = ((M + BM - 1) / BM) * (Cout / BN);
+    const int lds_bytes = 2 * (BM * 64 * 2 + BN * 64 * 2);
+    hipLaunchKernelGGL((conv3x3_kernel<BM, BN, 2, 2>), dim3(grid), dim3(256),
+                       lds_bytes, s, (const bf16*)xp, (const bf16*)w3, (bf16*)y,
+                       M, Cout, Cin, HW_out, W_out, Hp, Wp, stride);
+  } else {
+    constexpr int BM = 256, BN = 64;
+    const int grid = ((M + BM - 1) / BM) * (Cout / BN);
+    const int lds_bytes = 2 * (BM * 64 * 2 + BN * 64 * 2);
+    hipLaunchKernelGGL((conv3x3_kernel<BM, BN, 4, 1>), dim3(grid), dim3(256),
+                       lds_bytes, s, (const bf16*)xp, (const bf16*)w3, (bf16*)y,
+                       M, Cout, Cin, HW_out, W_out, Hp, Wp, stride);
+  }
+}

@@ -31,6 +31,10 @@ extern "C" void launch_bn_bwd_reduce(const void*, const void*, const void*,
 extern "C" void launch_bn_bwd_finalize(const float*, int, float*, int, hipStream_t);
 extern "C" void launch_gemm_bt(const void*, const void*, void*, int, int, int,
                                hipStream_t);
+extern "C" void launch_pad_nhwc(const void*, void*, int, int, int, int, int, int,
+                                hipStream_t);
+extern "C" void launch_conv3x3(const void*, const void*, void*, int, int, int,
+                               int, int, int, int, int, hipStream_t);
 extern "C" void launch_bn_bwd_dx(const void*, const void*, const void*, const float*,
                                  const float*, const float*, const float*, void*,
                                  void*, long long, int, bool, bool, bool,
@@ -196,6 +200,32 @@ torch::Tensor gemm_bt(torch::Tensor a, torch::Tensor b) {
   return c;
 }
 
+torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w3, int64_t stride) {
+  // x: 4-D channels_last bf16 [N, C, H, W]; w3: [Cout, 9*Cin] bf16
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 &&
+                  x.scalar_type() == torch::kBFloat16,
+              "conv3x3: 4-D bf16 GPU tensor required");
+  TORCH_CHECK(x.is_contiguous(torch::MemoryFormat::ChannelsLast),
+              "conv3x3: channels_last required");
+  TORCH_CHECK(stride == 1 || stride == 2, "conv3x3: stride 1 or 2");
+  const int Nimg = (int)x.size(0), Cin = (int)x.size(1);
+  const int H = (int)x.size(2), W = (int)x.size(3);
+  const int Cout = (int)w3.size(0);
+  TORCH_CHECK(w3.size(1) == 9 * Cin && w3.is_contiguous(), "conv3x3: w3 shape");
+  TORCH_CHECK(Cin % 64 == 0 && Cout % 64 == 0, "conv3x3: C % 64");
+  const int Hp = H + 2, Wp = W + 2;
+  const int Hout = (H + 2 - 3) / (int)stride + 1;
+  const int Wout = (W + 2 - 3) / (int)stride + 1;
+  const long long M = (long long)Nimg * Hout * Wout;
+  auto s = cur_stream();
+  auto xp = torch::empty({(long long)Nimg * Hp * Wp * Cin}, x.options());
+  launch_pad_nhwc(x.data_ptr(), xp.data_ptr(), Nimg, H, W, Hp, Wp, Cin, s);
+  auto y = torch::empty({M, Cout}, x.options());
+  launch_conv3x3(xp.data_ptr(), w3.data_ptr(), y.data_ptr(), (int)M, Cout, Cin,
+                 Hout * Wout, Wout, Hp, Wp, (int)stride, s);
+  return y;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -209,5 +239,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_bwd", &bn_bwd,
         "fused BN(+add)+ReLU bwd -> (dx, dgamma, dbeta, dres?)");
   m.def("gemm_bt", &gemm_bt, "bf16 MFMA GEMM: C[M,N] = A[M,K] @ B[N,K]^T");
+  m.def("conv3x3_fwd", &conv3x3_fwd,
+        "implicit-GEMM 3x3 same-pad conv (stride 1/2) -> y2d [M, Cout]");
   m.attr("_arch") = "gfx950";
 }

@@ -42,10 +42,78 @@ class _Conv1x1Hip(torch.autograd.Function):
         return dx, dw
 
 
+_CONV3X3 = os.environ.get("EDL_CONV3X3", "hip")
+
+
+def _repack_w3(weight):
+    """[Cout, Cin, 3, 3] -> s-major [Cout, 9*Cin] bf16 contiguous."""
+    co, ci = weight.shape[0], weight.shape[1]
+    return weight.permute(0, 2, 3, 1).reshape(co, 9 * ci).to(
+        torch.bfloat16).contiguous()
+
+
+class _Conv3x3Hip(torch.autograd.Function):
+    """3x3 same-pad conv on the implicit-GEMM kernel.
+
+    dgrad (stride 1) is ANOTHER 3x3 stride-1 conv with rotated/transposed
+    weights on the same kernel; stride-2 dgrad and all wgrads go through
+    torch.nn.grad (MIOpen) until the TN/transposed kernels land."""
+
+    @staticmethod
+    def forward(ctx, x, weight, stride):
+        ctx.save_for_backward(x, weight)
+        ctx.stride = stride
+        y2d = ext().conv3x3_fwd(x, _repack_w3(weight), stride)
+        n, _, h, w = x.shape
+        ho = (h - 1) // stride + 1
+        wo = (w - 1) // stride + 1
+        co = weight.shape[0]
+        return y2d.view(n, ho, wo, co).permute(0, 3, 1, 2)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight = ctx.saved_tensors
+        stride = ctx.stride
+        dy = dy.contiguous(memory_format=torch.channels_last)
+        if stride == 1:
+            wrot = weight.permute(1, 0, 2, 3).flip(2, 3)
+            dx2d = ext().conv3x3_fwd(
+                dy.to(torch.bfloat16), _repack_w3(wrot), 1)
+            n, _, h, w = x.shape
+            dx = dx2d.view(n, h, w, x.shape[1]).permute(0, 3, 1, 2)
+        else:
+            dx = torch.nn.grad.conv2d_input(
+                list(x.shape), weight, dy, stride=(stride, stride),
+                padding=(1, 1))
+        dw = torch.nn.grad.conv2d_weight(
+            x, list(weight.shape), dy, stride=(stride, stride), padding=(1, 1))
+        return dx, dw, None
+
+
 class Conv2dFast(nn.Conv2d):
     """nn.Conv2d drop-in; 1x1/group-1 convs on CUDA bypass MIOpen."""
 
     def forward(self, x):
+        if (
+            _CONV3X3 == "hip"
+            and x.is_cuda
+            and available()
+            and x.dtype == torch.bfloat16
+            and self.kernel_size == (3, 3)
+            and self.groups == 1
+            and self.padding == (1, 1)
+            and self.stride[0] == self.stride[1]
+            and self.stride[0] in (1, 2)
+            and self.bias is None
+            and self.in_channels % 64 == 0
+            and self.out_channels % 64 == 0
+        ):
+            if not x.is_contiguous(memory_format=torch.channels_last):
+                x = x.contiguous(memory_format=torch.channels_last)
+            w = self.weight
+            if w.dtype != torch.bfloat16:
+                w = w.to(torch.bfloat16)
+            return _Conv3x3Hip.apply(x, w, self.stride[0])
         if (
             _CONV1X1 != "miopen"
             and x.is_cuda

@@ -235,7 +235,14 @@ class TrainerEngine:
             return False
         want = self.graph_capture
         if want is None:
-            want = os.environ.get("EDL_GRAPH_CAPTURE", "1") == "1"
+            env = os.environ.get("EDL_GRAPH_CAPTURE")
+            if env is not None:
+                want = env == "1"
+            else:
+                # default: capture only single-GPU steps — capturing RCCL
+                # collectives inside a graph is riskier (a hang on one rank
+                # stalls the job); opt in with EDL_GRAPH_CAPTURE=1
+                want = self.world_size == 1
         if not want:
             return False
         try:

@@ -53,3 +53,47 @@ def test_conv1x1_hip_matches_matmul():
     assert torch.allclose(x.grad.float(), xr.grad, atol=0.5, rtol=0.05)
     assert torch.allclose(conv.weight.grad.float(), wr.grad, atol=2.0, rtol=0.05), \
         (conv.weight.grad.float() - wr.grad).abs().max().item()
+
+
+@pytest.mark.parametrize("stride", [1, 2])
+@pytest.mark.parametrize("shape", [(2, 64, 16, 16, 64), (2, 128, 14, 14, 256),
+                                   (3, 64, 9, 11, 64)])
+def test_conv3x3_fwd_numerics(stride, shape):
+    import torch.nn.functional as F
+
+    from edl_amd.ops.conv import Conv2dFast
+
+    n, cin, h, w, cout = shape
+    torch.manual_seed(0)
+    conv = Conv2dFast(cin, cout, 3, stride=stride, padding=1, bias=False
+                      ).cuda().to(torch.bfloat16)
+    x = (torch.randn(n, cin, h, w, device="cuda") * 1.5).to(torch.bfloat16)
+    x = x.contiguous(memory_format=torch.channels_last)
+    y = conv(x)
+    ref = F.conv2d(x.float(), conv.weight.float(), stride=stride, padding=1)
+    assert y.shape == ref.shape
+    err = (y.float() - ref).abs()
+    scale = ref.abs().mean().clamp(min=0.5)
+    assert (err / scale).max() < 0.1, (err.max().item(), scale.item())
+
+
+def test_conv3x3_backward_numerics():
+    import torch.nn.functional as F
+
+    from edl_amd.ops.conv import Conv2dFast
+
+    torch.manual_seed(1)
+    conv = Conv2dFast(64, 128, 3, padding=1, bias=False).cuda().to(torch.bfloat16)
+    x = torch.randn(2, 64, 12, 12, device="cuda").to(torch.bfloat16)
+    x = x.contiguous(memory_format=torch.channels_last).requires_grad_(True)
+    y = conv(x)
+    g = torch.randn_like(y).contiguous(memory_format=torch.channels_last)
+    y.backward(g)
+
+    xr = x.detach().float().requires_grad_(True)
+    wr = conv.weight.detach().float().requires_grad_(True)
+    yr = F.conv2d(xr, wr, stride=1, padding=1)
+    yr.backward(g.float())
+    assert torch.allclose(x.grad.float(), xr.grad, atol=0.5, rtol=0.1), \
+        (x.grad.float() - xr.grad).abs().max().item()
+    assert torch.allclose(conv.weight.grad.float(), wr.grad, atol=2.0, rtol=0.1)
