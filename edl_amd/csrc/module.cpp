@@ -35,6 +35,8 @@ extern "C" void launch_pad_nhwc(const void*, void*, int, int, int, int, int, int
                                 hipStream_t);
 extern "C" void launch_conv3x3(const void*, const void*, void*, int, int, int,
                                int, int, int, int, int, hipStream_t);
+extern "C" void launch_transpose_pad(const void*, void*, int, int, int,
+                                     hipStream_t);
 extern "C" void launch_bn_bwd_dx(const void*, const void*, const void*, const float*,
                                  const float*, const float*, const float*, void*,
                                  void*, long long, int, bool, bool, bool,
@@ -200,6 +202,18 @@ torch::Tensor gemm_bt(torch::Tensor a, torch::Tensor b) {
   return c;
 }
 
+torch::Tensor transpose_pad(torch::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 2 &&
+                  x.scalar_type() == torch::kBFloat16 && x.is_contiguous(),
+              "transpose_pad: contiguous 2-D bf16 GPU tensor");
+  const int M = (int)x.size(0), C = (int)x.size(1);
+  TORCH_CHECK(C % 8 == 0, "transpose_pad: C % 8");
+  const int Mp = (M + 63) / 64 * 64;
+  auto y = torch::empty({C, Mp}, x.options());
+  launch_transpose_pad(x.data_ptr(), y.data_ptr(), M, C, Mp, cur_stream());
+  return y;
+}
+
 torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w3, int64_t stride) {
   // x: 4-D channels_last bf16 [N, C, H, W]; w3: [Cout, 9*Cin] bf16
   TORCH_CHECK(x.is_cuda() && x.dim() == 4 &&
@@ -241,5 +255,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_bt", &gemm_bt, "bf16 MFMA GEMM: C[M,N] = A[M,K] @ B[N,K]^T");
   m.def("conv3x3_fwd", &conv3x3_fwd,
         "implicit-GEMM 3x3 same-pad conv (stride 1/2) -> y2d [M, Cout]");
+  m.def("transpose_pad", &transpose_pad,
+        "bf16 [M,C] -> [C, ceil64(M)] transpose with zero pad");
   m.attr("_arch") = "gfx950";
 }

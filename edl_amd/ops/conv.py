@@ -37,8 +37,12 @@ class _Conv1x1Hip(torch.autograd.Function):
     def backward(ctx, dy2d):
         x2d, w = ctx.saved_tensors
         dy2d = dy2d.contiguous()
-        dx = ext().gemm_bt(dy2d, w.t().contiguous())
-        dw = dy2d.t() @ x2d  # TN: [Cout, M] @ [M, Cin]
+        e = ext()
+        dx = e.gemm_bt(dy2d, w.t().contiguous())
+        # wgrad (TN, reduction over huge M): transpose-pad both operands
+        # and reuse the bt kernel — hipBLASLt's TN heuristics measured
+        # 272 us on these shapes (profiles/)
+        dw = e.gemm_bt(e.transpose_pad(dy2d), e.transpose_pad(x2d))
         return dx, dw
 
 

@@ -97,3 +97,16 @@ def test_conv3x3_backward_numerics():
     assert torch.allclose(x.grad.float(), xr.grad, atol=0.5, rtol=0.1), \
         (x.grad.float() - xr.grad).abs().max().item()
     assert torch.allclose(conv.weight.grad.float(), wr.grad, atol=2.0, rtol=0.1)
+
+
+@pytest.mark.parametrize("M,C", [(256, 64), (1000, 128), (100, 64), (1568, 512)])
+def test_transpose_pad(M, C):
+    from edl_amd import ops
+
+    torch.manual_seed(2)
+    x = (torch.randn(M, C, device="cuda")).to(torch.bfloat16)
+    y = ops.ext().transpose_pad(x)
+    Mp = (M + 63) // 64 * 64
+    assert y.shape == (C, Mp)
+    assert torch.equal(y[:, :M], x.t().contiguous())
+    assert (y[:, M:] == 0).all()
