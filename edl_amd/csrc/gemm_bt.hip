@@ -27,10 +27,15 @@ typedef float f32x4 __attribute__((ext_vector_type(4)));
 
 extern __shared__ __attribute__((aligned(16))) char smem[];
 
-template <int BM, int BN, int WAVES_M, int WAVES_N>
+template <int BM, int BN, int WAVES_M, int WAVES_N, bool SPLITK = false>
 __global__ __launch_bounds__(256, 2) void gemm_bt_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B,
-    bf16* __restrict__ C, const int M, const int N, const int K) {
+    void* __restrict__ C_any, const int M, const int N, const int K) {
+  // SPLITK: grid.y k-slices; each block atomically folds its fp32 partial
+  // tile into C (fp32, pre-zeroed). Wgrad-shaped GEMMs (tiny [Cout, Cin]
+  // output, K = M ~ 1e5) otherwise serialize on 1-2 blocks.
+  bf16* C = SPLITK ? nullptr : (bf16*)C_any;
+  float* Cf = SPLITK ? (float*)C_any : nullptr;
   constexpr int BK = 64;
   constexpr int A_BYTES = BM * BK * 2;  // row stride 128 B
   constexpr int B_BYTES = BN * BK * 2;
@@ -103,13 +108,19 @@ __global__ __launch_bounds__(256, 2) void gemm_bt_kernel(
         ((c ^ (r & 7)) << 4));
   };
 
-  const int KT = K / BK;
-  stage(0, 0);
+  int kt_begin = 0, kt_end = K / BK;
+  if (SPLITK) {
+    const int per = (kt_end + gridDim.y - 1) / gridDim.y;
+    kt_begin = blockIdx.y * per;
+    kt_end = min(kt_end, kt_begin + per);
+    if (kt_begin >= kt_end) return;
+  }
+  stage(0, kt_begin);
   __syncthreads();
 
-  for (int kt = 0; kt < KT; ++kt) {
-    const int cur = kt & 1;
-    if (kt + 1 < KT) stage(cur ^ 1, kt + 1);
+  for (int kt = kt_begin; kt < kt_end; ++kt) {
+    const int cur = (kt - kt_begin) & 1;
+    if (kt + 1 < kt_end) stage(cur ^ 1, kt + 1);
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
       bf16x8 a[4], b[4];
@@ -137,10 +148,18 @@ __global__ __launch_bounds__(256, 2) void gemm_bt_kernel(
     for (int reg = 0; reg < 4; ++reg) {
       const int m = m0 + wm + mf * 16 + r4 + reg;
       if (m < M) {
-        bf16* crow = C + (long long)m * N + n0 + wn + cn;
+        if (SPLITK) {
+          float* crow = Cf + (long long)m * N + n0 + wn + cn;
 #pragma unroll
-        for (int nf = 0; nf < 4; ++nf) {
-          crow[nf * 16] = __float2bfloat16(acc[mf][nf][reg]);
+          for (int nf = 0; nf < 4; ++nf) {
+            atomicAdd(&crow[nf * 16], acc[mf][nf][reg]);
+          }
+        } else {
+          bf16* crow = C + (long long)m * N + n0 + wn + cn;
+#pragma unroll
+          for (int nf = 0; nf < 4; ++nf) {
+            crow[nf * 16] = __float2bfloat16(acc[mf][nf][reg]);
+          }
         }
       }
     }
@@ -154,14 +173,34 @@ extern "C" void launch_gemm_bt(const void* A, const void* B, void* C, int M,
     const int grid = ((M + BM - 1) / BM) * (N / BN);
     const int lds_bytes = 2 * (BM * 64 * 2 + BN * 64 * 2);
     hipLaunchKernelGGL((gemm_bt_kernel<BM, BN, 2, 2>), dim3(grid), dim3(256),
-                       lds_bytes, s, (const bf16*)A, (const bf16*)B, (bf16*)C,
-                       M, N, K);
+                       lds_bytes, s, (const bf16*)A, (const bf16*)B, C, M, N, K);
   } else {  // N % 64 == 0
     constexpr int BM = 256, BN = 64;
     const int grid = ((M + BM - 1) / BM) * (N / BN);
     const int lds_bytes = 2 * (BM * 64 * 2 + BN * 64 * 2);
     hipLaunchKernelGGL((gemm_bt_kernel<BM, BN, 4, 1>), dim3(grid), dim3(256),
-                       lds_bytes, s, (const bf16*)A, (const bf16*)B, (bf16*)C,
+                       lds_bytes, s, (const bf16*)A, (const bf16*)B, C, M, N, K);
+  }
+}
+
+extern "C" void launch_gemm_bt_splitk(const void* A, const void* B, float* Cf32,
+                                      int M, int N, int K, int splitk,
+                                      hipStream_t s) {
+  // wgrad shapes: small M (= Cout) and N (= Cin); 128x64 tiles via the
+  // 256x64 path would waste rows — use 128x128 when possible, else 256x64.
+  if (M >= 128 && N % 128 == 0) {
+    constexpr int BM = 128, BN = 128;
+    const dim3 grid(((M + BM - 1) / BM) * (N / BN), splitk);
+    const int lds_bytes = 2 * (BM * 64 * 2 + BN * 64 * 2);
+    hipLaunchKernelGGL((gemm_bt_kernel<BM, BN, 2, 2, true>), grid, dim3(256),
+                       lds_bytes, s, (const bf16*)A, (const bf16*)B, Cf32,
+                       M, N, K);
+  } else {
+    constexpr int BM = 256, BN = 64;
+    const dim3 grid(((M + BM - 1) / BM) * (N / BN), splitk);
+    const int lds_bytes = 2 * (BM * 64 * 2 + BN * 64 * 2);
+    hipLaunchKernelGGL((gemm_bt_kernel<BM, BN, 4, 1, true>), grid, dim3(256),
+                       lds_bytes, s, (const bf16*)A, (const bf16*)B, Cf32,
                        M, N, K);
   }
 }

@@ -37,6 +37,8 @@ extern "C" void launch_conv3x3(const void*, const void*, void*, int, int, int,
                                int, int, int, int, int, hipStream_t);
 extern "C" void launch_transpose_pad(const void*, void*, int, int, int,
                                      hipStream_t);
+extern "C" void launch_gemm_bt_splitk(const void*, const void*, float*, int, int,
+                                      int, int, hipStream_t);
 extern "C" void launch_bn_bwd_dx(const void*, const void*, const void*, const float*,
                                  const float*, const float*, const float*, void*,
                                  void*, long long, int, bool, bool, bool,
@@ -202,6 +204,30 @@ torch::Tensor gemm_bt(torch::Tensor a, torch::Tensor b) {
   return c;
 }
 
+torch::Tensor gemm_bt_splitk(torch::Tensor a, torch::Tensor b, int64_t splitk) {
+  // fp32 C[M,N] = A[M,K] @ B[N,K]^T with grid.y k-slices (atomic combine).
+  TORCH_CHECK(a.is_cuda() && b.is_cuda() && a.dim() == 2 && b.dim() == 2 &&
+                  a.size(1) == b.size(1),
+              "gemm_bt_splitk: [M,K] x [N,K] GPU");
+  TORCH_CHECK(a.scalar_type() == torch::kBFloat16 &&
+                  b.scalar_type() == torch::kBFloat16,
+              "gemm_bt_splitk: bf16 only");
+  auto ac = a.contiguous();
+  auto bc = b.contiguous();
+  const int M = (int)a.size(0), K = (int)a.size(1), N = (int)b.size(0);
+  TORCH_CHECK(K % 64 == 0 && N % 64 == 0, "gemm_bt_splitk: K,N % 64");
+  if (splitk <= 0) {
+    // pick splitk so total blocks ~ 2x CUs, bounded by k-tiles
+    const int tiles = ((M + 127) / 128) * ((N + 63) / 64);
+    splitk = std::max<int64_t>(1, 512 / std::max(1, tiles));
+    splitk = std::min<int64_t>(splitk, K / 64);
+  }
+  auto c = torch::zeros({M, N}, a.options().dtype(torch::kFloat32));
+  launch_gemm_bt_splitk(ac.data_ptr(), bc.data_ptr(), c.data_ptr<float>(), M, N,
+                        K, (int)splitk, cur_stream());
+  return c;
+}
+
 torch::Tensor transpose_pad(torch::Tensor x) {
   TORCH_CHECK(x.is_cuda() && x.dim() == 2 &&
                   x.scalar_type() == torch::kBFloat16 && x.is_contiguous(),
@@ -257,5 +283,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "implicit-GEMM 3x3 same-pad conv (stride 1/2) -> y2d [M, Cout]");
   m.def("transpose_pad", &transpose_pad,
         "bf16 [M,C] -> [C, ceil64(M)] transpose with zero pad");
+  m.def("gemm_bt_splitk", &gemm_bt_splitk,
+        "split-K bt GEMM -> fp32 C (wgrad shapes)", pybind11::arg("a"),
+        pybind11::arg("b"), pybind11::arg("splitk") = 0);
   m.attr("_arch") = "gfx950";
 }

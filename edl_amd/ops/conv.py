@@ -39,11 +39,12 @@ class _Conv1x1Hip(torch.autograd.Function):
         dy2d = dy2d.contiguous()
         e = ext()
         dx = e.gemm_bt(dy2d, w.t().contiguous())
-        # wgrad (TN, reduction over huge M) stays on hipBLASLt: gemm_bt has
-        # no split-K yet, and a [Cout, Cin] output gives it only 1-2 blocks
-        # against K=M~100k (measured 24 vs 15 ms/step end-to-end)
-        dw = dy2d.t() @ x2d
-        return dx, dw
+        # wgrad: TN with reduction over huge M -> transpose-pad both
+        # operands and run the split-K bt kernel (fp32 combine); hipBLASLt's
+        # TN picks measured 272 us on these shapes, a non-split bt kernel
+        # serialized on 1-2 blocks (24 ms/step end-to-end)
+        dw = e.gemm_bt_splitk(e.transpose_pad(dy2d), e.transpose_pad(x2d), 0)
+        return dx, dw.to(dy2d.dtype)
 
 
 _CONV3X3 = os.environ.get("EDL_CONV3X3", "hip")

@@ -110,3 +110,21 @@ def test_transpose_pad(M, C):
     assert y.shape == (C, Mp)
     assert torch.equal(y[:, :M], x.t().contiguous())
     assert (y[:, M:] == 0).all()
+
+
+@pytest.mark.parametrize("M,N,K,splitk", [
+    (64, 64, 6400, 0), (128, 256, 4096, 8), (256, 64, 1280, 4),
+    (512, 128, 64, 0),
+])
+def test_gemm_bt_splitk_numerics(M, N, K, splitk):
+    from edl_amd import ops
+
+    torch.manual_seed(3)
+    a = (torch.randn(M, K, device="cuda") * 0.5).to(torch.bfloat16)
+    b = (torch.randn(N, K, device="cuda") * 0.5).to(torch.bfloat16)
+    c = ops.ext().gemm_bt_splitk(a, b, splitk)
+    assert c.dtype == torch.float32
+    ref = a.float() @ b.float().t()
+    err = (c - ref).abs()
+    scale = ref.abs().mean().clamp(min=1)
+    assert (err / scale).max() < 0.05, (err.max().item(), scale.item())
