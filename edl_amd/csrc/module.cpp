@@ -39,6 +39,8 @@ extern "C" void launch_transpose_pad(const void*, void*, int, int, int,
                                      hipStream_t);
 extern "C" void launch_gemm_bt_splitk(const void*, const void*, float*, int, int,
                                       int, int, hipStream_t);
+extern "C" void launch_shift9_transpose(const void*, void*, int, int, int, int,
+                                        int, int, int, int, hipStream_t);
 extern "C" void launch_bn_bwd_dx(const void*, const void*, const void*, const float*,
                                  const float*, const float*, const float*, void*,
                                  void*, long long, int, bool, bool, bool,
@@ -204,6 +206,29 @@ torch::Tensor gemm_bt(torch::Tensor a, torch::Tensor b) {
   return c;
 }
 
+torch::Tensor conv3x3_wgrad_operand(torch::Tensor x, int64_t stride) {
+  // x: 4-D channels_last bf16 input of the conv -> [9*Cin, Mp] shifted
+  // transpose (pads internally).
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 &&
+                  x.scalar_type() == torch::kBFloat16 &&
+                  x.is_contiguous(torch::MemoryFormat::ChannelsLast),
+              "wgrad_operand: 4-D channels_last bf16");
+  const int Nimg = (int)x.size(0), Cin = (int)x.size(1);
+  const int H = (int)x.size(2), W = (int)x.size(3);
+  const int Hp = H + 2, Wp = W + 2;
+  const int Hout = (H - 1) / (int)stride + 1;
+  const int Wout = (W - 1) / (int)stride + 1;
+  const int M = Nimg * Hout * Wout;
+  const int Mp = (M + 63) / 64 * 64;
+  auto s = cur_stream();
+  auto xp = torch::empty({(long long)Nimg * Hp * Wp * Cin}, x.options());
+  launch_pad_nhwc(x.data_ptr(), xp.data_ptr(), Nimg, H, W, Hp, Wp, Cin, s);
+  auto out = torch::empty({(long long)9 * Cin, Mp}, x.options());
+  launch_shift9_transpose(xp.data_ptr(), out.data_ptr(), M, Cin, Mp,
+                          Hout * Wout, Wout, Hp, Wp, (int)stride, s);
+  return out;
+}
+
 torch::Tensor gemm_bt_splitk(torch::Tensor a, torch::Tensor b, int64_t splitk) {
   // fp32 C[M,N] = A[M,K] @ B[N,K]^T with grid.y k-slices (atomic combine).
   TORCH_CHECK(a.is_cuda() && b.is_cuda() && a.dim() == 2 && b.dim() == 2 &&
@@ -286,5 +311,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_bt_splitk", &gemm_bt_splitk,
         "split-K bt GEMM -> fp32 C (wgrad shapes)", pybind11::arg("a"),
         pybind11::arg("b"), pybind11::arg("splitk") = 0);
+  m.def("conv3x3_wgrad_operand", &conv3x3_wgrad_operand,
+        "padded shifted transpose of conv3x3 input -> [9*Cin, Mp]");
   m.attr("_arch") = "gfx950";
 }

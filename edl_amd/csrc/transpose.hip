@@ -52,3 +52,54 @@ extern "C" void launch_transpose_pad(const void* in, void* out, int M, int C,
   hipLaunchKernelGGL(transpose_pad_kernel, grid, dim3(256), 0, s,
                      (const bf16*)in, (bf16*)out, M, C, Mp);
 }
+
+// Shifted 9-way transpose for 3x3 wgrad:
+//   out[(s*C + c)][m] = xp[prow(m) + shift(s)][c],  out[:, M..Mp) = 0
+// where xp is the zero-halo padded NHWC image (pad_nhwc) and prow(m) is
+// the conv3x3 output-position -> padded-row map (stride folded in). The
+// result is the B operand of dW3 = gemm_bt_splitk(dyT, out).
+extern "C" __global__ void shift9_transpose_kernel(
+    const bf16* __restrict__ xp, bf16* __restrict__ out, const int M,
+    const int C, const int Mp, const int HW_out, const int W_out, const int Hp,
+    const int Wp, const int stride_hw) {
+  __shared__ bf16 tile[64][64 + 4];
+  const int s = blockIdx.z;  // 0..8
+  const int shift = ((s / 3) * Wp + (s % 3)) * C;
+  const int tm0 = blockIdx.x * 64;
+  const int tc0 = blockIdx.y * 64;
+  const int lr = threadIdx.x >> 3;
+  const int lc = (threadIdx.x & 7) * 8;
+  for (int half = 0; half < 2; ++half) {
+    const int m = tm0 + half * 32 + lr;
+    uint4 v = {0, 0, 0, 0};
+    if (m < M && tc0 + lc < C) {
+      const int n_img = m / HW_out;
+      const int rem = m % HW_out;
+      const int h = rem / W_out, w = rem % W_out;
+      const long long prow =
+          ((long long)n_img * Hp + h * stride_hw) * Wp + w * stride_hw;
+      v = *reinterpret_cast<const uint4*>(xp + prow * C + shift + tc0 + lc);
+    }
+    *reinterpret_cast<uint4*>(&tile[half * 32 + lr][lc]) = v;
+  }
+  __syncthreads();
+  for (int half = 0; half < 2; ++half) {
+    const int c = tc0 + half * 32 + lr;
+    if (c < C) {
+      bf16 v[8];
+#pragma unroll
+      for (int i = 0; i < 8; ++i) v[i] = tile[lc + i][half * 32 + lr];
+      bf16* dst = out + ((long long)s * C + c) * Mp + tm0 + lc;
+      *reinterpret_cast<uint4*>(dst) = *reinterpret_cast<const uint4*>(v);
+    }
+  }
+}
+
+extern "C" void launch_shift9_transpose(const void* xp, void* out, int M, int C,
+                                        int Mp, int HW_out, int W_out, int Hp,
+                                        int Wp, int stride, hipStream_t s) {
+  dim3 grid((Mp + 63) / 64, (C + 63) / 64, 9);
+  hipLaunchKernelGGL(shift9_transpose_kernel, grid, dim3(256), 0, s,
+                     (const bf16*)xp, (bf16*)out, M, C, Mp, HW_out, W_out, Hp,
+                     Wp, stride);
+}

@@ -90,8 +90,16 @@ class _Conv3x3Hip(torch.autograd.Function):
             dx = torch.nn.grad.conv2d_input(
                 list(x.shape), weight, dy, stride=(stride, stride),
                 padding=(1, 1))
-        dw = torch.nn.grad.conv2d_weight(
-            x, list(weight.shape), dy, stride=(stride, stride), padding=(1, 1))
+        # wgrad: dW3[Cout, 9Cin] = dyT @ shift9(x)^T on the split-K kernel
+        e = ext()
+        n = dy.shape[0]
+        co = weight.shape[0]
+        ci = weight.shape[1]
+        dy2d = dy.permute(0, 2, 3, 1).reshape(-1, co)
+        dw3 = e.gemm_bt_splitk(
+            e.transpose_pad(dy2d.to(torch.bfloat16)),
+            e.conv3x3_wgrad_operand(x, stride), 0)
+        dw = dw3.view(co, 3, 3, ci).permute(0, 3, 1, 2).to(dy.dtype)
         return dx, dw, None
 
 

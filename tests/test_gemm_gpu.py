@@ -77,13 +77,15 @@ def test_conv3x3_fwd_numerics(stride, shape):
     assert (err / scale).max() < 0.1, (err.max().item(), scale.item())
 
 
-def test_conv3x3_backward_numerics():
+@pytest.mark.parametrize("stride", [1, 2])
+def test_conv3x3_backward_numerics(stride):
     import torch.nn.functional as F
 
     from edl_amd.ops.conv import Conv2dFast
 
     torch.manual_seed(1)
-    conv = Conv2dFast(64, 128, 3, padding=1, bias=False).cuda().to(torch.bfloat16)
+    conv = Conv2dFast(64, 128, 3, stride=stride, padding=1, bias=False
+                      ).cuda().to(torch.bfloat16)
     x = torch.randn(2, 64, 12, 12, device="cuda").to(torch.bfloat16)
     x = x.contiguous(memory_format=torch.channels_last).requires_grad_(True)
     y = conv(x)
@@ -92,11 +94,12 @@ def test_conv3x3_backward_numerics():
 
     xr = x.detach().float().requires_grad_(True)
     wr = conv.weight.detach().float().requires_grad_(True)
-    yr = F.conv2d(xr, wr, stride=1, padding=1)
+    yr = F.conv2d(xr, wr, stride=stride, padding=1)
     yr.backward(g.float())
     assert torch.allclose(x.grad.float(), xr.grad, atol=0.5, rtol=0.1), \
         (x.grad.float() - xr.grad).abs().max().item()
-    assert torch.allclose(conv.weight.grad.float(), wr.grad, atol=2.0, rtol=0.1)
+    assert torch.allclose(conv.weight.grad.float(), wr.grad, atol=2.0, rtol=0.1), \
+        (conv.weight.grad.float() - wr.grad).abs().max().item()
 
 
 @pytest.mark.parametrize("M,C", [(256, 64), (1000, 128), (100, 64), (1568, 512)])
