@@ -294,8 +294,10 @@ extern "C" int bn_stats_grid(long long M, int C) {
   const int c8 = C >> 3;
   long long rows_per_block = 256 / c8 > 0 ? 256 / c8 : 1;
   long long want = (M + rows_per_block - 1) / rows_per_block;
-  long long cap = 131072 / (2 * (long long)C);  // 512 KiB of partials
-  if (cap > 512) cap = 512;
+  // cap bounds the finalize kernels' serial partial-read loop (they were
+  // latency-bound at ~12-14 us with 512-block partials; ~5 us at 192)
+  long long cap = 131072 / (2 * (long long)C);
+  if (cap > 192) cap = 192;
   if (cap < 8) cap = 8;
   long long g = want < cap ? want : cap;
   return (int)(g > 0 ? g : 1);

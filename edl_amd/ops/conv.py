@@ -25,29 +25,41 @@ _CONV1X1 = os.environ.get("EDL_CONV1X1", "hip")
 
 class _Conv1x1Hip(torch.autograd.Function):
     """y2d = x2d @ W^T via the gemm_bt MFMA kernel; dgrad reuses the same
-    kernel on a transposed-weight copy; wgrad (TN shape) goes through
-    torch.matmul until the TN kernel lands."""
+    kernel on a (per-step cached) transposed-weight copy; wgrad runs the
+    split-K bt kernel on transpose-padded operands."""
 
     @staticmethod
-    def forward(ctx, x2d, w):  # w: [Cout, Cin] bf16
-        ctx.save_for_backward(x2d, w)
+    def forward(ctx, x2d, w, wt_cached):  # w: [Cout, Cin] bf16 (tracked)
+        ctx.save_for_backward(x2d)
+        ctx.wt = wt_cached  # [Cin, Cout] bf16, derived per weight epoch
         return ext().gemm_bt(x2d, w)
 
     @staticmethod
     def backward(ctx, dy2d):
-        x2d, w = ctx.saved_tensors
+        (x2d,) = ctx.saved_tensors
         dy2d = dy2d.contiguous()
         e = ext()
-        dx = e.gemm_bt(dy2d, w.t().contiguous())
+        dx = e.gemm_bt(dy2d, ctx.wt)
         # wgrad: TN with reduction over huge M -> transpose-pad both
         # operands and run the split-K bt kernel (fp32 combine); hipBLASLt's
         # TN picks measured 272 us on these shapes, a non-split bt kernel
         # serialized on 1-2 blocks (24 ms/step end-to-end)
         dw = e.gemm_bt_splitk(e.transpose_pad(dy2d), e.transpose_pad(x2d), 0)
-        return dx, dw.to(dy2d.dtype)
+        return dx, dw.to(dy2d.dtype), None
 
 
 _CONV3X3 = os.environ.get("EDL_CONV3X3", "hip")
+
+# Weight-derived tensors (bf16 casts, transposed/repacked layouts) are
+# immutable within one optimizer step; FusedSGD.step() bumps this epoch and
+# Conv2dFast caches per-epoch, removing ~100 small cast/copy kernels per
+# step (profiles/r01_step4).
+_weight_epoch = 0
+
+
+def bump_weight_epoch():
+    global _weight_epoch
+    _weight_epoch += 1
 
 
 def _repack_w3(weight):
@@ -65,10 +77,11 @@ class _Conv3x3Hip(torch.autograd.Function):
     torch.nn.grad (MIOpen) until the TN/transposed kernels land."""
 
     @staticmethod
-    def forward(ctx, x, weight, stride):
+    def forward(ctx, x, weight, w3_cached, w3rot_cached, stride):
         ctx.save_for_backward(x, weight)
         ctx.stride = stride
-        y2d = ext().conv3x3_fwd(x, _repack_w3(weight), stride)
+        ctx.w3rot = w3rot_cached
+        y2d = ext().conv3x3_fwd(x, w3_cached, stride)
         n, _, h, w = x.shape
         ho = (h - 1) // stride + 1
         wo = (w - 1) // stride + 1
@@ -81,9 +94,7 @@ class _Conv3x3Hip(torch.autograd.Function):
         stride = ctx.stride
         dy = dy.contiguous(memory_format=torch.channels_last)
         if stride == 1:
-            wrot = weight.permute(1, 0, 2, 3).flip(2, 3)
-            dx2d = ext().conv3x3_fwd(
-                dy.to(torch.bfloat16), _repack_w3(wrot), 1)
+            dx2d = ext().conv3x3_fwd(dy.to(torch.bfloat16), ctx.w3rot, 1)
             n, _, h, w = x.shape
             dx = dx2d.view(n, h, w, x.shape[1]).permute(0, 3, 1, 2)
         else:
@@ -100,11 +111,21 @@ class _Conv3x3Hip(torch.autograd.Function):
             e.transpose_pad(dy2d.to(torch.bfloat16)),
             e.conv3x3_wgrad_operand(x, stride), 0)
         dw = dw3.view(co, 3, 3, ci).permute(0, 3, 1, 2).to(dy.dtype)
-        return dx, dw, None
+        return dx, dw, None, None, None
 
 
 class Conv2dFast(nn.Conv2d):
     """nn.Conv2d drop-in; 1x1/group-1 convs on CUDA bypass MIOpen."""
+
+    def _cached(self, key, fn):
+        cache = getattr(self, "_w_cache", None)
+        if cache is None or cache[0] != _weight_epoch:
+            cache = (_weight_epoch, {})
+            self._w_cache = cache
+        d = cache[1]
+        if key not in d:
+            d[key] = fn()
+        return d[key]
 
     def forward(self, x):
         if (
@@ -126,7 +147,12 @@ class Conv2dFast(nn.Conv2d):
             w = self.weight
             if w.dtype != torch.bfloat16:
                 w = w.to(torch.bfloat16)
-            return _Conv3x3Hip.apply(x, w, self.stride[0])
+            w3 = self._cached("w3", lambda: _repack_w3(self.weight.detach()))
+            w3rot = None
+            if self.stride[0] == 1:
+                w3rot = self._cached("w3rot", lambda: _repack_w3(
+                    self.weight.detach().permute(1, 0, 2, 3).flip(2, 3)))
+            return _Conv3x3Hip.apply(x, w, w3, w3rot, self.stride[0])
         if (
             _CONV1X1 != "miopen"
             and x.is_cuda
@@ -153,7 +179,9 @@ class Conv2dFast(nn.Conv2d):
             ):
                 if wt.dtype != torch.bfloat16:
                     wt = wt.to(torch.bfloat16)
-                y2d = _Conv1x1Hip.apply(x2d, wt)
+                wt_t = self._cached("wt_t", lambda: self.weight.detach().view(
+                    self.out_channels, c).to(torch.bfloat16).t().contiguous())
+                y2d = _Conv1x1Hip.apply(x2d, wt, wt_t)
             else:
                 y2d = x2d @ wt.t()
             return (

@@ -57,6 +57,9 @@ class FusedSGD:
 
     @torch.no_grad()
     def step(self):
+        from .conv import bump_weight_epoch
+
+        bump_weight_epoch()  # invalidate per-step weight-derived caches
         g0 = self.param_groups[0]
         lr, mu, wd = g0["lr"], g0["momentum"], g0["weight_decay"]
         for bk in self._materialize():

@@ -127,6 +127,9 @@ class TrainerEngine:
             self._resume()
         # everyone starts from rank-0's weights (fresh or resumed)
         self.reducer.broadcast_params(src=0)
+        from ..ops.conv import bump_weight_epoch
+
+        bump_weight_epoch()  # weights changed outside an optimizer step
         edist.barrier(self.device)
         return self
 
