@@ -24,6 +24,16 @@ static inline int elementwise_grid(long long n, int block) {
   return (int)(want < cap ? want : cap);
 }
 
+// XCD-aware bijective blockIdx remap (guide T1): the dispatcher places
+// block b on XCD b%8; remapping gives each XCD a CONTIGUOUS tile range so
+// neighbouring tiles (sharing operand panels) hit the same private L2.
+__device__ __forceinline__ int xcd_swizzle(int orig, int nwg) {
+  const int nx = 8;
+  const int q = nwg / nx, r = nwg % nx;
+  const int xcd = orig % nx, local = orig / nx;
+  return (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + local;
+}
+
 __device__ __forceinline__ float wave_reduce_sum(float v) {
 #pragma unroll
   for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, 64);

@@ -67,8 +67,9 @@ __global__ __launch_bounds__(256, 2) void conv3x3_kernel(
   const int wave = threadIdx.x >> 6;
 
   const int tiles_n = N / BN;
-  const int m0 = (blockIdx.x / tiles_n) * BM;
-  const int n0 = (blockIdx.x % tiles_n) * BN;
+  const int bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  const int m0 = (bid / tiles_n) * BM;
+  const int n0 = (bid % tiles_n) * BN;
   const int wm = (wave / WAVES_N) * 64;
   const int wn = (wave % WAVES_N) * 64;
 

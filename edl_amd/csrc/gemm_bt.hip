@@ -44,10 +44,11 @@ __global__ __launch_bounds__(256, 2) void gemm_bt_kernel(
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
 
-  // tile coordinates (grid.x = tiles_m * tiles_n, n fastest)
+  // tile coordinates (grid.x = tiles_m * tiles_n, n fastest; XCD-swizzled)
   const int tiles_n = N / BN;
-  const int tile_m = blockIdx.x / tiles_n;
-  const int tile_n = blockIdx.x % tiles_n;
+  const int bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  const int tile_m = bid / tiles_n;
+  const int tile_n = bid % tiles_n;
   const int m0 = tile_m * BM;
   const int n0 = tile_n * BN;
 
