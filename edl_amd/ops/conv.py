@@ -29,10 +29,15 @@ class _Conv1x1Hip(torch.autograd.Function):
     split-K bt kernel on transpose-padded operands."""
 
     @staticmethod
-    def forward(ctx, x2d, w, wt_cached):  # w: [Cout, Cin] bf16 (tracked)
+    def forward(ctx, x2d, w_param, w_bf16, wt_cached):
+        # w_param: the fp32 (or bf16) parameter view — the DIFFERENTIABLE
+        # input; w_bf16/wt_cached: per-step cached compute copies. Grads
+        # come back in fp32 straight from the split-K kernel: no cast
+        # nodes on the weight path at all.
         ctx.save_for_backward(x2d)
         ctx.wt = wt_cached  # [Cin, Cout] bf16, derived per weight epoch
-        return ext().gemm_bt(x2d, w)
+        ctx.w_dtype = w_param.dtype
+        return ext().gemm_bt(x2d, w_bf16)
 
     @staticmethod
     def backward(ctx, dy2d):
@@ -45,7 +50,9 @@ class _Conv1x1Hip(torch.autograd.Function):
         # TN picks measured 272 us on these shapes, a non-split bt kernel
         # serialized on 1-2 blocks (24 ms/step end-to-end)
         dw = e.gemm_bt_splitk(e.transpose_pad(dy2d), e.transpose_pad(x2d), 0)
-        return dx, dw.to(dy2d.dtype), None
+        if dw.dtype != ctx.w_dtype:
+            dw = dw.to(ctx.w_dtype)
+        return dx, dw, None, None
 
 
 _CONV3X3 = os.environ.get("EDL_CONV3X3", "hip")
@@ -77,15 +84,16 @@ class _Conv3x3Hip(torch.autograd.Function):
     torch.nn.grad (MIOpen) until the TN/transposed kernels land."""
 
     @staticmethod
-    def forward(ctx, x, weight, w3_cached, w3rot_cached, stride):
-        ctx.save_for_backward(x, weight)
+    def forward(ctx, x, w_param, w_bf16, w3_cached, w3rot_cached, stride):
+        ctx.save_for_backward(x, w_bf16)
         ctx.stride = stride
         ctx.w3rot = w3rot_cached
+        ctx.w_dtype = w_param.dtype
         y2d = ext().conv3x3_fwd(x, w3_cached, stride)
         n, _, h, w = x.shape
         ho = (h - 1) // stride + 1
         wo = (w - 1) // stride + 1
-        co = weight.shape[0]
+        co = w_param.shape[0]
         return y2d.view(n, ho, wo, co).permute(0, 3, 1, 2)
 
     @staticmethod
@@ -110,8 +118,10 @@ class _Conv3x3Hip(torch.autograd.Function):
         dw3 = e.gemm_bt_splitk(
             e.transpose_pad(dy2d.to(torch.bfloat16)),
             e.conv3x3_wgrad_operand(x, stride), 0)
-        dw = dw3.view(co, 3, 3, ci).permute(0, 3, 1, 2).to(dy.dtype)
-        return dx, dw, None, None, None
+        dw = dw3.view(co, 3, 3, ci).permute(0, 3, 1, 2)  # fp32
+        if dw.dtype != ctx.w_dtype:
+            dw = dw.to(ctx.w_dtype)
+        return dx, dw, None, None, None, None
 
 
 class Conv2dFast(nn.Conv2d):
@@ -144,15 +154,15 @@ class Conv2dFast(nn.Conv2d):
         ):
             if not x.is_contiguous(memory_format=torch.channels_last):
                 x = x.contiguous(memory_format=torch.channels_last)
-            w = self.weight
-            if w.dtype != torch.bfloat16:
-                w = w.to(torch.bfloat16)
+            w_bf16 = self._cached("w_bf16", lambda: self.weight.detach().to(
+                torch.bfloat16).contiguous())
             w3 = self._cached("w3", lambda: _repack_w3(self.weight.detach()))
             w3rot = None
             if self.stride[0] == 1:
                 w3rot = self._cached("w3rot", lambda: _repack_w3(
                     self.weight.detach().permute(1, 0, 2, 3).flip(2, 3)))
-            return _Conv3x3Hip.apply(x, w, w3, w3rot, self.stride[0])
+            return _Conv3x3Hip.apply(x, self.weight, w_bf16, w3, w3rot,
+                                     self.stride[0])
         if (
             _CONV1X1 != "miopen"
             and x.is_cuda
@@ -177,11 +187,12 @@ class Conv2dFast(nn.Conv2d):
                 and c % 64 == 0
                 and self.out_channels % 64 == 0
             ):
-                if wt.dtype != torch.bfloat16:
-                    wt = wt.to(torch.bfloat16)
+                w_bf16 = self._cached("w_bf16", lambda: self.weight.detach()
+                                      .view(self.out_channels, c)
+                                      .to(torch.bfloat16).contiguous())
                 wt_t = self._cached("wt_t", lambda: self.weight.detach().view(
                     self.out_channels, c).to(torch.bfloat16).t().contiguous())
-                y2d = _Conv1x1Hip.apply(x2d, wt, wt_t)
+                y2d = _Conv1x1Hip.apply(x2d, wt, w_bf16, wt_t)
             else:
                 y2d = x2d @ wt.t()
             return (
