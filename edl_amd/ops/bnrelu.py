@@ -63,6 +63,10 @@ class BNReLU2d(nn.Module):
         self.register_buffer("num_batches_tracked", torch.tensor(0, dtype=torch.long))
 
     def _use_fused(self, x, res=None):
+        import os
+
+        if os.environ.get("EDL_FUSED_BN", "1") != "1":
+            return False
         if not (x.is_cuda and x.dtype == torch.bfloat16 and available()):
             return False
         if self.num_features % 8 != 0 or self.num_features > 2048:

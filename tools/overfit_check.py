@@ -38,10 +38,13 @@ def main():
     eng.model.train()
     eng.set_lr(0.02)
     loss0 = None
+    curve = []
     for i in range(args.steps):
         loss = eng.train_step(x, y)
         if i == 0:
             loss0 = loss.item()
+        if i % 25 == 0 or i == args.steps - 1:
+            curve.append(round(loss.item(), 3))
     if dev.type == "cuda":
         torch.cuda.synchronize()
     with torch.no_grad():
@@ -49,7 +52,8 @@ def main():
             logits = eng.model(x)
         acc = (logits.argmax(1) == y).float().mean().item()
     print(json.dumps({"loss0": round(loss0, 4), "loss_end": round(loss.item(), 4),
-                      "acc_end": round(acc, 4), "steps": args.steps}))
+                      "acc_end": round(acc, 4), "steps": args.steps,
+                      "curve": curve}))
     assert loss.item() < loss0 * 0.5, "did not learn"
     assert acc > 0.9, "did not memorize"
 
