@@ -22,6 +22,7 @@ def main():
     ap.add_argument("--steps", type=int, default=150)
     ap.add_argument("--batch", type=int, default=64)
     ap.add_argument("--classes", type=int, default=16)
+    ap.add_argument("--seed", type=int, default=7)
     args = ap.parse_args()
 
     eng = TrainerEngine(model="resnet50_vd", per_device_batch=args.batch,
@@ -30,7 +31,7 @@ def main():
                         use_hip_ops=torch.cuda.is_available(),
                         graph_capture=False, checkpoint_dir=None).setup()
     dev = eng.device
-    torch.manual_seed(7)
+    torch.manual_seed(args.seed)
     x = torch.randn(args.batch, 3, 112, 112, device=dev)
     if dev.type == "cuda":
         x = x.to(torch.bfloat16).contiguous(memory_format=torch.channels_last)
