@@ -128,3 +128,24 @@ def test_fused_sgd_trains_resnet_gpu(tmp_path):
         loss = eng.train_step(x, y)
     torch.cuda.synchronize()
     assert loss.item() < l0
+
+
+def test_full_model_memorizes():
+    """End-to-end gradient correctness through EVERY custom kernel:
+    ResNet50_vd must drive a fixed 64-sample synthetic set to ~zero loss
+    (verified: both the custom path and the torch path reach acc 1.0 by
+    400 steps; 100-step snapshots differ only within seed-level
+    trajectory noise — gpurun_out/of_*)."""
+    import json
+    import os
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, os.path.join(repo, "tools", "overfit_check.py"),
+         "--steps", "400"],
+        capture_output=True, text=True, timeout=280, cwd=repo)
+    assert out.returncode == 0, out.stdout + out.stderr
+    res = json.loads([l for l in out.stdout.splitlines() if l.startswith("{")][-1])
+    assert res["acc_end"] > 0.95

@@ -19,7 +19,7 @@ from edl_amd.train.engine import TrainerEngine  # noqa: E402
 
 def main():
     ap = argparse.ArgumentParser()
-    ap.add_argument("--steps", type=int, default=150)
+    ap.add_argument("--steps", type=int, default=400)
     ap.add_argument("--batch", type=int, default=64)
     ap.add_argument("--classes", type=int, default=16)
     ap.add_argument("--seed", type=int, default=7)
@@ -55,8 +55,8 @@ def main():
     print(json.dumps({"loss0": round(loss0, 4), "loss_end": round(loss.item(), 4),
                       "acc_end": round(acc, 4), "steps": args.steps,
                       "curve": curve}))
-    assert loss.item() < loss0 * 0.5, "did not learn"
-    assert acc > 0.9, "did not memorize"
+    assert loss.item() < 0.1, "did not learn"
+    assert acc > 0.95, "did not memorize"
 
 
 if __name__ == "__main__":
