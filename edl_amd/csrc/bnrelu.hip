@@ -70,15 +70,19 @@ extern "C" __global__ void bn_stats_kernel(
 
   float s[8] = {0}, q[8] = {0};
   long long r = row0;
-  // 2x manual unroll: two independent row streams keep more loads in
-  // flight per thread (the single-stream loop measured latency-bound)
-  for (; r + rstride < M; r += 2 * rstride) {
+  // 4 independent row streams: the grid is capped (finalize reads the
+  // partials serially), so per-thread in-flight bytes must cover HBM
+  // latency — 1 stream measured ~1.2 TB/s, latency-bound
+  for (; r + 3 * rstride < M; r += 4 * rstride) {
     F8 v0 = load8(x + r * C + c0);
     F8 v1 = load8(x + (r + rstride) * C + c0);
+    F8 v2 = load8(x + (r + 2 * rstride) * C + c0);
+    F8 v3 = load8(x + (r + 3 * rstride) * C + c0);
 #pragma unroll
     for (int i = 0; i < 8; ++i) {
-      s[i] += v0.v[i] + v1.v[i];
-      q[i] = fmaf(v0.v[i], v0.v[i], fmaf(v1.v[i], v1.v[i], q[i]));
+      s[i] += (v0.v[i] + v1.v[i]) + (v2.v[i] + v3.v[i]);
+      q[i] = fmaf(v0.v[i], v0.v[i], fmaf(v1.v[i], v1.v[i],
+                  fmaf(v2.v[i], v2.v[i], fmaf(v3.v[i], v3.v[i], q[i]))));
     }
   }
   for (; r < M; r += rstride) {
@@ -211,23 +215,28 @@ __global__ void bn_bwd_reduce_kernel(
   }
   float s1[8] = {0}, s2[8] = {0};
   long long r = row0;
-  for (; r + rstride < M; r += 2 * rstride) {
-    const long long e0 = r * C + c0, e1 = (r + rstride) * C + c0;
-    F8 g0 = load8(dy + e0), g1 = load8(dy + e1);
-    F8 x0 = load8(x + e0), x1 = load8(x + e1);
-    if (RELU) {
-      F8 y0 = load8(y + e0), y1 = load8(y + e1);
+  // 4 row streams x 3 tensors = 12 outstanding 16-B loads per thread
+  for (; r + 3 * rstride < M; r += 4 * rstride) {
+#pragma unroll
+    for (int u = 0; u < 2; ++u) {
+      const long long e0 = (r + 2 * u * rstride) * C + c0;
+      const long long e1 = (r + (2 * u + 1) * rstride) * C + c0;
+      F8 g0 = load8(dy + e0), g1 = load8(dy + e1);
+      F8 x0 = load8(x + e0), x1 = load8(x + e1);
+      if (RELU) {
+        F8 y0 = load8(y + e0), y1 = load8(y + e1);
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+          g0.v[i] = y0.v[i] > 0.0f ? g0.v[i] : 0.0f;
+          g1.v[i] = y1.v[i] > 0.0f ? g1.v[i] : 0.0f;
+        }
+      }
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
-        g0.v[i] = y0.v[i] > 0.0f ? g0.v[i] : 0.0f;
-        g1.v[i] = y1.v[i] > 0.0f ? g1.v[i] : 0.0f;
+        s1[i] += g0.v[i] + g1.v[i];
+        s2[i] = fmaf(g0.v[i], (x0.v[i] - mu[i]) * is[i],
+                     fmaf(g1.v[i], (x1.v[i] - mu[i]) * is[i], s2[i]));
       }
-    }
-#pragma unroll
-    for (int i = 0; i < 8; ++i) {
-      s1[i] += g0.v[i] + g1.v[i];
-      s2[i] = fmaf(g0.v[i], (x0.v[i] - mu[i]) * is[i],
-                   fmaf(g1.v[i], (x1.v[i] - mu[i]) * is[i], s2[i]));
     }
   }
   for (; r < M; r += rstride) {
