@@ -27,7 +27,8 @@ typedef float f32x4 __attribute__((ext_vector_type(4)));
 
 extern __shared__ __attribute__((aligned(16))) char smem[];
 
-template <int BM, int BN, int WAVES_M, int WAVES_N, bool SPLITK = false>
+template <int BM, int BN, int WAVES_M, int WAVES_N, bool SPLITK = false,
+          bool V2 = false>
 __global__ __launch_bounds__(256, 2) void gemm_bt_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B,
     void* __restrict__ C_any, const int M, const int N, const int K) {
@@ -109,6 +110,37 @@ __global__ __launch_bounds__(256, 2) void gemm_bt_kernel(
         ((c ^ (r & 7)) << 4));
   };
 
+  // partial staging: issue only chunk-range [part*PARTS..] of the tile
+  auto stage_part = [&](int buf, int kt, int part, int nparts) {
+    const long long k0 = (long long)kt * BK;
+    char* abase = lds + buf * (A_BYTES + B_BYTES);
+    char* bbase = abase + A_BYTES;
+    // chunk ranges per part: [part*tot/nparts, (part+1)*tot/nparts)
+    const int a_tot = A_CHUNKS / 4, b_tot = B_CHUNKS / 4;
+#pragma unroll
+    for (int i = part * a_tot / nparts; i < (part + 1) * a_tot / nparts; ++i) {
+      const int ch = wave * (A_CHUNKS / 4) + i;
+      const int r = ch * 8 + (lane >> 3);
+      const int gslot = (lane & 7) ^ (r & 7);
+      long long grow = m0 + r;
+      if (grow >= M) grow = M - 1;
+      const bf16* src = A + grow * K + k0 + gslot * 8;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)(abase + ch * 1024), 16, 0, 0);
+    }
+#pragma unroll
+    for (int i = part * b_tot / nparts; i < (part + 1) * b_tot / nparts; ++i) {
+      const int ch = wave * (B_CHUNKS / 4) + i;
+      const int r = ch * 8 + (lane >> 3);
+      const int gslot = (lane & 7) ^ (r & 7);
+      const bf16* src = B + (long long)(n0 + r) * K + k0 + gslot * 8;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)(bbase + ch * 1024), 16, 0, 0);
+    }
+  };
+
   int kt_begin = 0, kt_end = K / BK;
   if (SPLITK) {
     const int per = (kt_end + gridDim.y - 1) / gridDim.y;
@@ -119,6 +151,38 @@ __global__ __launch_bounds__(256, 2) void gemm_bt_kernel(
   stage(0, kt_begin);
   __syncthreads();
 
+  // v2 main loop: 4 phases per K-tile — each phase prefetches a quarter of
+  // the NEXT tile, reads this phase's fragments, and runs an MFMA cluster
+  // under s_setprio(1); waves free-run across phases (role diversity), one
+  // explicit vmcnt(0) + raw barrier at the tile boundary.
+  if constexpr (V2) {
+    for (int kt = kt_begin; kt < kt_end; ++kt) {
+      const int cur = (kt - kt_begin) & 1;
+      const bool pre = kt + 1 < kt_end;
+#pragma unroll
+      for (int ph = 0; ph < 4; ++ph) {
+        const int kk = ph >> 1, mh = ph & 1;
+        if (pre) stage_part(cur ^ 1, kt + 1, ph, 4);
+        bf16x8 a[2], b[4];
+#pragma unroll
+        for (int m2 = 0; m2 < 2; ++m2) a[m2] = read_a(cur, mh * 2 + m2, kk);
+#pragma unroll
+        for (int nf = 0; nf < 4; ++nf) b[nf] = read_b(cur, nf, kk);
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int m2 = 0; m2 < 2; ++m2)
+#pragma unroll
+          for (int nf = 0; nf < 4; ++nf)
+            acc[mh * 2 + m2][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a[m2], b[nf], acc[mh * 2 + m2][nf], 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
+      }
+      // tile boundary: next tile's stages must land; all waves done
+      // reading buf cur before it is restaged next iteration
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+    }
+  } else
   for (int kt = kt_begin; kt < kt_end; ++kt) {
     const int cur = (kt - kt_begin) & 1;
     if (kt + 1 < kt_end) stage(cur ^ 1, kt + 1);
@@ -167,20 +231,38 @@ __global__ __launch_bounds__(256, 2) void gemm_bt_kernel(
   }
 }
 
+static bool use_v2() {
+  static const bool v = []() {
+    const char* e = getenv("EDL_GEMM_V2");
+    return e == nullptr || e[0] != '0';  // default ON; EDL_GEMM_V2=0 => v1
+  }();
+  return v;
+}
+
 extern "C" void launch_gemm_bt(const void* A, const void* B, void* C, int M,
                                int N, int K, hipStream_t s) {
   if (N % 128 == 0) {
     constexpr int BM = 128, BN = 128;
     const int grid = ((M + BM - 1) / BM) * (N / BN);
     const int lds_bytes = 2 * (BM * 64 * 2 + BN * 64 * 2);
-    hipLaunchKernelGGL((gemm_bt_kernel<BM, BN, 2, 2>), dim3(grid), dim3(256),
-                       lds_bytes, s, (const bf16*)A, (const bf16*)B, C, M, N, K);
+    if (use_v2())
+      hipLaunchKernelGGL((gemm_bt_kernel<BM, BN, 2, 2, false, true>), dim3(grid),
+                         dim3(256), lds_bytes, s, (const bf16*)A, (const bf16*)B,
+                         C, M, N, K);
+    else
+      hipLaunchKernelGGL((gemm_bt_kernel<BM, BN, 2, 2>), dim3(grid), dim3(256),
+                         lds_bytes, s, (const bf16*)A, (const bf16*)B, C, M, N, K);
   } else {  // N % 64 == 0
     constexpr int BM = 256, BN = 64;
     const int grid = ((M + BM - 1) / BM) * (N / BN);
     const int lds_bytes = 2 * (BM * 64 * 2 + BN * 64 * 2);
-    hipLaunchKernelGGL((gemm_bt_kernel<BM, BN, 4, 1>), dim3(grid), dim3(256),
-                       lds_bytes, s, (const bf16*)A, (const bf16*)B, C, M, N, K);
+    if (use_v2())
+      hipLaunchKernelGGL((gemm_bt_kernel<BM, BN, 4, 1, false, true>), dim3(grid),
+                         dim3(256), lds_bytes, s, (const bf16*)A, (const bf16*)B,
+                         C, M, N, K);
+    else
+      hipLaunchKernelGGL((gemm_bt_kernel<BM, BN, 4, 1>), dim3(grid), dim3(256),
+                         lds_bytes, s, (const bf16*)A, (const bf16*)B, C, M, N, K);
   }
 }
 
