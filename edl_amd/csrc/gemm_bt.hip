@@ -232,9 +232,13 @@ __global__ __launch_bounds__(256, 2) void gemm_bt_kernel(
 }
 
 static bool use_v2() {
+  // Measured (gpurun_out/gb_v1 vs gb_v2): the 4-phase interleave is 3-16%
+  // SLOWER than the plain 2-phase loop on every conv shape — consistent
+  // with the guide's finding that phase-splitting pays only as part of the
+  // full 8-phase/256-tile co-design. Kept for A/B; default OFF.
   static const bool v = []() {
     const char* e = getenv("EDL_GEMM_V2");
-    return e == nullptr || e[0] != '0';  // default ON; EDL_GEMM_V2=0 => v1
+    return e != nullptr && e[0] == '1';
   }();
   return v;
 }
