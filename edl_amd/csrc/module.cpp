@@ -41,6 +41,10 @@ extern "C" void launch_gemm_bt_splitk(const void*, const void*, float*, int, int
                                       int, int, hipStream_t);
 extern "C" void launch_shift9_transpose(const void*, void*, int, int, int, int,
                                         int, int, int, int, hipStream_t);
+extern "C" void launch_avgpool2x2_fwd(const void*, void*, int, int, int, int,
+                                      int, int, hipStream_t);
+extern "C" void launch_avgpool2x2_bwd(const void*, void*, int, int, int, int,
+                                      int, int, hipStream_t);
 extern "C" void launch_bn_bwd_dx(const void*, const void*, const void*, const float*,
                                  const float*, const float*, const float*, void*,
                                  void*, long long, int, bool, bool, bool,
@@ -206,6 +210,34 @@ torch::Tensor gemm_bt(torch::Tensor a, torch::Tensor b) {
   return c;
 }
 
+torch::Tensor avgpool2x2_fwd(torch::Tensor x) {
+  // x: 4-D channels_last bf16; 2x2 stride-2 ceil_mode pool
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 &&
+                  x.scalar_type() == torch::kBFloat16 &&
+                  x.is_contiguous(torch::MemoryFormat::ChannelsLast),
+              "avgpool2x2: 4-D channels_last bf16");
+  const int N = (int)x.size(0), C = (int)x.size(1);
+  const int H = (int)x.size(2), W = (int)x.size(3);
+  TORCH_CHECK(C % 8 == 0, "avgpool2x2: C % 8");
+  const int Ho = (H + 1) / 2, Wo = (W + 1) / 2;
+  auto y = torch::empty({N, C, Ho, Wo},
+                        x.options().memory_format(torch::MemoryFormat::ChannelsLast));
+  launch_avgpool2x2_fwd(x.data_ptr(), y.data_ptr(), N, H, W, Ho, Wo, C,
+                        cur_stream());
+  return y;
+}
+
+torch::Tensor avgpool2x2_bwd(torch::Tensor dy, int64_t H, int64_t W) {
+  const int N = (int)dy.size(0), C = (int)dy.size(1);
+  const int Ho = (int)dy.size(2), Wo = (int)dy.size(3);
+  auto dyc = dy.contiguous(torch::MemoryFormat::ChannelsLast);
+  auto dx = torch::empty({N, C, H, W},
+                         dy.options().memory_format(torch::MemoryFormat::ChannelsLast));
+  launch_avgpool2x2_bwd(dyc.data_ptr(), dx.data_ptr(), N, (int)H, (int)W, Ho,
+                        Wo, C, cur_stream());
+  return dx;
+}
+
 torch::Tensor conv3x3_wgrad_operand(torch::Tensor x, int64_t stride) {
   // x: 4-D channels_last bf16 input of the conv -> [9*Cin, Mp] shifted
   // transpose (pads internally).
@@ -313,5 +345,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("b"), pybind11::arg("splitk") = 0);
   m.def("conv3x3_wgrad_operand", &conv3x3_wgrad_operand,
         "padded shifted transpose of conv3x3 input -> [9*Cin, Mp]");
+  m.def("avgpool2x2_fwd", &avgpool2x2_fwd, "2x2/s2 ceil avg pool (NHWC bf16)");
+  m.def("avgpool2x2_bwd", &avgpool2x2_bwd, "2x2/s2 ceil avg pool backward");
   m.attr("_arch") = "gfx950";
 }

@@ -15,6 +15,7 @@ import torch.nn as nn
 
 from ..ops.bnrelu import BNAddReLU2d, BNReLU2d
 from ..ops.conv import Conv2dFast
+from ..ops.pool import AvgPool2x2
 
 
 class ConvBN(nn.Module):
@@ -38,7 +39,7 @@ class VdShortcut(nn.Module):
         if cin != cout or stride != 1:
             if stride != 1 and not if_first:
                 # the "vd" trick: downsample by avg-pool, then 1x1 stride 1
-                self.pool = nn.AvgPool2d(2, 2, ceil_mode=True)
+                self.pool = AvgPool2x2()
                 conv_stride = 1
             else:
                 conv_stride = stride

@@ -1,0 +1,35 @@
+"""AvgPool2x2 — the vd-shortcut downsample pool on CDNA4 kernels.
+
+Drop-in for nn.AvgPool2d(2, 2, ceil_mode=True) with no padding: with
+kernel == stride every input position feeds exactly one window, so the
+backward is elementwise (torch's NHWC avg_pool2d_backward measured
+95 us/dispatch). Plain-torch fallback everywhere else."""
+import torch
+import torch.nn as nn
+
+from . import available, ext
+
+
+class _AvgPool2x2Fn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        ctx.hw = (x.shape[2], x.shape[3])
+        return ext().avgpool2x2_fwd(x)
+
+    @staticmethod
+    def backward(ctx, dy):
+        h, w = ctx.hw
+        return ext().avgpool2x2_bwd(dy, h, w)
+
+
+class AvgPool2x2(nn.Module):
+    def __init__(self):
+        super().__init__()
+        self._fallback = nn.AvgPool2d(2, 2, ceil_mode=True)
+
+    def forward(self, x):
+        if (x.is_cuda and x.dtype == torch.bfloat16 and available()
+                and x.shape[1] % 8 == 0
+                and x.is_contiguous(memory_format=torch.channels_last)):
+            return _AvgPool2x2Fn.apply(x)
+        return self._fallback(x)

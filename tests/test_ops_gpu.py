@@ -149,3 +149,25 @@ def test_full_model_memorizes():
     assert out.returncode == 0, out.stdout + out.stderr
     res = json.loads([l for l in out.stdout.splitlines() if l.startswith("{")][-1])
     assert res["acc_end"] > 0.95
+
+
+@pytest.mark.parametrize("H,W", [(56, 56), (7, 7), (9, 11)])
+def test_avgpool2x2_numerics(H, W):
+    from edl_amd.ops.pool import AvgPool2x2
+
+    torch.manual_seed(4)
+    pool = AvgPool2x2().cuda()
+    x = torch.randn(3, 64, H, W, device="cuda").to(torch.bfloat16)
+    x = x.contiguous(memory_format=torch.channels_last).requires_grad_(True)
+    y = pool(x)
+    g = torch.randn_like(y).contiguous(memory_format=torch.channels_last)
+    y.backward(g)
+
+    xr = x.detach().float().requires_grad_(True)
+    yr = torch.nn.AvgPool2d(2, 2, ceil_mode=True)(xr)
+    yr.backward(g.float())
+    assert y.shape == yr.shape
+    assert torch.allclose(y.float(), yr.detach(), atol=2e-2, rtol=2e-2), \
+        (y.float() - yr.detach()).abs().max().item()
+    assert torch.allclose(x.grad.float(), xr.grad, atol=2e-2, rtol=2e-2), \
+        (x.grad.float() - xr.grad).abs().max().item()
