@@ -19,8 +19,6 @@ import os
 import queue
 import threading
 
-import numpy as np
-
 from ..utils.log import get_logger
 from . import worker as W
 from .discovery import DiscoveryClient

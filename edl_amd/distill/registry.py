@@ -5,7 +5,6 @@ TCP-alive-checked TTL heartbeat) + server_alive.py:19-33. Teachers appear
 under /<job>/service/nodes/<service_name>/<endpoint>; their key expires
 with the lease when they die."""
 import socket
-import threading
 import time
 
 from ..coord.register import Register

@@ -20,7 +20,6 @@ from ..utils.log import get_logger
 from . import dist as edist
 from .bucketed_ddp import BucketedAllReducer
 from .checkpoint import CheckpointManager
-from .env import TrainerEnv
 
 log = get_logger("edl.engine")
 

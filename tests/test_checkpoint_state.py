@@ -1,5 +1,4 @@
 """Checkpoint manager + elastic train state tests."""
-import json
 import os
 
 import torch

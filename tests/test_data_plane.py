@@ -1,9 +1,5 @@
 """Data-plane tests (mirror reference test_data_server.py: file-list
 slicing + meta balance over a real server on 127.0.0.1)."""
-import pickle
-
-import pytest
-
 from edl_amd.data.data_server import DataClient, DataServer, PodsData
 from edl_amd.data.dataset import TxtFileSplitter
 from edl_amd.data.reader import Reader
