@@ -18,6 +18,7 @@ and fills slots for new ones (reference predict_manage_worker 58-171)."""
 import os
 import queue
 import threading
+import time
 
 from ..utils.log import get_logger
 from . import worker as W
@@ -113,17 +114,14 @@ class DistillReader:
 
         assigned = {s: None for s in range(n)}
         teachers = []
-        deadline = 120
-        import time as _t
-
-        t0 = _t.monotonic()
+        t0 = time.monotonic()
         while not teachers:
             teachers = self._current_teachers()
             if teachers or W._NOP_PREDICT_TEST:
                 break
-            if _t.monotonic() - t0 > deadline:
+            if time.monotonic() - t0 > 120:
                 raise TimeoutError("no teachers discovered")
-            _t.sleep(0.5)
+            time.sleep(0.5)
 
         procs = []
         reader = threading.Thread(
@@ -168,7 +166,7 @@ class DistillReader:
         )
         manager.start()
         # seed initial assignment immediately
-        for slot, ep in zip(range(n), sorted(teachers) or [None] * 0):
+        for slot, ep in zip(range(n), sorted(teachers)):
             assigned[slot] = ep
             server_queue.put(W.ServerItem(slot, ep))
         if W._NOP_PREDICT_TEST and not teachers:
