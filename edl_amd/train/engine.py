@@ -57,6 +57,9 @@ class TrainerEngine:
         kd_teacher=None,
         kd_alpha=1.0,
         recompute=False,
+        dgc=False,
+        dgc_ratio=0.01,
+        dgc_rampup=0,
     ):
         self.model_name = model
         self.per_device_batch = per_device_batch
@@ -76,6 +79,8 @@ class TrainerEngine:
         self.kd_alpha = kd_alpha
         self.recompute = recompute  # activation checkpointing (reference
         # dist_strategy.forward_recompute, train_with_fleet.py:322-325)
+        self.dgc_cfg = (dgc, dgc_ratio, dgc_rampup)  # optional sparse allreduce
+        self.dgc = None
 
         self.env = None
         self.device = None
@@ -120,6 +125,11 @@ class TrainerEngine:
         self.reducer = BucketedAllReducer(
             self.model.parameters(), bucket_cap_mb=self.bucket_mb
         )
+        if self.dgc_cfg[0]:
+            from .dgc import DGCCompressor
+
+            self.dgc = DGCCompressor(self.reducer, self.dgc_cfg[1],
+                                     self.dgc_cfg[2])
         self.opt = self._build_optimizer()
         if self.checkpoint_dir:
             self.ckpt = CheckpointManager(self.checkpoint_dir)
@@ -217,7 +227,10 @@ class TrainerEngine:
             logits = self.model(images)
         loss = self._loss(logits, labels, teacher_logits)
         loss.backward()
-        self.reducer.finalize()
+        if self.dgc is not None:
+            self.dgc.step()
+        else:
+            self.reducer.finalize()
         if not getattr(self.opt, "handles_grad_scale", False):
             scale = self.reducer.grad_scale
             if scale != 1.0:

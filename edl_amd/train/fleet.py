@@ -80,6 +80,8 @@ def distributed_engine(model_name="resnet50_vd", strategy=None, **kwargs):
     kwargs.setdefault("bucket_mb", strategy.fuse_all_reduce_ops_mb)
     kwargs.setdefault("dtype", "bf16" if strategy.use_amp and
                       torch.cuda.is_available() else "fp32")
+    kwargs.setdefault("dgc", strategy.use_dgc)
+    kwargs.setdefault("recompute", strategy.use_recompute)
     eng = TrainerEngine(model=model_name, **kwargs)
     eng.setup(_state["env"])
     return eng
