@@ -51,7 +51,12 @@ extern "C" __global__ void pad_nhwc_kernel(
   (void)total;
 }
 
-template <int BM, int BN, int WAVES_M, int WAVES_N>
+// GROUPED: grouped conv with 16 in/out channels per group (ResNeXt 32x16d
+// shapes). An n-tile of 64 output channels = 4 groups = ONE contiguous
+// 64-channel input block at channel base n0; the gemm runs K = 9*64
+// against a block-diagonal zero-padded weight repack (4x MFMA work on
+// zeros, but ~MFMA rate vs MIOpen's grouped path).
+template <int BM, int BN, int WAVES_M, int WAVES_N, bool GROUPED = false>
 __global__ __launch_bounds__(256, 2) void conv3x3_kernel(
     const bf16* __restrict__ XP, const bf16* __restrict__ B,
     bf16* __restrict__ C_out, const int M, const int N, const int Cin,
@@ -60,8 +65,9 @@ __global__ __launch_bounds__(256, 2) void conv3x3_kernel(
   constexpr int BK = 64;
   constexpr int A_BYTES = BM * BK * 2;
   constexpr int B_BYTES = BN * BK * 2;
-  const int K = 9 * Cin;          // gemm K
-  const int cb_per_s = Cin >> 6;  // 64-wide channel blocks per shift
+  const int Cin_k = GROUPED ? 64 : Cin;  // channels entering the gemm K
+  const int K = 9 * Cin_k;               // gemm K
+  const int cb_per_s = Cin_k >> 6;       // 64-wide channel blocks per shift
   char* lds = smem;
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -96,6 +102,7 @@ __global__ __launch_bounds__(256, 2) void conv3x3_kernel(
     shift_elems[s] = ((s / 3) * Wp + (s % 3)) * Cin;
   }
 
+  const int a_ch_base = GROUPED ? n0 : 0;  // group-block channel base
   auto stage = [&](int buf, int kt) {
     const int s = kt / cb_per_s;
     const int cb = kt % cb_per_s;
@@ -106,7 +113,8 @@ __global__ __launch_bounds__(256, 2) void conv3x3_kernel(
       const int ch = wave * (A_CHUNKS / 4) + i;
       const int r = ch * 8 + (lane >> 3);
       const int gslot = (lane & 7) ^ (r & 7);
-      const bf16* src = XP + arow[i] + shift_elems[s] + cb * 64 + gslot * 8;
+      const bf16* src = XP + arow[i] + shift_elems[s] + a_ch_base + cb * 64 +
+                        gslot * 8;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) void*)src,
           (__attribute__((address_space(3))) void*)(abase + ch * 1024), 16, 0, 0);
@@ -189,6 +197,19 @@ extern "C" void launch_pad_nhwc(const void* x, void* xp, int Nimg, int H, int W,
   if (gx > 1024) gx = 1024;
   hipLaunchKernelGGL(pad_nhwc_kernel, dim3(gx, Nimg), dim3(256), 0, s,
                      (const bf16*)x, (bf16*)xp, H * W, H, W, Hp, Wp, C);
+}
+
+extern "C" void launch_conv3x3_grouped(const void* xp, const void* w3g, void* y,
+                                       int M, int Cout, int Cin, int HW_out,
+                                       int W_out, int Hp, int Wp, int stride,
+                                       hipStream_t s) {
+  // BN must be 64 (= 4 groups of 16); w3g is [Cout, 9*64] block-diagonal
+  constexpr int BM = 256, BN = 64;
+  const int grid = ((M + BM - 1) / BM) * (Cout / BN);
+  const int lds_bytes = 2 * (BM * 64 * 2 + BN * 64 * 2);
+  hipLaunchKernelGGL((conv3x3_kernel<BM, BN, 4, 1, true>), dim3(grid),
+                     dim3(256), lds_bytes, s, (const bf16*)xp, (const bf16*)w3g,
+                     (bf16*)y, M, Cout, Cin, HW_out, W_out, Hp, Wp, stride);
 }
 
 extern "C" void launch_conv3x3(const void* xp, const void* w3, void* y, int M,

@@ -35,6 +35,8 @@ extern "C" void launch_pad_nhwc(const void*, void*, int, int, int, int, int, int
                                 hipStream_t);
 extern "C" void launch_conv3x3(const void*, const void*, void*, int, int, int,
                                int, int, int, int, int, hipStream_t);
+extern "C" void launch_conv3x3_grouped(const void*, const void*, void*, int, int,
+                                       int, int, int, int, int, int, hipStream_t);
 extern "C" void launch_transpose_pad(const void*, void*, int, int, int,
                                      hipStream_t);
 extern "C" void launch_gemm_bt_splitk(const void*, const void*, float*, int, int,
@@ -210,6 +212,33 @@ torch::Tensor gemm_bt(torch::Tensor a, torch::Tensor b) {
   return c;
 }
 
+torch::Tensor conv3x3_grouped_fwd(torch::Tensor x, torch::Tensor w3g,
+                                  int64_t stride) {
+  // grouped 3x3 (16 in/out channels per group): w3g [Cout, 9*64]
+  // block-diagonal repack; 4 groups per 64-wide n-tile.
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 &&
+                  x.scalar_type() == torch::kBFloat16 &&
+                  x.is_contiguous(torch::MemoryFormat::ChannelsLast),
+              "conv3x3g: 4-D channels_last bf16");
+  const int Nimg = (int)x.size(0), Cin = (int)x.size(1);
+  const int H = (int)x.size(2), W = (int)x.size(3);
+  const int Cout = (int)w3g.size(0);
+  TORCH_CHECK(w3g.size(1) == 9 * 64 && w3g.is_contiguous(), "conv3x3g: w3g");
+  TORCH_CHECK(Cin % 64 == 0 && Cout % 64 == 0 && Cin == Cout,
+              "conv3x3g: C % 64, equal in/out");
+  const int Hp = H + 2, Wp = W + 2;
+  const int Hout = (H + 2 - 3) / (int)stride + 1;
+  const int Wout = (W + 2 - 3) / (int)stride + 1;
+  const long long M = (long long)Nimg * Hout * Wout;
+  auto s = cur_stream();
+  auto xp = torch::empty({(long long)Nimg * Hp * Wp * Cin}, x.options());
+  launch_pad_nhwc(x.data_ptr(), xp.data_ptr(), Nimg, H, W, Hp, Wp, Cin, s);
+  auto y = torch::empty({M, Cout}, x.options());
+  launch_conv3x3_grouped(xp.data_ptr(), w3g.data_ptr(), y.data_ptr(), (int)M,
+                         Cout, Cin, Hout * Wout, Wout, Hp, Wp, (int)stride, s);
+  return y;
+}
+
 torch::Tensor avgpool2x2_fwd(torch::Tensor x) {
   // x: 4-D channels_last bf16; 2x2 stride-2 ceil_mode pool
   TORCH_CHECK(x.is_cuda() && x.dim() == 4 &&
@@ -346,6 +375,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv3x3_wgrad_operand", &conv3x3_wgrad_operand,
         "padded shifted transpose of conv3x3 input -> [9*Cin, Mp]");
   m.def("avgpool2x2_fwd", &avgpool2x2_fwd, "2x2/s2 ceil avg pool (NHWC bf16)");
+  m.def("conv3x3_grouped_fwd", &conv3x3_grouped_fwd,
+        "grouped (16ch/group) 3x3 conv fwd -> y2d [M, Cout]");
   m.def("avgpool2x2_bwd", &avgpool2x2_bwd, "2x2/s2 ceil avg pool backward");
   m.attr("_arch") = "gfx950";
 }

@@ -124,6 +124,23 @@ class _Conv3x3Hip(torch.autograd.Function):
         return dx, dw, None, None, None, None
 
 
+def _repack_w3_grouped(weight):
+    """Grouped 3x3 (16 in/out ch per group) -> block-diagonal [Cout, 9*64]:
+    the 4 groups of each 64-wide output tile share one contiguous 64-channel
+    input block; out-channel n's 16 true input channels sit at lane block
+    (n//16)%4 and the rest are zeros (4x MFMA work on zeros ~ still far
+    above MIOpen's grouped path)."""
+    co, cpg, _, _ = weight.shape  # [Cout, 16, 3, 3]
+    assert cpg == 16
+    w = weight.detach().to(torch.bfloat16)
+    out = torch.zeros(co, 3, 3, 64, dtype=torch.bfloat16, device=w.device)
+    pos = (torch.arange(co, device=w.device) // 16) % 4
+    for p in range(4):
+        m = pos == p
+        out[m, :, :, p * 16:(p + 1) * 16] = w[m].permute(0, 2, 3, 1)
+    return out.reshape(co, 9 * 64).contiguous()
+
+
 class Conv2dFast(nn.Conv2d):
     """nn.Conv2d drop-in; 1x1/group-1 convs on CUDA bypass MIOpen."""
 
@@ -163,6 +180,32 @@ class Conv2dFast(nn.Conv2d):
                     self.weight.detach().permute(1, 0, 2, 3).flip(2, 3)))
             return _Conv3x3Hip.apply(x, self.weight, w_bf16, w3, w3rot,
                                      self.stride[0])
+        if (
+            _CONV3X3 == "hip"
+            and x.is_cuda
+            and available()
+            and x.dtype == torch.bfloat16
+            and self.kernel_size == (3, 3)
+            and self.groups > 1
+            and self.in_channels == self.out_channels
+            and self.in_channels // self.groups == 16
+            and self.in_channels % 64 == 0
+            and self.padding == (1, 1)
+            and self.stride[0] == self.stride[1]
+            and self.stride[0] in (1, 2)
+            and self.bias is None
+            and not (torch.is_grad_enabled() and
+                     (x.requires_grad or self.weight.requires_grad))
+        ):
+            # grouped teacher convs: eval/inference-only fast path
+            if not x.is_contiguous(memory_format=torch.channels_last):
+                x = x.contiguous(memory_format=torch.channels_last)
+            w3g = self._cached("w3g", lambda: _repack_w3_grouped(self.weight))
+            y2d = ext().conv3x3_grouped_fwd(x, w3g, self.stride[0])
+            n, _, h, w = x.shape
+            ho = (h - 1) // self.stride[0] + 1
+            wo = (w - 1) // self.stride[0] + 1
+            return y2d.view(n, ho, wo, self.out_channels).permute(0, 3, 1, 2)
         if (
             _CONV1X1 != "miopen"
             and x.is_cuda

@@ -131,3 +131,39 @@ def test_gemm_bt_splitk_numerics(M, N, K, splitk):
     err = (c - ref).abs()
     scale = ref.abs().mean().clamp(min=1)
     assert (err / scale).max() < 0.05, (err.max().item(), scale.item())
+
+
+@pytest.mark.parametrize("stride", [1, 2])
+def test_conv3x3_grouped_eval_numerics(stride):
+    """Grouped 3x3 (ResNeXt 16ch/group) inference fast path vs F.conv2d."""
+    import torch.nn.functional as F
+
+    from edl_amd.ops.conv import Conv2dFast
+
+    torch.manual_seed(5)
+    C, groups = 128, 8  # 16 ch/group
+    conv = Conv2dFast(C, C, 3, stride=stride, padding=1, groups=groups,
+                      bias=False).cuda().to(torch.bfloat16)
+    x = torch.randn(2, C, 14, 14, device="cuda").to(torch.bfloat16)
+    x = x.contiguous(memory_format=torch.channels_last)
+    with torch.no_grad():
+        y = conv(x)
+        ref = F.conv2d(x.float(), conv.weight.float(), stride=stride,
+                       padding=1, groups=groups)
+    assert y.shape == ref.shape
+    err = (y.float() - ref).abs()
+    scale = ref.abs().mean().clamp(min=0.2)
+    assert (err / scale).max() < 0.1, (err.max().item(), scale.item())
+
+
+def test_conv3x3_grouped_training_falls_back():
+    """With grad enabled the grouped fast path must NOT engage (fwd-only)."""
+    from edl_amd.ops.conv import Conv2dFast
+
+    conv = Conv2dFast(64, 64, 3, padding=1, groups=4, bias=False
+                      ).cuda().to(torch.bfloat16)
+    x = torch.randn(1, 64, 8, 8, device="cuda").to(torch.bfloat16)
+    x = x.contiguous(memory_format=torch.channels_last).requires_grad_(True)
+    y = conv(x)  # MIOpen path; must be differentiable
+    y.sum().backward()
+    assert x.grad is not None and conv.weight.grad is not None
