@@ -181,7 +181,10 @@ class Conv2dFast(nn.Conv2d):
             return _Conv3x3Hip.apply(x, self.weight, w_bf16, w3, w3rot,
                                      self.stride[0])
         if (
-            _CONV3X3 == "hip"
+            # Measured SLOWER than MIOpen's grouped path on the ResNeXt
+            # teacher (distill 486 -> 415 img/s): the 4x block-diagonal
+            # zero work does not pay at these widths. Opt-in for A/B.
+            os.environ.get("EDL_CONV3X3_GROUPED") == "hip"
             and x.is_cuda
             and available()
             and x.dtype == torch.bfloat16

@@ -134,11 +134,15 @@ def test_gemm_bt_splitk_numerics(M, N, K, splitk):
 
 
 @pytest.mark.parametrize("stride", [1, 2])
-def test_conv3x3_grouped_eval_numerics(stride):
-    """Grouped 3x3 (ResNeXt 16ch/group) inference fast path vs F.conv2d."""
+def test_conv3x3_grouped_eval_numerics(stride, monkeypatch):
+    """Grouped 3x3 (ResNeXt 16ch/group) inference fast path vs F.conv2d.
+    Opt-in path (EDL_CONV3X3_GROUPED=hip): measured slower than MIOpen on
+    the teacher shapes, kept correct for future tile work."""
     import torch.nn.functional as F
 
     from edl_amd.ops.conv import Conv2dFast
+
+    monkeypatch.setenv("EDL_CONV3X3_GROUPED", "hip")
 
     torch.manual_seed(5)
     C, groups = 128, 8  # 16 ch/group
