@@ -112,6 +112,8 @@ def main():
     img_per_s = total_imgs / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
 
+    peak_gb = (round(torch.cuda.max_memory_allocated() / 2**30, 2)
+               if use_cuda else None)
     if rank == 0:
         result = {
             "metric": "img/s",
@@ -132,6 +134,7 @@ def main():
                 "image_shape": "3x224x224",
                 "parallelism": "dp%d" % world,
                 "graph_capture": engine._graph is not None,
+                "peak_mem_gb_per_gpu": peak_gb,
             },
         }
         print(json.dumps(result))
