@@ -89,6 +89,7 @@ class TrainerEngine:
         self.opt = None
         self.ckpt = None
         self.start_epoch = 0
+        self.start_step = 0  # step within start_epoch (mid-epoch resume)
         self.global_step = 0
         self._graph = None
         self._static = {}
@@ -184,12 +185,21 @@ class TrainerEngine:
                 self.opt.load_state_dict(opt_state)
             except (ValueError, KeyError) as e:
                 log.warning("optimizer state mismatch (%s); reset", e)
-        self.start_epoch = int(ts.get("epoch_no", -1)) + 1
+        if ts.get("mid_epoch"):
+            # step-level checkpoint: resume INSIDE the epoch (beyond the
+            # reference's per-epoch granularity, doc/fault_tolerance.md
+            # "step level ... next version")
+            self.start_epoch = int(ts.get("epoch_no", 0))
+            self.start_step = int(ts.get("step_in_epoch", 0))
+        else:
+            self.start_epoch = int(ts.get("epoch_no", -1)) + 1
+            self.start_step = 0
         self.global_step = int(ts.get("global_step", 0))
         if self.env.is_rank0:
             log.info(
-                "resumed from checkpoint v%s: next epoch %d, global_step %d (world=%d)",
-                ts.get("_version"), self.start_epoch, self.global_step, self.world_size,
+                "resumed from checkpoint v%s: epoch %d step %d, global_step %d (world=%d)",
+                ts.get("_version"), self.start_epoch, self.start_step,
+                self.global_step, self.world_size,
             )
 
     def save_checkpoint(self, epoch, extra=None, blocking=False):
@@ -297,12 +307,13 @@ class TrainerEngine:
         return self.train_step(images, labels)
 
     # ---- epoch driver ----
-    def train_epoch(self, epoch, loader, steps, log_every=50, on_step=None):
+    def train_epoch(self, epoch, loader, steps, log_every=50, on_step=None,
+                    start_step=0):
         self.model.train()
         t0 = time.monotonic()
         imgs_done = 0
         loss = None
-        for it in range(steps):
+        for it in range(start_step, steps):
             self.set_lr(self.scaled_lr(epoch, it / max(1, steps)))
             x, y = loader.next()
             loss = self.replay_step(x, y)

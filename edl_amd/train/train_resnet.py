@@ -93,13 +93,16 @@ def main(argv=None):
 
     def on_step(epoch, it):
         if args.checkpoint_steps and (engine.global_step % args.checkpoint_steps == 0):
-            engine.save_checkpoint(epoch - 1, extra={"mid_epoch": True})
+            engine.save_checkpoint(
+                epoch, extra={"mid_epoch": True, "step_in_epoch": it + 1})
 
     t_start = time.monotonic()
     for epoch in range(engine.start_epoch, args.num_epochs):
         if epoch == args.num_epochs - 1:
             report(TrainStatus.NEARTHEEND)
-        stats = engine.train_epoch(epoch, loader, args.steps_per_epoch, on_step=on_step)
+        stats = engine.train_epoch(
+            epoch, loader, args.steps_per_epoch, on_step=on_step,
+            start_step=engine.start_step if epoch == engine.start_epoch else 0)
         if engine.env.is_rank0:
             log.info("epoch %d done: %.1f img/s (world=%d, global_batch=%d)",
                      epoch, stats["img_per_s"], engine.world_size, engine.global_batch)

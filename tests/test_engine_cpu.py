@@ -205,3 +205,35 @@ def test_bucketed_allreduce_world2_matches_fullbatch(tmp_path):
         opt.step()
     ref = float(sum(p.sum() for p in m.parameters()))
     assert res["params_sum"] == pytest.approx(ref, abs=1e-4)
+
+
+def test_mid_epoch_step_resume(tmp_path):
+    """Step-level checkpoint: resume INSIDE an epoch (beyond the
+    reference's per-epoch granularity)."""
+    eng = TrainerEngine(model="resnet18_vd", per_device_batch=2, num_classes=8,
+                        dtype="fp32", channels_last=False, use_hip_ops=False,
+                        base_lr=0.001, checkpoint_dir=str(tmp_path / "ck"))
+    eng.setup()
+    x = torch.randn(2, 3, 32, 32)
+    y = torch.randint(0, 8, (2,))
+    for _ in range(5):
+        eng.train_step(x, y)
+    eng.save_checkpoint(3, extra={"mid_epoch": True, "step_in_epoch": 5},
+                        blocking=True)
+    eng.ckpt.wait()
+
+    eng2 = TrainerEngine(model="resnet18_vd", per_device_batch=2, num_classes=8,
+                         dtype="fp32", channels_last=False, use_hip_ops=False,
+                         base_lr=0.001, checkpoint_dir=str(tmp_path / "ck"))
+    eng2.setup()
+    assert eng2.start_epoch == 3
+    assert eng2.start_step == 5
+    assert eng2.global_step == 5
+    # epoch-level checkpoints still resume at the NEXT epoch
+    eng2.save_checkpoint(3, blocking=True)
+    eng2.ckpt.wait()
+    eng3 = TrainerEngine(model="resnet18_vd", per_device_batch=2, num_classes=8,
+                         dtype="fp32", channels_last=False, use_hip_ops=False,
+                         base_lr=0.001, checkpoint_dir=str(tmp_path / "ck"))
+    eng3.setup()
+    assert eng3.start_epoch == 4 and eng3.start_step == 0
