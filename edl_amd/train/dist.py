@@ -23,10 +23,10 @@ def init_from_env(env=None, timeout_s=120):
     tenv = env or TrainerEnv()
     use_cuda = torch.cuda.is_available()
     if use_cuda:
-        local_gpu = tenv.rank_in_pod if not tenv.gpus else tenv.rank_in_pod
         # FLAGS_selected_gpus indexes into CUDA_VISIBLE_DEVICES order;
-        # each trainer owns one visible device slot.
-        device = torch.device("cuda", local_gpu % max(1, torch.cuda.device_count()))
+        # each trainer owns one visible device slot (= its local rank).
+        device = torch.device(
+            "cuda", tenv.rank_in_pod % max(1, torch.cuda.device_count()))
         torch.cuda.set_device(device)
         backend = "nccl"
     else:
