@@ -156,10 +156,13 @@ extern "C" __global__ void bn_finalize_kernel(
 }
 
 // K3: y = relu?(x*scale + shift [+ res]); NHWC vectorized by channel octets.
-template <bool RELU, bool ADD>
+// MASK: also emit a 1-bit-per-channel ReLU mask (one byte per octet) so the
+// backward kernels never re-read y (16x fewer bytes on that stream).
+template <bool RELU, bool ADD, bool MASK = false>
 __global__ void bn_apply_kernel(
     const bf16* __restrict__ x, const bf16* __restrict__ res,
-    bf16* __restrict__ y, const float* __restrict__ scale,
+    bf16* __restrict__ y, unsigned char* __restrict__ mask,
+    const float* __restrict__ scale,
     const float* __restrict__ shift, const long long M, const int C) {
   const int c8 = C >> 3;
   const long long total = M * c8;
@@ -178,8 +181,13 @@ __global__ void bn_apply_kernel(
       for (int k = 0; k < 8; ++k) v.v[k] += rv.v[k];
     }
     if (RELU) {
+      unsigned char mb = 0;
 #pragma unroll
-      for (int k = 0; k < 8; ++k) v.v[k] = fmaxf(v.v[k], 0.0f);
+      for (int k = 0; k < 8; ++k) {
+        if (v.v[k] > 0.0f) mb |= (1u << k);
+        v.v[k] = fmaxf(v.v[k], 0.0f);
+      }
+      if (MASK) mask[i] = mb;
     }
     store8(y + eoff, v);
   }
@@ -188,10 +196,10 @@ __global__ void bn_apply_kernel(
 // ---------------- backward ----------------
 
 // B1: per-channel reductions s1 = sum(dy_eff), s2 = sum(dy_eff * xhat),
-// where dy_eff applies the fused ReLU mask (y > 0).
+// where dy_eff applies the fused ReLU mask (bitmask from the fwd apply).
 template <bool RELU>
 __global__ void bn_bwd_reduce_kernel(
-    const bf16* __restrict__ dy, const bf16* __restrict__ y,
+    const bf16* __restrict__ dy, const unsigned char* __restrict__ mask,
     const bf16* __restrict__ x, const float* __restrict__ mean,
     const float* __restrict__ invstd, float* __restrict__ sums,  // [2, C]
     const long long M, const int C) {
@@ -219,16 +227,17 @@ __global__ void bn_bwd_reduce_kernel(
   for (; r + 3 * rstride < M; r += 4 * rstride) {
 #pragma unroll
     for (int u = 0; u < 2; ++u) {
-      const long long e0 = (r + 2 * u * rstride) * C + c0;
-      const long long e1 = (r + (2 * u + 1) * rstride) * C + c0;
+      const long long r0 = r + 2 * u * rstride, r1 = r + (2 * u + 1) * rstride;
+      const long long e0 = r0 * C + c0, e1 = r1 * C + c0;
       F8 g0 = load8(dy + e0), g1 = load8(dy + e1);
       F8 x0 = load8(x + e0), x1 = load8(x + e1);
       if (RELU) {
-        F8 y0 = load8(y + e0), y1 = load8(y + e1);
+        const unsigned m0 = mask[r0 * c8 + lane_c];
+        const unsigned m1 = mask[r1 * c8 + lane_c];
 #pragma unroll
         for (int i = 0; i < 8; ++i) {
-          g0.v[i] = y0.v[i] > 0.0f ? g0.v[i] : 0.0f;
-          g1.v[i] = y1.v[i] > 0.0f ? g1.v[i] : 0.0f;
+          g0.v[i] = (m0 >> i) & 1 ? g0.v[i] : 0.0f;
+          g1.v[i] = (m1 >> i) & 1 ? g1.v[i] : 0.0f;
         }
       }
 #pragma unroll
@@ -244,9 +253,9 @@ __global__ void bn_bwd_reduce_kernel(
     F8 g = load8(dy + eoff);
     F8 xv = load8(x + eoff);
     if (RELU) {
-      F8 yv = load8(y + eoff);
+      const unsigned mb = mask[r * c8 + lane_c];
 #pragma unroll
-      for (int i = 0; i < 8; ++i) g.v[i] = yv.v[i] > 0.0f ? g.v[i] : 0.0f;
+      for (int i = 0; i < 8; ++i) g.v[i] = (mb >> i) & 1 ? g.v[i] : 0.0f;
     }
 #pragma unroll
     for (int i = 0; i < 8; ++i) {
@@ -288,7 +297,7 @@ extern "C" __global__ void bn_bwd_finalize_kernel(
 // B3: dx = scale * (dy_eff - s1/M - xhat * s2/M); optional dres = dy_eff.
 template <bool RELU, bool ADD, bool TRAINING>
 __global__ void bn_bwd_dx_kernel(
-    const bf16* __restrict__ dy, const bf16* __restrict__ y,
+    const bf16* __restrict__ dy, const unsigned char* __restrict__ mask,
     const bf16* __restrict__ x, const float* __restrict__ mean,
     const float* __restrict__ invstd, const float* __restrict__ gamma,
     const float* __restrict__ sums, bf16* __restrict__ dx,
@@ -304,9 +313,9 @@ __global__ void bn_bwd_dx_kernel(
     const long long eoff = (i / c8) * C + c0;
     F8 g = load8(dy + eoff);
     if (RELU) {
-      F8 yv = load8(y + eoff);
+      const unsigned mb = mask[i];
 #pragma unroll
-      for (int k = 0; k < 8; ++k) g.v[k] = yv.v[k] > 0.0f ? g.v[k] : 0.0f;
+      for (int k = 0; k < 8; ++k) g.v[k] = (mb >> k) & 1 ? g.v[k] : 0.0f;
     }
     if (ADD) store8(dres + eoff, g);
     F8 xv = load8(x + eoff);
@@ -368,32 +377,46 @@ extern "C" void launch_bn_bwd_finalize(const float* partial, int nblocks,
 }
 
 extern "C" void launch_bn_apply(const void* x, const void* res, void* y,
+                                unsigned char* mask,
                                 const float* scale, const float* shift,
                                 long long M, int C, bool relu, bool add,
                                 hipStream_t s) {
   const long long total = M * (C >> 3);
   const int grid = elementwise_grid(total, 256);
-  if (relu && add)
-    hipLaunchKernelGGL((bn_apply_kernel<true, true>), dim3(grid), dim3(256), 0, s,
-                       (const bf16*)x, (const bf16*)res, (bf16*)y, scale, shift, M, C);
-  else if (relu)
-    hipLaunchKernelGGL((bn_apply_kernel<true, false>), dim3(grid), dim3(256), 0, s,
-                       (const bf16*)x, nullptr, (bf16*)y, scale, shift, M, C);
-  else if (add)
-    hipLaunchKernelGGL((bn_apply_kernel<false, true>), dim3(grid), dim3(256), 0, s,
-                       (const bf16*)x, (const bf16*)res, (bf16*)y, scale, shift, M, C);
-  else
-    hipLaunchKernelGGL((bn_apply_kernel<false, false>), dim3(grid), dim3(256), 0, s,
-                       (const bf16*)x, nullptr, (bf16*)y, scale, shift, M, C);
+  const bf16* xb = (const bf16*)x;
+  const bf16* rb = (const bf16*)res;
+  bf16* yb = (bf16*)y;
+  if (relu && add) {
+    if (mask)
+      hipLaunchKernelGGL((bn_apply_kernel<true, true, true>), dim3(grid),
+                         dim3(256), 0, s, xb, rb, yb, mask, scale, shift, M, C);
+    else
+      hipLaunchKernelGGL((bn_apply_kernel<true, true>), dim3(grid), dim3(256),
+                         0, s, xb, rb, yb, nullptr, scale, shift, M, C);
+  } else if (relu) {
+    if (mask)
+      hipLaunchKernelGGL((bn_apply_kernel<true, false, true>), dim3(grid),
+                         dim3(256), 0, s, xb, nullptr, yb, mask, scale, shift, M, C);
+    else
+      hipLaunchKernelGGL((bn_apply_kernel<true, false>), dim3(grid), dim3(256),
+                         0, s, xb, nullptr, yb, nullptr, scale, shift, M, C);
+  } else if (add) {
+    hipLaunchKernelGGL((bn_apply_kernel<false, true>), dim3(grid), dim3(256), 0,
+                       s, xb, rb, yb, nullptr, scale, shift, M, C);
+  } else {
+    hipLaunchKernelGGL((bn_apply_kernel<false, false>), dim3(grid), dim3(256), 0,
+                       s, xb, nullptr, yb, nullptr, scale, shift, M, C);
+  }
 }
 
-extern "C" void launch_bn_bwd_reduce(const void* dy, const void* y, const void* x,
+extern "C" void launch_bn_bwd_reduce(const void* dy, const unsigned char* mask,
+                                     const void* x,
                                      const float* mean, const float* invstd,
                                      float* partial, int grid, long long M, int C,
                                      bool relu, hipStream_t s) {
   if (relu)
     hipLaunchKernelGGL((bn_bwd_reduce_kernel<true>), dim3(grid), dim3(256), 0, s,
-                       (const bf16*)dy, (const bf16*)y, (const bf16*)x, mean, invstd,
+                       (const bf16*)dy, mask, (const bf16*)x, mean, invstd,
                        partial, M, C);
   else
     hipLaunchKernelGGL((bn_bwd_reduce_kernel<false>), dim3(grid), dim3(256), 0, s,
@@ -401,7 +424,8 @@ extern "C" void launch_bn_bwd_reduce(const void* dy, const void* y, const void* 
                        partial, M, C);
 }
 
-extern "C" void launch_bn_bwd_dx(const void* dy, const void* y, const void* x,
+extern "C" void launch_bn_bwd_dx(const void* dy, const unsigned char* mask,
+                                 const void* x,
                                  const float* mean, const float* invstd,
                                  const float* gamma, const float* sums, void* dx,
                                  void* dres, long long M, int C, bool relu,
@@ -410,7 +434,7 @@ extern "C" void launch_bn_bwd_dx(const void* dy, const void* y, const void* x,
   const int grid = elementwise_grid(total, 256);
 #define CASE(R, A, T)                                                          \
   hipLaunchKernelGGL((bn_bwd_dx_kernel<R, A, T>), dim3(grid), dim3(256), 0, s, \
-                     (const bf16*)dy, (const bf16*)y, (const bf16*)x, mean,    \
+                     (const bf16*)dy, mask, (const bf16*)x, mean,              \
                      invstd, gamma, sums, (bf16*)dx, (bf16*)dres, M, C)
   if (training) {
     if (relu && add) CASE(true, true, true);

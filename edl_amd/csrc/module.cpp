@@ -22,12 +22,13 @@ extern "C" void launch_bn_stats(const void*, float*, int, long long, int,
 extern "C" void launch_bn_finalize(const float*, int, const float*, const float*,
                                    float*, float*, float*, float*, float*, float*,
                                    float, float, long long, int, hipStream_t);
-extern "C" void launch_bn_apply(const void*, const void*, void*, const float*,
-                                const float*, long long, int, bool, bool,
-                                hipStream_t);
-extern "C" void launch_bn_bwd_reduce(const void*, const void*, const void*,
-                                     const float*, const float*, float*, int,
-                                     long long, int, bool, hipStream_t);
+extern "C" void launch_bn_apply(const void*, const void*, void*, unsigned char*,
+                                const float*, const float*, long long, int,
+                                bool, bool, hipStream_t);
+extern "C" void launch_bn_bwd_reduce(const void*, const unsigned char*,
+                                     const void*, const float*, const float*,
+                                     float*, int, long long, int, bool,
+                                     hipStream_t);
 extern "C" void launch_bn_bwd_finalize(const float*, int, float*, int, hipStream_t);
 extern "C" void launch_gemm_bt(const void*, const void*, void*, int, int, int,
                                hipStream_t);
@@ -47,10 +48,10 @@ extern "C" void launch_avgpool2x2_fwd(const void*, void*, int, int, int, int,
                                       int, int, hipStream_t);
 extern "C" void launch_avgpool2x2_bwd(const void*, void*, int, int, int, int,
                                       int, int, hipStream_t);
-extern "C" void launch_bn_bwd_dx(const void*, const void*, const void*, const float*,
-                                 const float*, const float*, const float*, void*,
-                                 void*, long long, int, bool, bool, bool,
-                                 hipStream_t);
+extern "C" void launch_bn_bwd_dx(const void*, const unsigned char*, const void*,
+                                 const float*, const float*, const float*,
+                                 const float*, void*, void*, long long, int,
+                                 bool, bool, bool, hipStream_t);
 
 namespace {
 
@@ -138,6 +139,9 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor gamma,
   auto scale = torch::empty({C}, opts);
   auto shift = torch::empty({C}, opts);
   auto y = torch::empty_like(x);
+  // 1-bit/channel ReLU mask for the backward (one byte per channel octet)
+  auto msk = relu ? torch::empty({M, C / 8}, x.options().dtype(torch::kUInt8))
+                  : torch::Tensor();
   auto s = cur_stream();
   launch_bn_stats(x.data_ptr(), partial.data_ptr<float>(), grid, M, (int)C, s);
   launch_bn_finalize(partial.data_ptr<float>(), grid, gamma.data_ptr<float>(),
@@ -148,9 +152,11 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor gamma,
                      rvar.defined() ? rvar.data_ptr<float>() : nullptr,
                      (float)momentum, (float)eps, M, (int)C, s);
   launch_bn_apply(x.data_ptr(), res.has_value() ? res->data_ptr() : nullptr,
-                  y.data_ptr(), scale.data_ptr<float>(), shift.data_ptr<float>(),
+                  y.data_ptr(),
+                  relu ? (unsigned char*)msk.data_ptr() : nullptr,
+                  scale.data_ptr<float>(), shift.data_ptr<float>(),
                   M, (int)C, relu, res.has_value(), s);
-  return {y, mean, invstd};
+  return {y, mean, invstd, msk};
 }
 
 torch::Tensor bn_fwd_eval(torch::Tensor x, torch::Tensor scale, torch::Tensor shift,
@@ -160,12 +166,14 @@ torch::Tensor bn_fwd_eval(torch::Tensor x, torch::Tensor scale, torch::Tensor sh
   const long long M = x.numel() / C;
   auto y = torch::empty_like(x);
   launch_bn_apply(x.data_ptr(), res.has_value() ? res->data_ptr() : nullptr,
-                  y.data_ptr(), scale.data_ptr<float>(), shift.data_ptr<float>(),
+                  y.data_ptr(), nullptr,
+                  scale.data_ptr<float>(), shift.data_ptr<float>(),
                   M, (int)C, relu, res.has_value(), cur_stream());
   return y;
 }
 
-std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y, torch::Tensor x,
+std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor msk,
+                                  torch::Tensor x,
                                   torch::Tensor mean, torch::Tensor invstd,
                                   torch::Tensor gamma, bool relu, bool add,
                                   bool training) {
@@ -180,12 +188,14 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y, torch::Tens
   auto dres = add ? torch::empty_like(x) : torch::Tensor();
   auto s = cur_stream();
   auto dyc = dy.is_contiguous() ? dy : dy.contiguous();
-  launch_bn_bwd_reduce(dyc.data_ptr(), relu ? y.data_ptr() : nullptr, x.data_ptr(),
+  const unsigned char* mp =
+      relu ? (const unsigned char*)msk.data_ptr() : nullptr;
+  launch_bn_bwd_reduce(dyc.data_ptr(), mp, x.data_ptr(),
                        mean.data_ptr<float>(), invstd.data_ptr<float>(),
                        partial.data_ptr<float>(), grid, M, (int)C, relu, s);
   launch_bn_bwd_finalize(partial.data_ptr<float>(), grid, sums.data_ptr<float>(),
                          (int)C, s);
-  launch_bn_bwd_dx(dyc.data_ptr(), relu ? y.data_ptr() : nullptr, x.data_ptr(),
+  launch_bn_bwd_dx(dyc.data_ptr(), mp, x.data_ptr(),
                    mean.data_ptr<float>(), invstd.data_ptr<float>(),
                    gamma.data_ptr<float>(), sums.data_ptr<float>(), dx.data_ptr(),
                    add ? dres.data_ptr() : nullptr, M, (int)C, relu, add, training,
@@ -360,7 +370,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("kd_ce_forward", &kd_ce_forward, "KD soft-label CE forward -> per-row loss");
   m.def("kd_ce_backward", &kd_ce_backward, "KD soft-label CE backward -> dlogits");
   m.def("bn_fwd_train", &bn_fwd_train,
-        "fused NHWC bf16 BN(+add)+ReLU train fwd -> (y, mean, invstd)");
+        "fused NHWC bf16 BN(+add)+ReLU train fwd -> (y, mean, invstd, mask)");
   m.def("bn_fwd_eval", &bn_fwd_eval, "fused NHWC bf16 BN(+add)+ReLU eval fwd");
   m.def("bn_bwd", &bn_bwd,
         "fused BN(+add)+ReLU bwd -> (dx, dgamma, dbeta, dres?)");
