@@ -141,7 +141,7 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor gamma,
   auto y = torch::empty_like(x);
   // 1-bit/channel ReLU mask for the backward (one byte per channel octet)
   auto msk = relu ? torch::empty({M, C / 8}, x.options().dtype(torch::kUInt8))
-                  : torch::Tensor();
+                  : torch::empty({0}, x.options().dtype(torch::kUInt8));
   auto s = cur_stream();
   launch_bn_stats(x.data_ptr(), partial.data_ptr<float>(), grid, M, (int)C, s);
   launch_bn_finalize(partial.data_ptr<float>(), grid, gamma.data_ptr<float>(),
@@ -172,7 +172,7 @@ torch::Tensor bn_fwd_eval(torch::Tensor x, torch::Tensor scale, torch::Tensor sh
   return y;
 }
 
-std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor msk,
+std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, c10::optional<torch::Tensor> msk,
                                   torch::Tensor x,
                                   torch::Tensor mean, torch::Tensor invstd,
                                   torch::Tensor gamma, bool relu, bool add,
@@ -188,8 +188,10 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor msk,
   auto dres = add ? torch::empty_like(x) : torch::Tensor();
   auto s = cur_stream();
   auto dyc = dy.is_contiguous() ? dy : dy.contiguous();
+  TORCH_CHECK(!relu || (msk.has_value() && msk->numel() == M * (C / 8)),
+              "bn_bwd: relu path needs the fwd mask");
   const unsigned char* mp =
-      relu ? (const unsigned char*)msk.data_ptr() : nullptr;
+      relu ? (const unsigned char*)msk->data_ptr() : nullptr;
   launch_bn_bwd_reduce(dyc.data_ptr(), mp, x.data_ptr(),
                        mean.data_ptr<float>(), invstd.data_ptr<float>(),
                        partial.data_ptr<float>(), grid, M, (int)C, relu, s);
