@@ -120,3 +120,41 @@ def test_unreachable_store_raises():
 def test_table_key_layout(coord_client):
     assert coord_client.table_key("resource", "p1") == "/test_job/resource/nodes/p1"
     assert coord_client.table_key("resource") == "/test_job/resource/nodes/"
+
+
+def test_concurrent_clients_stress(coord_server):
+    """16 threads x (leases + puts + CAS races + watches): the store must
+    stay consistent and exactly one CAS winner per key."""
+    import threading
+
+    winners = []
+    lock = threading.Lock()
+    errs = []
+
+    def worker(i):
+        try:
+            c = CoordClient(coord_server.endpoint, "stress")
+            lease = c.grant(5)
+            for j in range(30):
+                c.put("/stress/t%d/k%d" % (i, j), str(j), lease)
+            ok, _ = c.put_if_absent("/stress/winner", "t%d" % i)
+            if ok:
+                with lock:
+                    winners.append(i)
+            assert c.keepalive(lease)
+            kvs = c.range("/stress/t%d/" % i)
+            assert len(kvs) == 30
+            c.close()
+        except Exception as e:  # noqa: BLE001
+            errs.append(e)
+
+    threads = [threading.Thread(target=worker, args=(i,)) for i in range(16)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(30)
+    assert not errs, errs
+    assert len(winners) == 1
+    c = CoordClient(coord_server.endpoint, "stress")
+    assert len(c.range("/stress/")) == 16 * 30 + 1
+    c.close()
