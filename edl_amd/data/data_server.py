@@ -51,35 +51,35 @@ class PodsData:
             self._maybe_balance()
 
     def _maybe_balance(self):
-        live = [p for p in self.pod_ids if not self.finished[p]]
+        # "finished" = finished PRODUCING (data end); every pod still
+        # CONSUMES, so leveling always targets all pods. Gate: every pod
+        # has reported at least once (or already hit data end).
         if not all(self.reported[p] or self.finished[p] for p in self.pod_ids):
             return
         total = sum(len(q) for q in self.queues.values())
-        targets = live or self.pod_ids
+        if total == 0:
+            return
+        targets = self.pod_ids
         avg = total // len(targets)
         if avg == 0:
-            # too little to level: everyone keeps their own
+            # too little to level: hand out round-robin by current load
+            pool = [item for p in self.pod_ids for item in self.queues[p]]
             for p in self.pod_ids:
-                self.assignments[p].extend(self.queues[p])
                 self.queues[p] = []
+            order = sorted(targets, key=lambda p: len(self.assignments[p]))
+            for j, item in enumerate(pool):
+                self.assignments[order[j % len(order)]].append(item)
             return
-        # steal surplus above avg into a pool, then top up the deficit pods
+        # keep up to avg locally (locality), pool the surplus, top up the
+        # CURRENTLY least-loaded pod with each pooled item
         pool = []
         for p in self.pod_ids:
-            keep = avg if p in targets else 0
-            self.assignments[p].extend(self.queues[p][:keep])
-            pool.extend(self.queues[p][keep:])
+            self.assignments[p].extend(self.queues[p][:avg])
+            pool.extend(self.queues[p][avg:])
             self.queues[p] = []
-        i = 0
-        while pool:
-            p = targets[i % len(targets)]
-            if len(self.assignments[p]) <= avg:
-                self.assignments[p].append(pool.pop())
-            i += 1
-            if i > 10 * len(pool) + 100:  # all at/above avg: round-robin rest
-                for j, item in enumerate(pool):
-                    self.assignments[targets[j % len(targets)]].append(item)
-                pool = []
+        for item in pool:
+            p = min(targets, key=lambda q: len(self.assignments[q]))
+            self.assignments[p].append(item)
 
     def take_assignments(self, pod_id, max_n=64):
         with self.lock:

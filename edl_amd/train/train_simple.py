@@ -33,6 +33,11 @@ def main(argv=None):
     p.add_argument("--steps_per_epoch", type=int, default=100)
     p.add_argument("--lr", type=float, default=0.01)
     p.add_argument("--checkpoint", default=None)
+    p.add_argument("--data_dir", default=None,
+                   help="fit_a_line only: directory of text files "
+                        "('f1 ... f13 y' per line) read through the ELASTIC "
+                        "data plane (leader-balanced Reader) instead of "
+                        "synthetic data")
     args = p.parse_args(argv)
 
     tenv = TrainerEnv()
@@ -58,9 +63,84 @@ def main(argv=None):
             start_epoch = int(got[2]["epoch_no"]) + 1
     reducer.broadcast_params(0)
 
+    def data_plane_epoch():
+        """One pass over the shared file set via the elastic data plane:
+        trainer global rank 0 runs the leader DataServer; endpoints
+        rendezvous through the coordination store (dist_reader table)."""
+        import glob
+        import os as _os
+
+        from ..coord.client import CoordClient
+        from ..coord.tables import ETCD_DIST_READER
+        from ..data.data_server import DataServer
+        from ..data.reader import Reader
+
+        files = sorted(glob.glob(_os.path.join(args.data_dir, "*")))
+        pod_ids = [str(r) for r in range(tenv.world_size)]
+        me = str(tenv.global_rank)
+        srv = DataServer(file_list=files, pod_ids=pod_ids).start()
+        store = CoordClient(tenv.store_endpoints, tenv.job_id)
+        lease = store.grant(60)
+        store.put(store.table_key(ETCD_DIST_READER, me),
+                  "127.0.0.1:%d" % srv.port, lease)
+        eps = {}
+        deadline = time.monotonic() + 60
+        while len(eps) < tenv.world_size and time.monotonic() < deadline:
+            pfx = store.table_key(ETCD_DIST_READER)
+            eps = {k[len(pfx):]: v for k, v in store.range(pfx)}
+            time.sleep(0.1)
+        assert len(eps) == tenv.world_size, "data-plane rendezvous failed"
+        reader = Reader(me, eps["0"], srv, eps, batch_size=args.batch_size)
+        batch = []
+        for item in reader:
+            for rec in item["data"]:
+                vals = [float(v) for v in rec.split()]
+                batch.append(vals)
+                if len(batch) >= args.batch_size:
+                    yield torch.tensor(batch)
+                    batch = []
+        if batch:
+            yield torch.tensor(batch)
+        reader.close()
+        store.revoke(lease)
+        store.close()
+        srv.stop()
+
     g = torch.Generator().manual_seed(100 + tenv.global_rank)
     t0 = time.monotonic()
     for epoch in range(start_epoch, args.num_epochs):
+        if args.model == "fit_a_line" and args.data_dir:
+            # ranks may get UNEQUAL batch counts from the balancer: every
+            # step first agrees (all-reduce) whether anyone still has data;
+            # ranks without a batch contribute zero gradients (finalize
+            # all-reduces their zeroed buckets) so collectives stay matched
+            import torch.distributed as dist
+
+            n_recs = 0
+            it = iter(data_plane_epoch())
+            while True:
+                xy = next(it, None)
+                if dist.is_initialized():
+                    have = torch.tensor(
+                        [0.0 if xy is None else 1.0], device=device)
+                    dist.all_reduce(have)
+                    if have.item() == 0:
+                        break
+                elif xy is None:
+                    break
+                reducer.zero_grad()
+                if xy is not None:
+                    x, y = xy[:, :13].to(device), xy[:, 13:14].to(device)
+                    n_recs += x.shape[0]
+                    loss = F.mse_loss(model(x), y)
+                    loss.backward()
+                reducer.finalize()
+                opt.step()
+            log.info("rank %d consumed %d records this epoch",
+                     tenv.global_rank, n_recs)
+            if ckpt and tenv.is_rank0:
+                ckpt.save(model.state_dict(), {"epoch_no": epoch}, blocking=True)
+            continue
         for _ in range(args.steps_per_epoch):
             reducer.zero_grad()
             if args.model == "fit_a_line":

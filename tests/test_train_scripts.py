@@ -101,3 +101,36 @@ def test_resume_across_world_sizes(coord_server, tmp_path):
     assert (tmp_path / "ck" / "checkpoint.1").is_dir()
     with open(tmp_path / "ck" / "checkpoint.1" / "train_status.json") as f:
         assert json.load(f)["epoch_no"] == 1
+
+
+def test_fit_a_line_elastic_data_plane(coord_server, tmp_path):
+    """fit_a_line trained from FILES through the elastic data plane under
+    the launcher at world 2 (leader-balanced Reader; BASELINE config 1 with
+    real input plumbing)."""
+    data = tmp_path / "data"
+    data.mkdir()
+    import numpy as np
+
+    rng = np.random.RandomState(0)
+    for i in range(4):
+        rows = []
+        for _ in range(32):
+            x = rng.randn(13)
+            y = x.sum() * 0.5 + 1.0
+            rows.append(" ".join("%.5f" % v for v in list(x) + [y]))
+        (data / ("part%d.txt" % i)).write_text("\n".join(rows) + "\n")
+
+    run_edlrun(
+        coord_server, tmp_path,
+        ["-m", "edl_amd.train.train_simple", "--model", "fit_a_line",
+         "--num_epochs", "1", "--batch_size", "8",
+         "--data_dir", str(data), "--checkpoint", str(tmp_path / "ck")],
+    )
+    assert (tmp_path / "ck" / "checkpoint.0").is_dir()
+    # both ranks consumed a share of the 128 records
+    logs = "".join((tmp_path / ("agent%d.log" % i)).read_text() for i in (0, 1))
+    import re
+
+    counts = [int(m) for m in re.findall(r"consumed (\d+) records", logs)]
+    assert sum(counts) == 128, counts
+    assert all(c > 0 for c in counts), counts
