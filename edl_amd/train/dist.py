@@ -33,6 +33,12 @@ def init_from_env(env=None, timeout_s=120):
         device = torch.device("cpu")
         backend = "gloo"
 
+    # test hook: EDL_FORCE_BACKEND=gloo lets N ranks share one GPU (RCCL
+    # forbids duplicate devices) so the full multi-rank path is testable
+    # on a 1-GPU box
+    forced = os.environ.get("EDL_FORCE_BACKEND")
+    if forced:
+        backend = forced
     if tenv.world_size <= 1:
         return tenv, device  # single process: no communicator needed
     if not dist.is_initialized():
@@ -63,7 +69,8 @@ def rank():
 def barrier(device=None):
     if not dist.is_initialized():
         return
-    if device is not None and device.type == "cuda":
+    if (device is not None and device.type == "cuda"
+            and dist.get_backend() == "nccl"):
         dist.barrier(device_ids=[device.index])
     else:
         dist.barrier()
