@@ -53,22 +53,8 @@ __device__ __forceinline__ void store8(bf16* p, const F8& x) {
 // atomics at all (single-stage atomics measured 17-40 us/dispatch fixed
 // cost: same-word serialization at ~88 adds/us for small C, grid x 2C
 // traffic for large C).
-// The finalize is FUSED: each block publishes its partials (plain stores +
-// agent-scope release + ticket), and the LAST-arriving block acquires and
-// reduces them, computing mean/invstd/scale/shift + running stats — no
-// separate [C]-sized launch (was ~9 us launch+latency per BN). Recipe:
-// cdna_hip_programming.md §6 Guideline 16 (order: stores -> per-wave
-// vmcnt(0) -> __syncthreads -> lane0 release fence -> asm vmcnt(0) ->
-// relaxed ticket fetch_add; last block: acquire fence -> __syncthreads ->
-// plain loads). ticket: one zeroed u32 per call (module-cached; the last
-// block resets it for the next call).
 extern "C" __global__ void bn_stats_kernel(
     const bf16* __restrict__ x, float* __restrict__ partial,  // [grid, 2C]
-    unsigned* __restrict__ ticket, const float* __restrict__ gamma,
-    const float* __restrict__ beta, float* __restrict__ mean_out,
-    float* __restrict__ invstd_out, float* __restrict__ scale_out,
-    float* __restrict__ shift_out, float* __restrict__ running_mean,
-    float* __restrict__ running_var, const float momentum, const float eps,
     const long long M, const int C) {
   __shared__ float lsum[2 * 2048];
   const int c8 = C >> 3;  // channel-octet count
@@ -115,61 +101,57 @@ extern "C" __global__ void bn_stats_kernel(
   __syncthreads();
   float* out = partial + (long long)blockIdx.x * 2 * C;
   for (int i = threadIdx.x; i < 2 * C; i += blockDim.x) out[i] = lsum[i];
+}
 
-  // ---- last-arriver finalize ----
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // every storing wave
-  __syncthreads();
-  if (threadIdx.x == 0) {
-    __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    const unsigned t = __hip_atomic_fetch_add(ticket, 1u, __ATOMIC_RELAXED,
-                                              __HIP_MEMORY_SCOPE_AGENT);
-    lsum[0] = (t == gridDim.x - 1) ? 1.0f : 0.0f;  // reuse the ONE shared array
-  }
-  __syncthreads();
-  if (lsum[0] == 0.0f) return;  // not the last block
-  if (threadIdx.x == 0) {
-    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
-    __hip_atomic_store(ticket, 0u, __ATOMIC_RELAXED,
-                       __HIP_MEMORY_SCOPE_AGENT);  // ready for the next call
-  }
-  __syncthreads();
-  const int nblocks = (int)gridDim.x;
-  const float inv_m = 1.0f / (float)M;
-  for (int c = threadIdx.x; c < C; c += blockDim.x) {
-    float sa[8] = {0}, qa[8] = {0};
-    const long long st = 2 * C;
-    int b = 0;
-    for (; b + 8 <= nblocks; b += 8) {
-#pragma unroll
-      for (int u = 0; u < 8; ++u) {
-        sa[u] += partial[(long long)(b + u) * st + c];
-        qa[u] += partial[(long long)(b + u) * st + C + c];
-      }
-    }
-    for (; b < nblocks; ++b) {
-      sa[0] += partial[(long long)b * st + c];
-      qa[0] += partial[(long long)b * st + C + c];
-    }
-    float sm = 0.0f, qq = 0.0f;
+// K2: reduce partials over blocks, finalize mean/invstd, update running
+// stats. One thread per channel; thread c's reads of partial[b][c] are
+// coalesced across the warp for each fixed b.
+extern "C" __global__ void bn_finalize_kernel(
+    const float* __restrict__ partial, const int nblocks,
+    const float* __restrict__ gamma,
+    const float* __restrict__ beta, float* __restrict__ mean_out,
+    float* __restrict__ invstd_out, float* __restrict__ scale_out,
+    float* __restrict__ shift_out, float* __restrict__ running_mean,
+    float* __restrict__ running_var, const float momentum, const float eps,
+    const long long M, const int C) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  // 8 independent accumulators: a single s+= chain leaves every load
+  // latency-exposed (measured 125 us at nblocks=1024 vs ~5 us unrolled)
+  float sa[8] = {0}, qa[8] = {0};
+  const long long st = 2 * C;
+  int b = 0;
+  for (; b + 8 <= nblocks; b += 8) {
 #pragma unroll
     for (int u = 0; u < 8; ++u) {
-      sm += sa[u];
-      qq += qa[u];
+      sa[u] += partial[(long long)(b + u) * st + c];
+      qa[u] += partial[(long long)(b + u) * st + C + c];
     }
-    const float mu = sm * inv_m;
-    const float var = fmaxf(qq * inv_m - mu * mu, 0.0f);
-    const float is = rsqrtf(var + eps);
-    mean_out[c] = mu;
-    invstd_out[c] = is;
-    const float g = gamma[c];
-    scale_out[c] = g * is;
-    shift_out[c] = beta[c] - g * is * mu;
-    if (running_mean != nullptr) {
-      const float ub = (M > 1) ? var * (float)M / (float)(M - 1) : var;
-      running_mean[c] = fmaf(momentum, mu - running_mean[c], running_mean[c]);
-      running_var[c] = fmaf(momentum, ub - running_var[c], running_var[c]);
-    }
+  }
+  for (; b < nblocks; ++b) {
+    sa[0] += partial[(long long)b * st + c];
+    qa[0] += partial[(long long)b * st + C + c];
+  }
+  float s = 0.0f, q = 0.0f;
+#pragma unroll
+  for (int u = 0; u < 8; ++u) {
+    s += sa[u];
+    q += qa[u];
+  }
+  const float inv_m = 1.0f / (float)M;
+  const float mean = s * inv_m;
+  const float var = fmaxf(q * inv_m - mean * mean, 0.0f);
+  const float invstd = rsqrtf(var + eps);
+  mean_out[c] = mean;
+  invstd_out[c] = invstd;
+  const float g = gamma[c];
+  scale_out[c] = g * invstd;
+  shift_out[c] = beta[c] - g * invstd * mean;
+  if (running_mean != nullptr) {
+    // unbiased variance for the running estimate (torch semantics)
+    const float ub = (M > 1) ? var * (float)M / (float)(M - 1) : var;
+    running_mean[c] = fmaf(momentum, mean - running_mean[c], running_mean[c]);
+    running_var[c] = fmaf(momentum, ub - running_var[c], running_var[c]);
   }
 }
 
@@ -219,8 +201,7 @@ template <bool RELU>
 __global__ void bn_bwd_reduce_kernel(
     const bf16* __restrict__ dy, const unsigned char* __restrict__ mask,
     const bf16* __restrict__ x, const float* __restrict__ mean,
-    const float* __restrict__ invstd, float* __restrict__ partial,
-    unsigned* __restrict__ ticket, float* __restrict__ sums,  // out [2, C]
+    const float* __restrict__ invstd, float* __restrict__ sums,  // [2, C]
     const long long M, const int C) {
   __shared__ float lsum[2 * 2048];
   const int c8 = C >> 3;
@@ -288,40 +269,26 @@ __global__ void bn_bwd_reduce_kernel(
     atomicAdd(&lsum[C + c0 + i], s2[i]);
   }
   __syncthreads();
-  float* out = partial + (long long)blockIdx.x * 2 * C;  // [grid, 2C]
+  float* out = sums + (long long)blockIdx.x * 2 * C;  // [grid, 2C] partials
   for (int i = threadIdx.x; i < 2 * C; i += blockDim.x) out[i] = lsum[i];
+}
 
-  // ---- last-arriver reduce -> sums[2C] (= [dbeta; dgamma]) ----
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-  __syncthreads();
-  if (threadIdx.x == 0) {
-    __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    const unsigned t = __hip_atomic_fetch_add(ticket, 1u, __ATOMIC_RELAXED,
-                                              __HIP_MEMORY_SCOPE_AGENT);
-    lsum[0] = (t == gridDim.x - 1) ? 1.0f : 0.0f;
-  }
-  __syncthreads();
-  if (lsum[0] == 0.0f) return;
-  if (threadIdx.x == 0) {
-    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
-    __hip_atomic_store(ticket, 0u, __ATOMIC_RELAXED,
-                       __HIP_MEMORY_SCOPE_AGENT);
-  }
-  __syncthreads();
-  const int nblocks = (int)gridDim.x;
+// B1b: reduce bwd partials over blocks -> sums[2C] (= [dbeta; dgamma]).
+extern "C" __global__ void bn_bwd_finalize_kernel(
+    const float* __restrict__ partial, const int nblocks,
+    float* __restrict__ sums, const int C) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= 2 * C) return;
   const long long st = 2 * C;
-  for (int i = threadIdx.x; i < 2 * C; i += blockDim.x) {
-    float sa[8] = {0};
-    int b = 0;
-    for (; b + 8 <= nblocks; b += 8) {
+  float sa[8] = {0};
+  int b = 0;
+  for (; b + 8 <= nblocks; b += 8) {
 #pragma unroll
-      for (int u = 0; u < 8; ++u) sa[u] += partial[(long long)(b + u) * st + i];
-    }
-    for (; b < nblocks; ++b) sa[0] += partial[(long long)b * st + i];
-    sums[i] = ((sa[0] + sa[1]) + (sa[2] + sa[3])) +
-              ((sa[4] + sa[5]) + (sa[6] + sa[7]));
+    for (int u = 0; u < 8; ++u) sa[u] += partial[(long long)(b + u) * st + i];
   }
+  for (; b < nblocks; ++b) sa[0] += partial[(long long)b * st + i];
+  sums[i] = ((sa[0] + sa[1]) + (sa[2] + sa[3])) +
+            ((sa[4] + sa[5]) + (sa[6] + sa[7]));
 }
 
 // (no separate dgamma/dbeta kernel: the bwd-reduce workspace IS [dbeta; dgamma]
@@ -386,15 +353,27 @@ extern "C" int bn_stats_grid(long long M, int C) {
   return (int)(g > 0 ? g : 1);
 }
 
-extern "C" void launch_bn_stats(const void* x, float* partial, unsigned* ticket,
-                                const float* gamma, const float* beta,
-                                float* mean, float* invstd, float* scale,
-                                float* shift, float* rmean, float* rvar,
-                                float momentum, float eps, int grid,
+extern "C" void launch_bn_stats(const void* x, float* partial, int grid,
                                 long long M, int C, hipStream_t s) {
   hipLaunchKernelGGL(bn_stats_kernel, dim3(grid), dim3(256), 0, s,
-                     (const bf16*)x, partial, ticket, gamma, beta, mean, invstd,
-                     scale, shift, rmean, rvar, momentum, eps, M, C);
+                     (const bf16*)x, partial, M, C);
+}
+
+extern "C" void launch_bn_finalize(const float* partial, int nblocks,
+                                   const float* gamma,
+                                   const float* beta, float* mean, float* invstd,
+                                   float* scale, float* shift, float* rmean,
+                                   float* rvar, float momentum, float eps,
+                                   long long M, int C, hipStream_t s) {
+  hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 255) / 256), dim3(256), 0, s,
+                     partial, nblocks, gamma, beta, mean, invstd, scale, shift,
+                     rmean, rvar, momentum, eps, M, C);
+}
+
+extern "C" void launch_bn_bwd_finalize(const float* partial, int nblocks,
+                                       float* sums, int C, hipStream_t s) {
+  hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((2 * C + 255) / 256), dim3(256),
+                     0, s, partial, nblocks, sums, C);
 }
 
 extern "C" void launch_bn_apply(const void* x, const void* res, void* y,
@@ -433,17 +412,16 @@ extern "C" void launch_bn_apply(const void* x, const void* res, void* y,
 extern "C" void launch_bn_bwd_reduce(const void* dy, const unsigned char* mask,
                                      const void* x,
                                      const float* mean, const float* invstd,
-                                     float* partial, unsigned* ticket,
-                                     float* sums, int grid, long long M, int C,
+                                     float* partial, int grid, long long M, int C,
                                      bool relu, hipStream_t s) {
   if (relu)
     hipLaunchKernelGGL((bn_bwd_reduce_kernel<true>), dim3(grid), dim3(256), 0, s,
                        (const bf16*)dy, mask, (const bf16*)x, mean, invstd,
-                       partial, ticket, sums, M, C);
+                       partial, M, C);
   else
     hipLaunchKernelGGL((bn_bwd_reduce_kernel<false>), dim3(grid), dim3(256), 0, s,
                        (const bf16*)dy, nullptr, (const bf16*)x, mean, invstd,
-                       partial, ticket, sums, M, C);
+                       partial, M, C);
 }
 
 extern "C" void launch_bn_bwd_dx(const void* dy, const unsigned char* mask,

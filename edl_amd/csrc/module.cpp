@@ -17,17 +17,19 @@ extern "C" void launch_kd_ce_fwd_bf16(const void*, const void*, float*, int, int
 extern "C" void launch_kd_ce_bwd_bf16(const void*, const void*, void*, float, int,
                                       int, hipStream_t);
 extern "C" int bn_stats_grid(long long, int);
-extern "C" void launch_bn_stats(const void*, float*, unsigned*, const float*,
-                                const float*, float*, float*, float*, float*,
-                                float*, float*, float, float, int, long long,
-                                int, hipStream_t);
+extern "C" void launch_bn_stats(const void*, float*, int, long long, int,
+                                hipStream_t);
+extern "C" void launch_bn_finalize(const float*, int, const float*, const float*,
+                                   float*, float*, float*, float*, float*, float*,
+                                   float, float, long long, int, hipStream_t);
 extern "C" void launch_bn_apply(const void*, const void*, void*, unsigned char*,
                                 const float*, const float*, long long, int,
                                 bool, bool, hipStream_t);
 extern "C" void launch_bn_bwd_reduce(const void*, const unsigned char*,
                                      const void*, const float*, const float*,
-                                     float*, unsigned*, float*, int, long long,
-                                     int, bool, hipStream_t);
+                                     float*, int, long long, int, bool,
+                                     hipStream_t);
+extern "C" void launch_bn_bwd_finalize(const float*, int, float*, int, hipStream_t);
 extern "C" void launch_gemm_bt(const void*, const void*, void*, int, int, int,
                                hipStream_t);
 extern "C" void launch_pad_nhwc(const void*, void*, int, int, int, int, int, int,
@@ -122,7 +124,7 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor gamma,
                                         torch::Tensor rvar, double momentum,
                                         double eps,
                                         c10::optional<torch::Tensor> res,
-                                        bool relu, torch::Tensor ticket) {
+                                        bool relu) {
   // x: [M, C] contiguous view of an NHWC tensor (Python side reshapes)
   const int64_t C = gamma.numel();
   TORCH_CHECK(x.dim() == 2 && x.size(1) == C && x.is_contiguous(),
@@ -141,17 +143,14 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor gamma,
   auto msk = relu ? torch::empty({M, C / 8}, x.options().dtype(torch::kUInt8))
                   : torch::empty({0}, x.options().dtype(torch::kUInt8));
   auto s = cur_stream();
-  TORCH_CHECK(ticket.is_cuda() && ticket.scalar_type() == torch::kInt32 &&
-                  ticket.numel() >= 1,
-              "bn: ticket must be a zeroed int32 GPU scalar");
-  launch_bn_stats(x.data_ptr(), partial.data_ptr<float>(),
-                  (unsigned*)ticket.data_ptr(), gamma.data_ptr<float>(),
-                  beta.data_ptr<float>(), mean.data_ptr<float>(),
-                  invstd.data_ptr<float>(), scale.data_ptr<float>(),
-                  shift.data_ptr<float>(),
-                  rmean.defined() ? rmean.data_ptr<float>() : nullptr,
-                  rvar.defined() ? rvar.data_ptr<float>() : nullptr,
-                  (float)momentum, (float)eps, grid, M, (int)C, s);
+  launch_bn_stats(x.data_ptr(), partial.data_ptr<float>(), grid, M, (int)C, s);
+  launch_bn_finalize(partial.data_ptr<float>(), grid, gamma.data_ptr<float>(),
+                     beta.data_ptr<float>(), mean.data_ptr<float>(),
+                     invstd.data_ptr<float>(), scale.data_ptr<float>(),
+                     shift.data_ptr<float>(),
+                     rmean.defined() ? rmean.data_ptr<float>() : nullptr,
+                     rvar.defined() ? rvar.data_ptr<float>() : nullptr,
+                     (float)momentum, (float)eps, M, (int)C, s);
   launch_bn_apply(x.data_ptr(), res.has_value() ? res->data_ptr() : nullptr,
                   y.data_ptr(),
                   relu ? (unsigned char*)msk.data_ptr() : nullptr,
@@ -177,7 +176,7 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, c10::optional<torch::Tensor>
                                   torch::Tensor x,
                                   torch::Tensor mean, torch::Tensor invstd,
                                   torch::Tensor gamma, bool relu, bool add,
-                                  bool training, torch::Tensor ticket) {
+                                  bool training) {
   const int64_t C = gamma.numel();
   check_bn_inputs(x, C);
   const long long M = x.numel() / C;
@@ -195,9 +194,9 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, c10::optional<torch::Tensor>
       relu ? (const unsigned char*)msk->data_ptr() : nullptr;
   launch_bn_bwd_reduce(dyc.data_ptr(), mp, x.data_ptr(),
                        mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                       partial.data_ptr<float>(),
-                       (unsigned*)ticket.data_ptr(), sums.data_ptr<float>(),
-                       grid, M, (int)C, relu, s);
+                       partial.data_ptr<float>(), grid, M, (int)C, relu, s);
+  launch_bn_bwd_finalize(partial.data_ptr<float>(), grid, sums.data_ptr<float>(),
+                         (int)C, s);
   launch_bn_bwd_dx(dyc.data_ptr(), mp, x.data_ptr(),
                    mean.data_ptr<float>(), invstd.data_ptr<float>(),
                    gamma.data_ptr<float>(), sums.data_ptr<float>(), dx.data_ptr(),

@@ -17,27 +17,24 @@ from . import available, ext
 
 class _FusedBN(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x2d, gamma, beta, rmean, rvar, momentum, eps, res2d, relu,
-                ticket):
+    def forward(ctx, x2d, gamma, beta, rmean, rvar, momentum, eps, res2d, relu):
         y, mean, invstd, mask = ext().bn_fwd_train(
-            x2d, gamma, beta, rmean, rvar, momentum, eps, res2d, relu, ticket
+            x2d, gamma, beta, rmean, rvar, momentum, eps, res2d, relu
         )
         # backward reads the 1-bit ReLU mask instead of y (16x fewer bytes)
         ctx.save_for_backward(x2d, mask, mean, invstd, gamma)
         ctx.relu = relu
         ctx.has_res = res2d is not None
-        ctx.ticket = ticket
         return y
 
     @staticmethod
     def backward(ctx, dy):
         x2d, mask, mean, invstd, gamma = ctx.saved_tensors
         dx, dgamma, dbeta, dres = ext().bn_bwd(
-            dy, mask, x2d, mean, invstd, gamma, ctx.relu, ctx.has_res, True,
-            ctx.ticket
+            dy, mask, x2d, mean, invstd, gamma, ctx.relu, ctx.has_res, True
         )
         return (dx, dgamma, dbeta, None, None, None, None,
-                dres if ctx.has_res else None, None, None)
+                dres if ctx.has_res else None, None)
 
 
 def _to_2d(t):
@@ -95,22 +92,13 @@ class BNReLU2d(nn.Module):
             y = F.relu(y)
         return y.to(x.dtype)
 
-    def _ticket(self, device):
-        # one zeroed u32 per module: the last-arriver finalize resets it
-        # inside the kernel, so it is allocated (and zeroed) exactly once
-        t = getattr(self, "_ticket_buf", None)
-        if t is None or t.device != device:
-            t = torch.zeros(1, dtype=torch.int32, device=device)
-            self._ticket_buf = t
-        return t
-
     def _fused(self, x, res=None):
         x2d = _to_2d(x)
         res2d = _to_2d(res) if res is not None else None
         if self.training:
             y2d = _FusedBN.apply(
                 x2d, self.weight, self.bias, self.running_mean, self.running_var,
-                self.momentum, self.eps, res2d, self.act, self._ticket(x.device),
+                self.momentum, self.eps, res2d, self.act,
             )
         else:
             invstd = torch.rsqrt(self.running_var + self.eps)
