@@ -107,6 +107,22 @@ class Launcher:
             self._watcher.stop()
             self._watcher = None
 
+    def _await_cluster_change(self, factor=3.0):
+        """After a trainer failure, wait up to ~factor lease-TTLs for the
+        cluster watcher to report a membership change (a peer death that
+        broke the communicator). True -> resize; False -> genuine local
+        failure."""
+        from .coord.tables import ETCD_TTL
+
+        deadline = time.monotonic() + factor * ETCD_TTL + 1.0
+        while time.monotonic() < deadline:
+            if self._watcher.changed:
+                return True
+            if self._resource.failed or self._elector.lost:
+                return False
+            time.sleep(0.2)
+        return self._watcher.changed
+
     def _run(self):
         cluster = barrier(self._client, self._pod.pod_id, timeout=600, allow_join=True)
         log.info("initial barrier done: stage=%s world=%d",
@@ -119,13 +135,21 @@ class Launcher:
             self._procs.tail_rank0()
             alive, failed = self._procs.poll()
             if not alive:
-                if failed:
-                    log.error("trainer process failed; pod exits FAILED")
-                    self.final_status = Status.FAILED
-                else:
+                if not failed:
                     log.info("all trainers exited cleanly")
                     self.final_status = Status.SUCCEED
-                return
+                    return
+                # A trainer died. If a PEER pod just died, our trainer's
+                # collective fails near-instantly ("connection closed by
+                # peer") while the dead pod's lease takes up to one TTL to
+                # lapse — so before declaring the failure OURS, give the
+                # failure detector a grace window to publish the membership
+                # change; if the cluster changed, this is a resize, not a
+                # local fault (stop-resume below).
+                if not self._await_cluster_change():
+                    log.error("trainer process failed; pod exits FAILED")
+                    self.final_status = Status.FAILED
+                    return
 
             if self._resource.failed or self._elector.lost:
                 log.error("lost store lease/leadership; stopping trainers")

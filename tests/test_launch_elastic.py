@@ -145,6 +145,47 @@ def test_elastic_scale_in(coord_server, tmp_path, agent_reaper):
     c.close()
 
 
+def test_peer_death_crashes_trainer_then_resize(coord_server, tmp_path,
+                                                agent_reaper):
+    """Race regression (launch.py _await_cluster_change): when a peer pod
+    dies, the survivor's trainer crashes near-instantly (broken collective)
+    while the dead pod's lease takes up to one TTL to lapse. Simulate by
+    SIGKILLing agent B's tree AND both world-2 trainer pids in the same
+    instant: agent A must treat its trainer's death as a resize (grace
+    window sees the membership change) and finish at world 1 — not exit
+    FAILED."""
+    job = "job_peercrash"
+    a = spawn_agent(coord_server.endpoint, job, tmp_path, "a",
+                    extra_env={"EDL_DEMO_SLEEP": "10"})
+    b = spawn_agent(coord_server.endpoint, job, tmp_path, "b",
+                    extra_env={"EDL_DEMO_SLEEP": "10"})
+    agent_reaper.extend([a, b])
+    deadline = time.monotonic() + 30
+    while time.monotonic() < deadline:
+        if len([r for r in read_runs(tmp_path) if r["world"] == 2]) >= 2:
+            break
+        time.sleep(0.2)
+    else:
+        pytest.fail("both trainers did not start: %s" % read_runs(tmp_path))
+
+    kill_tree(b)
+    # trainers live in their OWN sessions (procs.py start_new_session), so
+    # kill them by the exact pids they recorded — this mimics the gloo/RCCL
+    # "connection closed by peer" crash hitting BOTH ranks at once
+    for r in read_runs(tmp_path):
+        if r["world"] == 2:
+            try:
+                os.kill(r["pid"], signal.SIGKILL)
+            except ProcessLookupError:
+                pass
+    assert a.wait(timeout=90) == 0, (tmp_path / "agent_a.log").read_text()
+    runs = read_runs(tmp_path)
+    assert any(r["world"] == 1 for r in runs), runs
+    c = CoordClient(coord_server.endpoint, job)
+    assert load_job_status(c) == Status.SUCCEED
+    c.close()
+
+
 def test_elastic_scale_out(coord_server, tmp_path, agent_reaper):
     """Start 1 agent with range 1:2 (trainer sleeps), add a second agent;
     the generator must append it (stage bump) and both finish at world 2."""

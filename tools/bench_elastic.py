@@ -26,7 +26,8 @@ sys.path.insert(0, REPO)
 
 from edl_amd.coord.server import CoordServer  # noqa: E402
 
-MARKER = os.path.join(tempfile.gettempdir(), "edl_elastic_steps.jsonl")
+MARKER = os.path.join(tempfile.gettempdir(),
+                      "edl_elastic_steps.%d.jsonl" % os.getpid())
 
 
 def spawn_agent(store_ep, job_id, idx, nodes_range, log_dir, gpu=None):
@@ -124,6 +125,13 @@ def main():
             try:
                 os.killpg(os.getpgid(p.pid), signal.SIGKILL)
             except ProcessLookupError:
+                pass
+        # trainer workers live in their OWN process groups (procs.py uses
+        # start_new_session): kill the exact pids they reported
+        for pid in {s["pid"] for s in read_steps() if "pid" in s}:
+            try:
+                os.kill(pid, signal.SIGKILL)
+            except (ProcessLookupError, PermissionError):
                 pass
         srv.stop()
 

@@ -20,7 +20,13 @@ def main():
     marker = os.environ.get("EDL_STEP_MARKER")
     t = torch.ones(1 << 18, device=device)
     step = 0
+    t_start = time.time()
     while True:
+        # self-terminate when orphaned (our launcher agent died) or after a
+        # hard TTL: trainers run in their OWN process group, so killing the
+        # agent does not reach them — a leaked worker once ran for hours
+        if os.getppid() == 1 or time.time() - t_start > 600:
+            return
         if dist.is_initialized():
             dist.all_reduce(t)
         if device.type == "cuda":
@@ -29,7 +35,7 @@ def main():
         if marker:
             with open(marker, "a") as f:
                 f.write(json.dumps({"rank": tenv.global_rank,
-                                    "world": tenv.world_size,
+                                    "world": tenv.world_size, "pid": os.getpid(),
                                     "step": step, "ts": time.time()}) + "\n")
         time.sleep(0.05)
 
