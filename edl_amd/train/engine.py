@@ -23,6 +23,27 @@ from .checkpoint import CheckpointManager
 
 log = get_logger("edl.engine")
 
+# EDL_PROFILE_MARKERS=1 emits roctx ranges (torch.cuda.nvtx IS roctx on
+# ROCm) around the step phases so `rocprofv3 --marker-trace` can attribute
+# kernels to forward/backward/allreduce/optimizer (SURVEY.md §5.1 — the
+# reference's env-gated _TimeLine, but rocprof-native)
+_MARKERS = os.environ.get("EDL_PROFILE_MARKERS") == "1"
+
+
+class _mark:
+    __slots__ = ("name",)
+
+    def __init__(self, name):
+        self.name = name
+
+    def __enter__(self):
+        if _MARKERS:
+            torch.cuda.nvtx.range_push(self.name)
+
+    def __exit__(self, *exc):
+        if _MARKERS:
+            torch.cuda.nvtx.range_pop()
+
 
 def piecewise_lr(base_lr, epoch, boundaries=(30, 60, 90), decay=0.1, warmup_epochs=5,
                  step_in_epoch=0.0):
@@ -232,21 +253,25 @@ class TrainerEngine:
 
     def train_step(self, images, labels, teacher_logits=None):
         self.reducer.zero_grad()
-        with torch.autocast(device_type=self.device.type, dtype=self.dtype,
-                            enabled=self.dtype != torch.float32):
-            logits = self.model(images)
-        loss = self._loss(logits, labels, teacher_logits)
-        loss.backward()
-        if self.dgc is not None:
-            self.dgc.step()
-        else:
-            self.reducer.finalize()
-        if not getattr(self.opt, "handles_grad_scale", False):
-            scale = self.reducer.grad_scale
-            if scale != 1.0:
-                for b in self.reducer._buckets:
-                    b.buffer.mul_(scale)
-        self.opt.step()
+        with _mark("edl.forward"):
+            with torch.autocast(device_type=self.device.type, dtype=self.dtype,
+                                enabled=self.dtype != torch.float32):
+                logits = self.model(images)
+            loss = self._loss(logits, labels, teacher_logits)
+        with _mark("edl.backward"):
+            loss.backward()
+        with _mark("edl.allreduce_finalize"):
+            if self.dgc is not None:
+                self.dgc.step()
+            else:
+                self.reducer.finalize()
+        with _mark("edl.optimizer"):
+            if not getattr(self.opt, "handles_grad_scale", False):
+                scale = self.reducer.grad_scale
+                if scale != 1.0:
+                    for b in self.reducer._buckets:
+                        b.buffer.mul_(scale)
+            self.opt.step()
         self.global_step += 1
         return loss
 
