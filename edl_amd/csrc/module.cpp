@@ -44,6 +44,11 @@ extern "C" void launch_gemm_bt_splitk(const void*, const void*, float*, int, int
                                       int, int, hipStream_t);
 extern "C" void launch_shift9_transpose(const void*, void*, int, int, int, int,
                                         int, int, int, int, hipStream_t);
+extern "C" void launch_gemm_tn_splitk(const void*, const void*, float*, int,
+                                      int, int, int, hipStream_t);
+extern "C" void launch_gemm_tn3x3_splitk(const void*, const void*, float*, int,
+                                         int, int, int, int, int, int, int,
+                                         int, hipStream_t);
 extern "C" void launch_avgpool2x2_fwd(const void*, void*, int, int, int, int,
                                       int, int, hipStream_t);
 extern "C" void launch_avgpool2x2_bwd(const void*, void*, int, int, int, int,
@@ -326,6 +331,70 @@ torch::Tensor gemm_bt_splitk(torch::Tensor a, torch::Tensor b, int64_t splitk) {
   return c;
 }
 
+torch::Tensor gemm_tn_splitk(torch::Tensor a, torch::Tensor b, int64_t splitk) {
+  // fp32 C[N1,N2] = A[K,N1]^T @ B[K,N2] — conv wgrad WITHOUT the
+  // transpose_pad materializations (operands in native activation layout)
+  TORCH_CHECK(a.is_cuda() && b.is_cuda() && a.dim() == 2 && b.dim() == 2 &&
+                  a.size(0) == b.size(0),
+              "gemm_tn_splitk: [K,N1] x [K,N2] GPU");
+  TORCH_CHECK(a.scalar_type() == torch::kBFloat16 &&
+                  b.scalar_type() == torch::kBFloat16,
+              "gemm_tn_splitk: bf16 only");
+  auto ac = a.contiguous();
+  auto bc = b.contiguous();
+  const int K = (int)a.size(0), N1 = (int)a.size(1), N2 = (int)b.size(1);
+  TORCH_CHECK(N1 % 64 == 0 && N2 % 64 == 0, "gemm_tn_splitk: N1,N2 % 64");
+  const int nchunks = (K + 63) / 64;
+  if (splitk <= 0) {
+    const int tiles = (N1 / (N1 % 128 == 0 ? 128 : 64)) *
+                      (N2 / (N2 % 128 == 0 ? 128 : 64));
+    splitk = std::max<int64_t>(1, 512 / std::max(1, tiles));
+  }
+  splitk = std::min<int64_t>(splitk, nchunks);
+  auto c = torch::zeros({N1, N2}, a.options().dtype(torch::kFloat32));
+  launch_gemm_tn_splitk(ac.data_ptr(), bc.data_ptr(), c.data_ptr<float>(), N1,
+                        N2, K, (int)splitk, cur_stream());
+  return c;
+}
+
+torch::Tensor gemm_tn3x3_splitk(torch::Tensor dy2d, torch::Tensor x,
+                                int64_t stride, int64_t splitk) {
+  // conv3x3 wgrad, fully direct: fp32 dW3[Cout, 9*Cin] = dY^T @
+  // gather3x3(pad(x)) — no transpose_pad / shift9 materializations.
+  // dy2d: [M, Cout] bf16 (NHWC-flattened dY); x: 4-D channels_last bf16.
+  TORCH_CHECK(dy2d.is_cuda() && dy2d.dim() == 2 &&
+                  dy2d.scalar_type() == torch::kBFloat16,
+              "gemm_tn3x3: dy2d [M,Cout] bf16");
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 &&
+                  x.scalar_type() == torch::kBFloat16 &&
+                  x.is_contiguous(torch::MemoryFormat::ChannelsLast),
+              "gemm_tn3x3: x 4-D channels_last bf16");
+  const int Nimg = (int)x.size(0), Cin = (int)x.size(1);
+  const int H = (int)x.size(2), W = (int)x.size(3);
+  const int Hp = H + 2, Wp = W + 2;
+  const int Ho = (H - 1) / (int)stride + 1, Wo = (W - 1) / (int)stride + 1;
+  const int M = Nimg * Ho * Wo;
+  const int Cout = (int)dy2d.size(1);
+  TORCH_CHECK((int)dy2d.size(0) == M, "gemm_tn3x3: dy2d rows != N*Ho*Wo");
+  TORCH_CHECK(Cin % 64 == 0 && Cout % 64 == 0, "gemm_tn3x3: C % 64");
+  auto dyc = dy2d.contiguous();
+  auto s = cur_stream();
+  auto xp = torch::empty({(long long)Nimg * Hp * Wp * Cin}, x.options());
+  launch_pad_nhwc(x.data_ptr(), xp.data_ptr(), Nimg, H, W, Hp, Wp, Cin, s);
+  const int nchunks = (M + 63) / 64;
+  if (splitk <= 0) {
+    const int tiles = (Cout / (Cout % 128 == 0 ? 128 : 64)) *
+                      (9 * Cin / (Cin % 128 == 0 ? 128 : 64));
+    splitk = std::max<int64_t>(1, 512 / std::max(1, tiles));
+  }
+  splitk = std::min<int64_t>(splitk, nchunks);
+  auto c = torch::zeros({Cout, 9 * Cin}, x.options().dtype(torch::kFloat32));
+  launch_gemm_tn3x3_splitk(dyc.data_ptr(), xp.data_ptr(), c.data_ptr<float>(),
+                           Cout, Cin, M, Ho, Wo, Hp, Wp, (int)stride,
+                           (int)splitk, s);
+  return c;
+}
+
 torch::Tensor transpose_pad(torch::Tensor x) {
   TORCH_CHECK(x.is_cuda() && x.dim() == 2 &&
                   x.scalar_type() == torch::kBFloat16 && x.is_contiguous(),
@@ -384,6 +453,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_bt_splitk", &gemm_bt_splitk,
         "split-K bt GEMM -> fp32 C (wgrad shapes)", pybind11::arg("a"),
         pybind11::arg("b"), pybind11::arg("splitk") = 0);
+  m.def("gemm_tn3x3_splitk", &gemm_tn3x3_splitk,
+        "direct conv3x3 wgrad: fp32 dW3[Cout,9Cin] = dY^T @ gather3x3(pad(x))",
+        pybind11::arg("dy2d"), pybind11::arg("x"), pybind11::arg("stride"),
+        pybind11::arg("splitk") = 0);
+  m.def("gemm_tn_splitk", &gemm_tn_splitk,
+        "split-K TN GEMM: fp32 C[N1,N2] = A[K,N1]^T @ B[K,N2] (direct wgrad)",
+        pybind11::arg("a"), pybind11::arg("b"), pybind11::arg("splitk") = 0);
   m.def("conv3x3_wgrad_operand", &conv3x3_wgrad_operand,
         "padded shifted transpose of conv3x3 input -> [9*Cin, Mp]");
   m.def("avgpool2x2_fwd", &avgpool2x2_fwd, "2x2/s2 ceil avg pool (NHWC bf16)");

@@ -22,6 +22,11 @@ from . import available, ext
 # (hipBLASLt/rocBLAS via torch.matmul), "miopen" (F.conv2d).
 _CONV1X1 = os.environ.get("EDL_CONV1X1", "hip")
 
+# wgrad path: "tn" = direct TN split-K kernel on the native [M, C]
+# activation layout (no transpose_pad materializations); "bt" = the
+# transpose_pad + gemm_bt_splitk pipeline (A/B fallback).
+_WGRAD = os.environ.get("EDL_WGRAD", "tn")
+
 
 class _Conv1x1Hip(torch.autograd.Function):
     """y2d = x2d @ W^T via the gemm_bt MFMA kernel; dgrad reuses the same
@@ -45,11 +50,18 @@ class _Conv1x1Hip(torch.autograd.Function):
         dy2d = dy2d.contiguous()
         e = ext()
         dx = e.gemm_bt(dy2d, ctx.wt)
-        # wgrad: TN with reduction over huge M -> transpose-pad both
-        # operands and run the split-K bt kernel (fp32 combine); hipBLASLt's
-        # TN picks measured 272 us on these shapes, a non-split bt kernel
-        # serialized on 1-2 blocks (24 ms/step end-to-end)
-        dw = e.gemm_bt_splitk(e.transpose_pad(dy2d), e.transpose_pad(x2d), 0)
+        # wgrad: TN with reduction over huge M. Default: the direct TN
+        # split-K kernel reading dY/X in their native [M, C] layout (the
+        # "transpose" happens in the LDS fragment reads — gemm_tn.hip).
+        # Fallback "bt": transpose-pad both operands and run the split-K
+        # bt kernel; that path costs an extra HBM round-trip per operand
+        # (~1.3 ms/step, profiles/r01_step5_final.txt). hipBLASLt's TN
+        # picks measured 272 us on these shapes, a non-split bt kernel
+        # serialized on 1-2 blocks (24 ms/step end-to-end).
+        if _WGRAD == "tn":
+            dw = e.gemm_tn_splitk(dy2d, x2d, 0)
+        else:
+            dw = e.gemm_bt_splitk(e.transpose_pad(dy2d), e.transpose_pad(x2d), 0)
         if dw.dtype != ctx.w_dtype:
             dw = dw.to(ctx.w_dtype)
         return dx, dw, None, None
@@ -109,15 +121,22 @@ class _Conv3x3Hip(torch.autograd.Function):
             dx = torch.nn.grad.conv2d_input(
                 list(x.shape), weight, dy, stride=(stride, stride),
                 padding=(1, 1))
-        # wgrad: dW3[Cout, 9Cin] = dyT @ shift9(x)^T on the split-K kernel
+        # wgrad: dW3[Cout, 9Cin] = dY^T @ shift9(x). Default: the direct
+        # TN gather kernel reading dY and pad(x) in native NHWC layout
+        # (gemm_tn.hip G3 path); fallback "bt": materialize
+        # transpose_pad(dy) [Cout, Mp] + shift9_transpose(x) [9Cin, Mp]
+        # and run the split-K bt kernel.
         e = ext()
         n = dy.shape[0]
         co = weight.shape[0]
         ci = weight.shape[1]
         dy2d = dy.permute(0, 2, 3, 1).reshape(-1, co)
-        dw3 = e.gemm_bt_splitk(
-            e.transpose_pad(dy2d.to(torch.bfloat16)),
-            e.conv3x3_wgrad_operand(x, stride), 0)
+        if _WGRAD == "tn":
+            dw3 = e.gemm_tn3x3_splitk(dy2d.to(torch.bfloat16), x, stride, 0)
+        else:
+            dw3 = e.gemm_bt_splitk(
+                e.transpose_pad(dy2d.to(torch.bfloat16)),
+                e.conv3x3_wgrad_operand(x, stride), 0)
         dw = dw3.view(co, 3, 3, ci).permute(0, 3, 1, 2)  # fp32
         if dw.dtype != ctx.w_dtype:
             dw = dw.to(ctx.w_dtype)

@@ -171,3 +171,69 @@ def test_conv3x3_grouped_training_falls_back():
     y = conv(x)  # MIOpen path; must be differentiable
     y.sum().backward()
     assert x.grad is not None and conv.weight.grad is not None
+
+
+# K values exercise all four (BN1, BN2) tile configs, the K tail
+# (guarded stage_tail path), single-partial-chunk K, and splitk clamping
+@pytest.mark.parametrize("K,N1,N2,splitk", [
+    (4096, 64, 64, 0),        # (64,64,KW=4) config, 128 KB LDS
+    (100352, 64, 64, 0),      # real stage-1 wgrad shape
+    (4096, 64, 256, 0),       # (64,128,KW=2)
+    (4096, 256, 64, 0),       # (128,64,KW=2)
+    (6272, 256, 1024, 0),     # (128,128,KW=1), real stage-3 shape
+    (1000, 128, 128, 0),      # K tail (1000 % 64 = 40)
+    (63, 64, 64, 0),          # single partial chunk
+    (130, 64, 128, 8),        # splitk > nchunks -> clamped
+    (1568, 512, 2048, 0),     # real stage-4 shape
+])
+def test_gemm_tn_splitk_numerics(K, N1, N2, splitk):
+    from edl_amd import ops
+
+    torch.manual_seed(2)
+    a = (torch.randn(K, N1, device="cuda") * 1.5).to(torch.bfloat16)
+    b = (torch.randn(K, N2, device="cuda") +
+         torch.arange(N2, device="cuda") * 0.003).to(torch.bfloat16)
+    c = ops.ext().gemm_tn_splitk(a, b, splitk)
+    assert c.dtype == torch.float32 and c.shape == (N1, N2)
+    ref = a.float().t() @ b.float()
+    err = (c - ref).abs()
+    scale = ref.abs().mean().clamp(min=1)
+    assert (err / scale).max() < 0.05, (err.max().item(), scale.item())
+
+
+def test_gemm_tn_matches_bt_pipeline():
+    """Direct TN wgrad == transpose_pad + gemm_bt_splitk on the same data."""
+    from edl_amd import ops
+
+    e = ops.ext()
+    torch.manual_seed(3)
+    K, N1, N2 = 12544, 128, 512
+    a = torch.randn(K, N1, device="cuda").to(torch.bfloat16)
+    b = torch.randn(K, N2, device="cuda").to(torch.bfloat16)
+    c_tn = e.gemm_tn_splitk(a, b, 0)
+    c_bt = e.gemm_bt_splitk(e.transpose_pad(a), e.transpose_pad(b), 0)
+    assert torch.allclose(c_tn, c_bt, atol=0.5, rtol=0.02), \
+        (c_tn - c_bt).abs().max().item()
+
+
+@pytest.mark.parametrize("stride", [1, 2])
+@pytest.mark.parametrize("shape", [(2, 64, 16, 16, 64), (2, 128, 14, 14, 256),
+                                   (3, 64, 9, 11, 64)])
+def test_gemm_tn3x3_matches_bt_pipeline(stride, shape):
+    """Direct-gather conv3x3 wgrad == transpose_pad + shift9 + bt pipeline."""
+    from edl_amd import ops
+
+    e = ops.ext()
+    torch.manual_seed(4)
+    n, ci, h, w, co = shape
+    x = torch.randn(n, ci, h, w, device="cuda").to(torch.bfloat16).contiguous(
+        memory_format=torch.channels_last)
+    ho = (h - 1) // stride + 1
+    wo = (w - 1) // stride + 1
+    dy2d = torch.randn(n * ho * wo, co, device="cuda").to(torch.bfloat16)
+    c_tn = e.gemm_tn3x3_splitk(dy2d, x, stride, 0)
+    c_bt = e.gemm_bt_splitk(e.transpose_pad(dy2d),
+                            e.conv3x3_wgrad_operand(x, stride), 0)
+    assert c_tn.shape == c_bt.shape == (co, 9 * ci)
+    assert torch.allclose(c_tn, c_bt, atol=0.5, rtol=0.02), \
+        (c_tn - c_bt).abs().max().item()

@@ -50,5 +50,52 @@ def main():
             t_blas / t_hip))
 
 
+def main_wgrad():
+    """Wgrad: direct TN kernel vs the transpose_pad + bt_splitk pipeline.
+    Shapes are the ResNet50_vd 1x1 wgrads at bs32: C[Cout,Cin] over K=M."""
+    e = ops.ext()
+    print("%-28s %10s %10s %8s" % ("shape (K,N1,N2)", "tn", "pad+bt", "ratio"))
+    for M, Cout, Cin in SHAPES:
+        a = torch.randn(M, Cout, device="cuda").to(torch.bfloat16)
+        b = torch.randn(M, Cin, device="cuda").to(torch.bfloat16)
+        flops = 2.0 * M * Cout * Cin
+        t_tn = bench(lambda: e.gemm_tn_splitk(a, b, 0))
+        t_bt = bench(lambda: e.gemm_bt_splitk(
+            e.transpose_pad(a), e.transpose_pad(b), 0))
+        print("%-28s %7.1f TF %7.1f TF %7.2fx" % (
+            str((M, Cout, Cin)), flops / t_tn / 1e12, flops / t_bt / 1e12,
+            t_bt / t_tn))
+
+
+def main_wgrad3():
+    """conv3x3 wgrad: direct TN gather vs transpose_pad+shift9+bt.
+    (N, Cin, H, W, Cout, stride) = bs32 ResNet50_vd 3x3 convs."""
+    e = ops.ext()
+    shapes = [
+        (32, 64, 56, 56, 64, 1),
+        (32, 128, 28, 28, 128, 1), (32, 128, 56, 56, 128, 2),
+        (32, 256, 14, 14, 256, 1), (32, 256, 28, 28, 256, 2),
+        (32, 512, 7, 7, 512, 1), (32, 512, 14, 14, 512, 2),
+    ]
+    print("%-28s %10s %10s %8s" % ("(N,C,H,W,s)", "tn3x3", "old", "ratio"))
+    for n, ci, h, w, co, s in shapes:
+        x = torch.randn(n, ci, h, w, device="cuda").to(torch.bfloat16)
+        x = x.contiguous(memory_format=torch.channels_last)
+        ho, wo = (h - 1) // s + 1, (w - 1) // s + 1
+        dy = torch.randn(n * ho * wo, co, device="cuda").to(torch.bfloat16)
+        flops = 2.0 * n * ho * wo * co * 9 * ci
+        t_tn = bench(lambda: e.gemm_tn3x3_splitk(dy, x, s, 0))
+        t_bt = bench(lambda: e.gemm_bt_splitk(
+            e.transpose_pad(dy), e.conv3x3_wgrad_operand(x, s), 0))
+        print("%-28s %7.1f TF %7.1f TF %7.2fx" % (
+            str((n, ci, h, w, s)), flops / t_tn / 1e12, flops / t_bt / 1e12,
+            t_bt / t_tn))
+
+
 if __name__ == "__main__":
-    main()
+    if "--wgrad" in sys.argv:
+        main_wgrad()
+    elif "--wgrad3" in sys.argv:
+        main_wgrad3()
+    else:
+        main()
