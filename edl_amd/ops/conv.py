@@ -26,6 +26,12 @@ _CONV1X1 = os.environ.get("EDL_CONV1X1", "hip")
 # activation layout (no transpose_pad materializations); "bt" = the
 # transpose_pad + gemm_bt_splitk pipeline (A/B fallback).
 _WGRAD = os.environ.get("EDL_WGRAD", "tn")
+# 3x3 gather wgrad wins where the pixel dim is large relative to C
+# (stage 1/2: 1.04-1.48x) but trails the materialized pipeline on the
+# deep small-HW shapes (Cin>=256: 0.90-0.96x, gemm_bench --wgrad3) — the
+# per-row div/mod gather addressing amortizes worse at small M. Route by
+# Cin; override with EDL_WGRAD3_TN_MAXC (0 = never, 9999 = always).
+_WGRAD3_MAXC = int(os.environ.get("EDL_WGRAD3_TN_MAXC", "128"))
 
 
 class _Conv1x1Hip(torch.autograd.Function):
@@ -131,7 +137,7 @@ class _Conv3x3Hip(torch.autograd.Function):
         co = weight.shape[0]
         ci = weight.shape[1]
         dy2d = dy.permute(0, 2, 3, 1).reshape(-1, co)
-        if _WGRAD == "tn":
+        if _WGRAD == "tn" and ci <= _WGRAD3_MAXC:
             dw3 = e.gemm_tn3x3_splitk(dy2d.to(torch.bfloat16), x, stride, 0)
         else:
             dw3 = e.gemm_bt_splitk(
