@@ -276,7 +276,12 @@ __global__ void bn_bwd_reduce_kernel(
 // B1b: reduce bwd partials over blocks -> sums[2C] (= [dbeta; dgamma]).
 extern "C" __global__ void bn_bwd_finalize_kernel(
     const float* __restrict__ partial, const int nblocks,
-    float* __restrict__ sums, const int C) {
+    float* __restrict__ sums, const int C, float* __restrict__ db_acc,
+    float* __restrict__ dg_acc) {
+  // db_acc/dg_acc: optional direct-grad targets (the param's bucket-view
+  // gradient) — accumulating here removes the per-param AccumulateGrad
+  // add kernels at world 1 (~106 tiny launches/step across ResNet50's
+  // BN layers). Single writer per channel, so a plain += suffices.
   const int i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= 2 * C) return;
   const long long st = 2 * C;
@@ -287,8 +292,14 @@ extern "C" __global__ void bn_bwd_finalize_kernel(
     for (int u = 0; u < 8; ++u) sa[u] += partial[(long long)(b + u) * st + i];
   }
   for (; b < nblocks; ++b) sa[0] += partial[(long long)b * st + i];
-  sums[i] = ((sa[0] + sa[1]) + (sa[2] + sa[3])) +
-            ((sa[4] + sa[5]) + (sa[6] + sa[7]));
+  const float v = ((sa[0] + sa[1]) + (sa[2] + sa[3])) +
+                  ((sa[4] + sa[5]) + (sa[6] + sa[7]));
+  sums[i] = v;
+  if (i < C) {
+    if (db_acc) db_acc[i] += v;
+  } else if (dg_acc) {
+    dg_acc[i - C] += v;
+  }
 }
 
 // (no separate dgamma/dbeta kernel: the bwd-reduce workspace IS [dbeta; dgamma]
@@ -371,9 +382,10 @@ extern "C" void launch_bn_finalize(const float* partial, int nblocks,
 }
 
 extern "C" void launch_bn_bwd_finalize(const float* partial, int nblocks,
-                                       float* sums, int C, hipStream_t s) {
+                                       float* sums, int C, float* db_acc,
+                                       float* dg_acc, hipStream_t s) {
   hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((2 * C + 255) / 256), dim3(256),
-                     0, s, partial, nblocks, sums, C);
+                     0, s, partial, nblocks, sums, C, db_acc, dg_acc);
 }
 
 extern "C" void launch_bn_apply(const void* x, const void* res, void* y,

@@ -68,7 +68,7 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B,
     float* __restrict__ C, const int N1, const int N2, const int K,
     const int Ho = 0, const int Wo = 0, const int Hp = 0, const int Wp = 0,
-    const int Cin = 0, const int stride = 1) {
+    const int Cin = 0, const int stride = 1, const int perm_cin = 0) {
   constexpr int BK = 64;
   constexpr int W1 = BN1 / 64, W2 = BN2 / 64;  // wave sub-tiles per dim
   constexpr int GW = W1 * W2;                  // waves per k-group
@@ -296,10 +296,22 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_kernel(
     for (int reg = 0; reg < 4; ++reg) {
       const int n1 = t1 + wm + mf * 16 + r4 + reg;
       if (n1 < N1) {
-        float* crow = C + (long long)n1 * N2 + t2 + wn + cn;
+        if (perm_cin > 0) {
+          // direct-grad layout: C is the conv weight [Cout, Cin, 3, 3];
+          // our n2 = s*Cin + ci (s-major) -> offset (n1*Cin + ci)*9 + s
 #pragma unroll
-        for (int nf = 0; nf < 4; ++nf) {
-          atomicAdd(&crow[nf * 16], acc[mf][nf][reg]);
+          for (int nf = 0; nf < 4; ++nf) {
+            const int n2 = t2 + wn + cn + nf * 16;
+            const int sidx = n2 / perm_cin, ci = n2 % perm_cin;
+            atomicAdd(C + ((long long)n1 * perm_cin + ci) * 9 + sidx,
+                      acc[mf][nf][reg]);
+          }
+        } else {
+          float* crow = C + (long long)n1 * N2 + t2 + wn + cn;
+#pragma unroll
+          for (int nf = 0; nf < 4; ++nf) {
+            atomicAdd(&crow[nf * 16], acc[mf][nf][reg]);
+          }
         }
       }
     }
@@ -331,7 +343,7 @@ extern "C" void launch_gemm_tn_splitk(const void* A, const void* B, float* C,
 extern "C" void launch_gemm_tn3x3_splitk(const void* dy, const void* xpad,
                                          float* C, int Cout, int Cin, int M,
                                          int Ho, int Wo, int Hp, int Wp,
-                                         int stride, int splitk,
+                                         int stride, int splitk, int perm,
                                          hipStream_t s) {
   // dW3[Cout, 9*Cin] = dY[M, Cout]^T @ gather3x3(xpad); BN2 must divide
   // Cin so each col tile sits inside one (sy, sx) shift
@@ -344,17 +356,17 @@ extern "C" void launch_gemm_tn3x3_splitk(const void* dy, const void* xpad,
   if (b1 && b2)
     hipLaunchKernelGGL((gemm_tn_kernel<128, 128, 1, true>), grid, dim3(256),
                        lds, s, (const bf16*)dy, (const bf16*)xpad, C, Cout, N2,
-                       M, Ho, Wo, Hp, Wp, Cin, stride);
+                       M, Ho, Wo, Hp, Wp, Cin, stride, perm);
   else if (b1)
     hipLaunchKernelGGL((gemm_tn_kernel<128, 64, 2, true>), grid, dim3(256),
                        lds, s, (const bf16*)dy, (const bf16*)xpad, C, Cout, N2,
-                       M, Ho, Wo, Hp, Wp, Cin, stride);
+                       M, Ho, Wo, Hp, Wp, Cin, stride, perm);
   else if (b2)
     hipLaunchKernelGGL((gemm_tn_kernel<64, 128, 2, true>), grid, dim3(256),
                        lds, s, (const bf16*)dy, (const bf16*)xpad, C, Cout, N2,
-                       M, Ho, Wo, Hp, Wp, Cin, stride);
+                       M, Ho, Wo, Hp, Wp, Cin, stride, perm);
   else
     hipLaunchKernelGGL((gemm_tn_kernel<64, 64, 4, true>), grid, dim3(256), lds,
                        s, (const bf16*)dy, (const bf16*)xpad, C, Cout, N2, M,
-                       Ho, Wo, Hp, Wp, Cin, stride);
+                       Ho, Wo, Hp, Wp, Cin, stride, perm);
 }

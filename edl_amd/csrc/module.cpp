@@ -29,7 +29,8 @@ extern "C" void launch_bn_bwd_reduce(const void*, const unsigned char*,
                                      const void*, const float*, const float*,
                                      float*, int, long long, int, bool,
                                      hipStream_t);
-extern "C" void launch_bn_bwd_finalize(const float*, int, float*, int, hipStream_t);
+extern "C" void launch_bn_bwd_finalize(const float*, int, float*, int, float*,
+                                       float*, hipStream_t);
 extern "C" void launch_gemm_bt(const void*, const void*, void*, int, int, int,
                                hipStream_t);
 extern "C" void launch_pad_nhwc(const void*, void*, int, int, int, int, int, int,
@@ -48,7 +49,7 @@ extern "C" void launch_gemm_tn_splitk(const void*, const void*, float*, int,
                                       int, int, int, hipStream_t);
 extern "C" void launch_gemm_tn3x3_splitk(const void*, const void*, float*, int,
                                          int, int, int, int, int, int, int,
-                                         int, hipStream_t);
+                                         int, int, hipStream_t);
 extern "C" void launch_avgpool2x2_fwd(const void*, void*, int, int, int, int,
                                       int, int, hipStream_t);
 extern "C" void launch_avgpool2x2_bwd(const void*, void*, int, int, int, int,
@@ -181,7 +182,9 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, c10::optional<torch::Tensor>
                                   torch::Tensor x,
                                   torch::Tensor mean, torch::Tensor invstd,
                                   torch::Tensor gamma, bool relu, bool add,
-                                  bool training) {
+                                  bool training,
+                                  c10::optional<torch::Tensor> dgamma_acc,
+                                  c10::optional<torch::Tensor> dbeta_acc) {
   const int64_t C = gamma.numel();
   check_bn_inputs(x, C);
   const long long M = x.numel() / C;
@@ -200,8 +203,24 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, c10::optional<torch::Tensor>
   launch_bn_bwd_reduce(dyc.data_ptr(), mp, x.data_ptr(),
                        mean.data_ptr<float>(), invstd.data_ptr<float>(),
                        partial.data_ptr<float>(), grid, M, (int)C, relu, s);
+  // direct-grad mode: accumulate dgamma/dbeta straight into the params'
+  // bucket-view gradients (skips the per-param AccumulateGrad kernels)
+  float* dg_acc = nullptr;
+  float* db_acc = nullptr;
+  if (dgamma_acc.has_value()) {
+    TORCH_CHECK(dgamma_acc->numel() == C && dgamma_acc->is_contiguous() &&
+                    dgamma_acc->scalar_type() == torch::kFloat32,
+                "bn_bwd: dgamma_acc fp32 [C]");
+    dg_acc = dgamma_acc->data_ptr<float>();
+  }
+  if (dbeta_acc.has_value()) {
+    TORCH_CHECK(dbeta_acc->numel() == C && dbeta_acc->is_contiguous() &&
+                    dbeta_acc->scalar_type() == torch::kFloat32,
+                "bn_bwd: dbeta_acc fp32 [C]");
+    db_acc = dbeta_acc->data_ptr<float>();
+  }
   launch_bn_bwd_finalize(partial.data_ptr<float>(), grid, sums.data_ptr<float>(),
-                         (int)C, s);
+                         (int)C, db_acc, dg_acc, s);
   launch_bn_bwd_dx(dyc.data_ptr(), mp, x.data_ptr(),
                    mean.data_ptr<float>(), invstd.data_ptr<float>(),
                    gamma.data_ptr<float>(), sums.data_ptr<float>(), dx.data_ptr(),
@@ -331,7 +350,8 @@ torch::Tensor gemm_bt_splitk(torch::Tensor a, torch::Tensor b, int64_t splitk) {
   return c;
 }
 
-torch::Tensor gemm_tn_splitk(torch::Tensor a, torch::Tensor b, int64_t splitk) {
+torch::Tensor gemm_tn_splitk(torch::Tensor a, torch::Tensor b, int64_t splitk,
+                             c10::optional<torch::Tensor> out) {
   // fp32 C[N1,N2] = A[K,N1]^T @ B[K,N2] — conv wgrad WITHOUT the
   // transpose_pad materializations (operands in native activation layout)
   TORCH_CHECK(a.is_cuda() && b.is_cuda() && a.dim() == 2 && b.dim() == 2 &&
@@ -351,14 +371,26 @@ torch::Tensor gemm_tn_splitk(torch::Tensor a, torch::Tensor b, int64_t splitk) {
     splitk = std::max<int64_t>(1, 512 / std::max(1, tiles));
   }
   splitk = std::min<int64_t>(splitk, nchunks);
-  auto c = torch::zeros({N1, N2}, a.options().dtype(torch::kFloat32));
+  torch::Tensor c;
+  if (out.has_value()) {
+    // direct-grad mode: ACCUMULATE into the caller's fp32 buffer (the
+    // param's bucket-view gradient, pre-zeroed by reducer.zero_grad) —
+    // no fresh zeros tensor, no AccumulateGrad add afterwards
+    c = *out;
+    TORCH_CHECK(c.is_contiguous() && c.scalar_type() == torch::kFloat32 &&
+                    c.numel() == (int64_t)N1 * N2,
+                "gemm_tn_splitk: out fp32 contiguous [N1,N2]");
+  } else {
+    c = torch::zeros({N1, N2}, a.options().dtype(torch::kFloat32));
+  }
   launch_gemm_tn_splitk(ac.data_ptr(), bc.data_ptr(), c.data_ptr<float>(), N1,
                         N2, K, (int)splitk, cur_stream());
   return c;
 }
 
 torch::Tensor gemm_tn3x3_splitk(torch::Tensor dy2d, torch::Tensor x,
-                                int64_t stride, int64_t splitk) {
+                                int64_t stride, int64_t splitk,
+                                c10::optional<torch::Tensor> out) {
   // conv3x3 wgrad, fully direct: fp32 dW3[Cout, 9*Cin] = dY^T @
   // gather3x3(pad(x)) — no transpose_pad / shift9 materializations.
   // dy2d: [M, Cout] bf16 (NHWC-flattened dY); x: 4-D channels_last bf16.
@@ -388,10 +420,23 @@ torch::Tensor gemm_tn3x3_splitk(torch::Tensor dy2d, torch::Tensor x,
     splitk = std::max<int64_t>(1, 512 / std::max(1, tiles));
   }
   splitk = std::min<int64_t>(splitk, nchunks);
-  auto c = torch::zeros({Cout, 9 * Cin}, x.options().dtype(torch::kFloat32));
+  torch::Tensor c;
+  int perm = 0;
+  if (out.has_value()) {
+    // direct-grad mode: accumulate straight into the conv weight's
+    // bucket-view gradient, laid out [Cout, Cin, 3, 3] (the kernel
+    // epilogue remaps from its native s-major [Cout, 9*Cin])
+    c = *out;
+    TORCH_CHECK(c.is_contiguous() && c.scalar_type() == torch::kFloat32 &&
+                    c.numel() == (int64_t)Cout * Cin * 9,
+                "gemm_tn3x3_splitk: out fp32 contiguous [Cout,Cin,3,3]");
+    perm = Cin;
+  } else {
+    c = torch::zeros({Cout, 9 * Cin}, x.options().dtype(torch::kFloat32));
+  }
   launch_gemm_tn3x3_splitk(dyc.data_ptr(), xp.data_ptr(), c.data_ptr<float>(),
                            Cout, Cin, M, Ho, Wo, Hp, Wp, (int)stride,
-                           (int)splitk, s);
+                           (int)splitk, perm, s);
   return c;
 }
 
@@ -444,7 +489,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused NHWC bf16 BN(+add)+ReLU train fwd -> (y, mean, invstd, mask)");
   m.def("bn_fwd_eval", &bn_fwd_eval, "fused NHWC bf16 BN(+add)+ReLU eval fwd");
   m.def("bn_bwd", &bn_bwd,
-        "fused BN(+add)+ReLU bwd -> (dx, dgamma, dbeta, dres?)");
+        "fused BN(+add)+ReLU bwd -> (dx, dgamma, dbeta, dres?)",
+        pybind11::arg("dy"), pybind11::arg("msk"), pybind11::arg("x"),
+        pybind11::arg("mean"), pybind11::arg("invstd"), pybind11::arg("gamma"),
+        pybind11::arg("relu"), pybind11::arg("add"), pybind11::arg("training"),
+        pybind11::arg("dgamma_acc") = pybind11::none(),
+        pybind11::arg("dbeta_acc") = pybind11::none());
   m.def("gemm_bt", &gemm_bt, "bf16 MFMA GEMM: C[M,N] = A[M,K] @ B[N,K]^T");
   m.def("conv3x3_fwd", &conv3x3_fwd,
         "implicit-GEMM 3x3 same-pad conv (stride 1/2) -> y2d [M, Cout]");
@@ -456,10 +506,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_tn3x3_splitk", &gemm_tn3x3_splitk,
         "direct conv3x3 wgrad: fp32 dW3[Cout,9Cin] = dY^T @ gather3x3(pad(x))",
         pybind11::arg("dy2d"), pybind11::arg("x"), pybind11::arg("stride"),
-        pybind11::arg("splitk") = 0);
+        pybind11::arg("splitk") = 0, pybind11::arg("out") = pybind11::none());
   m.def("gemm_tn_splitk", &gemm_tn_splitk,
         "split-K TN GEMM: fp32 C[N1,N2] = A[K,N1]^T @ B[K,N2] (direct wgrad)",
-        pybind11::arg("a"), pybind11::arg("b"), pybind11::arg("splitk") = 0);
+        pybind11::arg("a"), pybind11::arg("b"), pybind11::arg("splitk") = 0,
+        pybind11::arg("out") = pybind11::none());
   m.def("conv3x3_wgrad_operand", &conv3x3_wgrad_operand,
         "padded shifted transpose of conv3x3 input -> [9*Cin, Mp]");
   m.def("avgpool2x2_fwd", &avgpool2x2_fwd, "2x2/s2 ceil avg pool (NHWC bf16)");

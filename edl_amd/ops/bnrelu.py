@@ -15,6 +15,15 @@ import torch.nn.functional as F
 from . import available, ext
 
 
+def _direct_grad(p):
+    """Direct-grad mode (reducer sets _edl_direct_grad at world 1): the
+    param's bucket-view gradient to accumulate into in-kernel, or None."""
+    if (getattr(p, "_edl_direct_grad", False) and p.grad is not None
+            and p.grad.dtype == torch.float32 and p.grad.is_contiguous()):
+        return p.grad
+    return None
+
+
 class _FusedBN(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x2d, gamma, beta, rmean, rvar, momentum, eps, res2d, relu):
@@ -25,15 +34,21 @@ class _FusedBN(torch.autograd.Function):
         ctx.save_for_backward(x2d, mask, mean, invstd, gamma)
         ctx.relu = relu
         ctx.has_res = res2d is not None
+        ctx.dg_t = _direct_grad(gamma)
+        ctx.db_t = _direct_grad(beta)
         return y
 
     @staticmethod
     def backward(ctx, dy):
         x2d, mask, mean, invstd, gamma = ctx.saved_tensors
         dx, dgamma, dbeta, dres = ext().bn_bwd(
-            dy, mask, x2d, mean, invstd, gamma, ctx.relu, ctx.has_res, True
+            dy, mask, x2d, mean, invstd, gamma, ctx.relu, ctx.has_res, True,
+            ctx.dg_t, ctx.db_t
         )
-        return (dx, dgamma, dbeta, None, None, None, None,
+        return (dx,
+                None if ctx.dg_t is not None else dgamma,
+                None if ctx.db_t is not None else dbeta,
+                None, None, None, None,
                 dres if ctx.has_res else None, None)
 
 

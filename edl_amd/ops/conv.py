@@ -40,14 +40,17 @@ class _Conv1x1Hip(torch.autograd.Function):
     split-K bt kernel on transpose-padded operands."""
 
     @staticmethod
-    def forward(ctx, x2d, w_param, w_bf16, wt_cached):
+    def forward(ctx, x2d, w_param, w_bf16, wt_cached, grad_tgt):
         # w_param: the fp32 (or bf16) parameter view — the DIFFERENTIABLE
         # input; w_bf16/wt_cached: per-step cached compute copies. Grads
         # come back in fp32 straight from the split-K kernel: no cast
-        # nodes on the weight path at all.
+        # nodes on the weight path at all. grad_tgt (direct-grad mode,
+        # world 1): the weight's bucket-view .grad — backward accumulates
+        # into it in-kernel and returns None, skipping AccumulateGrad.
         ctx.save_for_backward(x2d)
         ctx.wt = wt_cached  # [Cin, Cout] bf16, derived per weight epoch
         ctx.w_dtype = w_param.dtype
+        ctx.grad_tgt = grad_tgt
         return ext().gemm_bt(x2d, w_bf16)
 
     @staticmethod
@@ -65,12 +68,17 @@ class _Conv1x1Hip(torch.autograd.Function):
         # picks measured 272 us on these shapes, a non-split bt kernel
         # serialized on 1-2 blocks (24 ms/step end-to-end).
         if _WGRAD == "tn":
+            if ctx.grad_tgt is not None:
+                e.gemm_tn_splitk(
+                    dy2d, x2d, 0,
+                    ctx.grad_tgt.view(ctx.grad_tgt.shape[0], -1))
+                return dx, None, None, None, None
             dw = e.gemm_tn_splitk(dy2d, x2d, 0)
         else:
             dw = e.gemm_bt_splitk(e.transpose_pad(dy2d), e.transpose_pad(x2d), 0)
         if dw.dtype != ctx.w_dtype:
             dw = dw.to(ctx.w_dtype)
-        return dx, dw, None, None
+        return dx, dw, None, None, None
 
 
 _CONV3X3 = os.environ.get("EDL_CONV3X3", "hip")
@@ -102,11 +110,13 @@ class _Conv3x3Hip(torch.autograd.Function):
     torch.nn.grad (MIOpen) until the TN/transposed kernels land."""
 
     @staticmethod
-    def forward(ctx, x, w_param, w_bf16, w3_cached, w3rot_cached, stride):
+    def forward(ctx, x, w_param, w_bf16, w3_cached, w3rot_cached, stride,
+                grad_tgt):
         ctx.save_for_backward(x, w_bf16)
         ctx.stride = stride
         ctx.w3rot = w3rot_cached
         ctx.w_dtype = w_param.dtype
+        ctx.grad_tgt = grad_tgt
         y2d = ext().conv3x3_fwd(x, w3_cached, stride)
         n, _, h, w = x.shape
         ho = (h - 1) // stride + 1
@@ -138,6 +148,12 @@ class _Conv3x3Hip(torch.autograd.Function):
         ci = weight.shape[1]
         dy2d = dy.permute(0, 2, 3, 1).reshape(-1, co)
         if _WGRAD == "tn" and ci <= _WGRAD3_MAXC:
+            if ctx.grad_tgt is not None:
+                # accumulate straight into the bucket-view grad, laid out
+                # [Cout, Cin, 3, 3] (kernel epilogue remaps)
+                e.gemm_tn3x3_splitk(dy2d.to(torch.bfloat16), x, stride, 0,
+                                    ctx.grad_tgt)
+                return dx, None, None, None, None, None, None
             dw3 = e.gemm_tn3x3_splitk(dy2d.to(torch.bfloat16), x, stride, 0)
         else:
             dw3 = e.gemm_bt_splitk(
@@ -146,7 +162,7 @@ class _Conv3x3Hip(torch.autograd.Function):
         dw = dw3.view(co, 3, 3, ci).permute(0, 3, 1, 2)  # fp32
         if dw.dtype != ctx.w_dtype:
             dw = dw.to(ctx.w_dtype)
-        return dx, dw, None, None, None, None
+        return dx, dw, None, None, None, None, None
 
 
 def _repack_w3_grouped(weight):
@@ -168,6 +184,17 @@ def _repack_w3_grouped(weight):
 
 class Conv2dFast(nn.Conv2d):
     """nn.Conv2d drop-in; 1x1/group-1 convs on CUDA bypass MIOpen."""
+
+    def _grad_tgt(self):
+        """Direct-grad mode (reducer sets _edl_direct_grad at world 1):
+        the weight's bucket-view gradient, or None. fp32 params only —
+        the wgrad kernels accumulate fp32."""
+        w = self.weight
+        if (getattr(w, "_edl_direct_grad", False) and w.grad is not None
+                and w.grad.dtype == torch.float32 and w.grad.is_contiguous()
+                and torch.is_grad_enabled() and self.training):
+            return w.grad
+        return None
 
     def _cached(self, key, fn):
         cache = getattr(self, "_w_cache", None)
@@ -203,8 +230,12 @@ class Conv2dFast(nn.Conv2d):
             if self.stride[0] == 1:
                 w3rot = self._cached("w3rot", lambda: _repack_w3(
                     self.weight.detach().permute(1, 0, 2, 3).flip(2, 3)))
+            # NOTE: direct-grad stays OFF for 3x3 — the [Cout,Cin,3,3]
+            # remap makes the split-K epilogue atomics stride-9 scattered
+            # (measured: slower end-to-end than the AccumulateGrad add it
+            # saves). 1x1/BN targets are layout-native and stay direct.
             return _Conv3x3Hip.apply(x, self.weight, w_bf16, w3, w3rot,
-                                     self.stride[0])
+                                     self.stride[0], None)
         if (
             # Measured SLOWER than MIOpen's grouped path on the ResNeXt
             # teacher (distill 486 -> 415 img/s): the 4x block-diagonal
@@ -263,7 +294,8 @@ class Conv2dFast(nn.Conv2d):
                                       .to(torch.bfloat16).contiguous())
                 wt_t = self._cached("wt_t", lambda: self.weight.detach().view(
                     self.out_channels, c).to(torch.bfloat16).t().contiguous())
-                y2d = _Conv1x1Hip.apply(x2d, wt, w_bf16, wt_t)
+                y2d = _Conv1x1Hip.apply(x2d, wt, w_bf16, wt_t,
+                                        self._grad_tgt())
             else:
                 y2d = x2d @ wt.t()
             return (

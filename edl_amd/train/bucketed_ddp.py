@@ -22,6 +22,8 @@ Gradient averaging (the 1/world factor) is NOT applied here; the fused
 SGD kernel folds it into its update (grad_scale), saving one pass over
 all gradients per step.
 """
+import os
+
 import torch
 import torch.distributed as dist
 
@@ -55,6 +57,7 @@ class BucketedAllReducer:
         self._buckets = []
         self._param_bucket = {}
         self._build(params, bucket_cap_mb)
+        self._mark_direct_grads()
         if self._enabled:
             for p in params:
                 h = p.register_post_accumulate_grad_hook(self._on_grad)
@@ -109,9 +112,23 @@ class BucketedAllReducer:
         self._buckets = []
         self._param_bucket = {}
         self._build(self._params, bucket_cap_mb)
+        self._mark_direct_grads()
         if self._enabled:
             for p in self._params:
                 self._hooks.append(p.register_post_accumulate_grad_hook(self._on_grad))
+
+    def _mark_direct_grads(self):
+        """World 1 (no collectives, no readiness hooks needed): flag every
+        param so the custom HIP backward ops accumulate straight into the
+        bucket-view .grad and return None — this removes one
+        AccumulateGrad add kernel per parameter per step (~150 small
+        launches / ~0.7 ms on ResNet50_vd) plus the fresh zeros the wgrad
+        kernels otherwise allocate. At world > 1 the flag stays off: the
+        overlap machinery needs the post-accumulate hooks to fire."""
+        direct = (not self._enabled
+                  and os.environ.get("EDL_DIRECT_GRAD", "1") != "0")
+        for p in self._params:
+            p._edl_direct_grad = direct
 
     # ---- per-step protocol ----
     def zero_grad(self):

@@ -237,3 +237,21 @@ def test_gemm_tn3x3_matches_bt_pipeline(stride, shape):
     assert c_tn.shape == c_bt.shape == (co, 9 * ci)
     assert torch.allclose(c_tn, c_bt, atol=0.5, rtol=0.02), \
         (c_tn - c_bt).abs().max().item()
+
+
+def test_gemm_tn3x3_accumulate_out_layout():
+    """out= path accumulates in the conv-weight [Cout,Cin,3,3] layout."""
+    from edl_amd import ops
+
+    e = ops.ext()
+    torch.manual_seed(5)
+    n, ci, h, w, co = 2, 64, 10, 10, 64
+    x = torch.randn(n, ci, h, w, device="cuda").to(torch.bfloat16).contiguous(
+        memory_format=torch.channels_last)
+    dy2d = torch.randn(n * h * w, co, device="cuda").to(torch.bfloat16)
+    native = e.gemm_tn3x3_splitk(dy2d, x, 1, 0)
+    out = torch.full((co, ci, 3, 3), 0.25, device="cuda")
+    e.gemm_tn3x3_splitk(dy2d, x, 1, 0, out)
+    ref = native.view(co, 3, 3, ci).permute(0, 3, 1, 2) + 0.25
+    assert torch.allclose(out, ref, atol=1e-3, rtol=1e-3), \
+        (out - ref).abs().max().item()

@@ -171,3 +171,41 @@ def test_avgpool2x2_numerics(H, W):
         (y.float() - yr.detach()).abs().max().item()
     assert torch.allclose(x.grad.float(), xr.grad, atol=2e-2, rtol=2e-2), \
         (x.grad.float() - xr.grad).abs().max().item()
+
+
+@pytest.mark.parametrize("seed", [0, 1])
+def test_direct_grad_matches_standard(seed, monkeypatch):
+    """World-1 direct-grad mode (kernels accumulate into bucket-view grads,
+    autograd sees None) must produce the same gradients as the standard
+    AccumulateGrad path."""
+    import torch.nn as nn
+
+    from edl_amd.ops.bnrelu import BNReLU2d
+    from edl_amd.ops.conv import Conv2dFast
+    from edl_amd.train.bucketed_ddp import BucketedAllReducer
+
+    def run(direct):
+        monkeypatch.setenv("EDL_DIRECT_GRAD", "1" if direct else "0")
+        torch.manual_seed(seed)
+        m = nn.Sequential(
+            Conv2dFast(64, 128, 1, bias=False),
+            BNReLU2d(128),
+            Conv2dFast(128, 64, 3, padding=1, bias=False),
+            BNReLU2d(64, act=False),
+        ).cuda().to(memory_format=torch.channels_last)
+        red = BucketedAllReducer(list(m.parameters()), flatten_params=True)
+        red.zero_grad()
+        torch.manual_seed(123)
+        x = (torch.randn(4, 64, 12, 12, device="cuda")
+             .to(torch.bfloat16).contiguous(memory_format=torch.channels_last))
+        y = x
+        for mod in m:
+            y = mod(y)
+        y.float().pow(2).mean().backward()
+        return [p.grad.detach().clone() for p in m.parameters()]
+
+    g_std = run(False)
+    g_dir = run(True)
+    for a, b in zip(g_std, g_dir):
+        assert torch.allclose(a, b, atol=1e-3, rtol=1e-3), \
+            (a - b).abs().max().item()
