@@ -356,9 +356,14 @@ extern "C" int bn_stats_grid(long long M, int C) {
   long long rows_per_block = 256 / c8 > 0 ? 256 / c8 : 1;
   long long want = (M + rows_per_block - 1) / rows_per_block;
   // cap bounds the finalize kernels' serial partial-read loop (they were
-  // latency-bound at ~12-14 us with 512-block partials; ~5 us at 192)
+  // latency-bound at ~12-14 us with 512-block partials; ~5 us at 192).
+  // EDL_BN_GRID_CAP overrides for A/B (192 = measured best tradeoff).
+  static const long long cap_env = []() {
+    const char* e = getenv("EDL_BN_GRID_CAP");
+    return e ? atoll(e) : 192LL;
+  }();
   long long cap = 131072 / (2 * (long long)C);
-  if (cap > 192) cap = 192;
+  if (cap > cap_env) cap = cap_env;
   if (cap < 8) cap = 8;
   long long g = want < cap ? want : cap;
   return (int)(g > 0 ? g : 1);
