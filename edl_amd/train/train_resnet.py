@@ -39,7 +39,12 @@ def parse_args(argv=None):
     p.add_argument("--checkpoint", default=None, help="checkpoint dir (resume + save)")
     p.add_argument("--checkpoint_steps", type=int, default=0,
                    help="also checkpoint every N steps (0 = per-epoch only)")
-    p.add_argument("--data_dir", default=None, help="unused (synthetic data)")
+    p.add_argument("--data_dir", default=None,
+                   help="txt record files: train through the elastic "
+                        "data plane (leader-balanced Reader) instead of "
+                        "the synthetic loader; records deterministically "
+                        "seed synthetic images (no dataset network)")
+    p.add_argument("--image_hw", type=int, default=224)
     p.add_argument("--use_hip_ops", type=int, default=1)
     p.add_argument("--graph_capture", type=int, default=None)
     p.add_argument("--bucket_mb", type=int, default=25)
@@ -85,11 +90,37 @@ def main(argv=None):
                 pass
 
     report(TrainStatus.RUNNING)
-    loader = SyntheticImageNet(
-        args.batch_size, engine.device,
-        channels_last=engine.channels_last and engine.device.type == "cuda",
-        seed=1234 + engine.env.global_rank,
-    )
+    shape = (3, args.image_hw, args.image_hw)
+    cl = engine.channels_last and engine.device.type == "cuda"
+    loader = None
+    if not args.data_dir:
+        loader = SyntheticImageNet(
+            args.batch_size, engine.device, image_shape=shape,
+            channels_last=cl, seed=1234 + engine.env.global_rank,
+        )
+
+    def plane_epoch_loader():
+        """Elastic data plane: leader-balanced record slices -> loader.
+        Ranks can receive slightly unequal record counts, so all ranks
+        agree on the MIN step count before training (collectives must
+        stay matched across the world)."""
+        import torch.distributed as dist
+
+        from ..data.plane import RecordImageSet, fetch_epoch_records
+
+        recs = fetch_epoch_records(tenv, args.data_dir, args.batch_size)
+        ds = RecordImageSet(recs, args.batch_size, engine.device,
+                            image_shape=shape, channels_last=cl)
+        steps = ds.steps()
+        if dist.is_initialized():
+            t = torch.tensor([steps], dtype=torch.long,
+                             device=engine.device
+                             if dist.get_backend() == "nccl" else "cpu")
+            dist.all_reduce(t, op=dist.ReduceOp.MIN)
+            steps = int(t.item())
+        if args.steps_per_epoch:
+            steps = min(steps, args.steps_per_epoch)
+        return ds, steps
 
     def on_step(epoch, it):
         if args.checkpoint_steps and (engine.global_step % args.checkpoint_steps == 0):
@@ -100,8 +131,15 @@ def main(argv=None):
     for epoch in range(engine.start_epoch, args.num_epochs):
         if epoch == args.num_epochs - 1:
             report(TrainStatus.NEARTHEEND)
+        if args.data_dir:
+            ep_loader, ep_steps = plane_epoch_loader()
+            if ep_steps == 0:
+                log.warning("epoch %d: no full batch from the data plane", epoch)
+                continue
+        else:
+            ep_loader, ep_steps = loader, args.steps_per_epoch
         stats = engine.train_epoch(
-            epoch, loader, args.steps_per_epoch, on_step=on_step,
+            epoch, ep_loader, ep_steps, on_step=on_step,
             start_step=engine.start_step if epoch == engine.start_epoch else 0)
         if engine.env.is_rank0:
             log.info("epoch %d done: %.1f img/s (world=%d, global_batch=%d)",

@@ -237,3 +237,19 @@ def test_mid_epoch_step_resume(tmp_path):
                          base_lr=0.001, checkpoint_dir=str(tmp_path / "ck"))
     eng3.setup()
     assert eng3.start_epoch == 4 and eng3.start_step == 0
+
+
+def test_direct_grad_flags_world1_only(monkeypatch):
+    """Reducer marks params for in-kernel grad accumulation only at
+    world 1 (no collectives); EDL_DIRECT_GRAD=0 disables."""
+    import torch
+
+    from edl_amd.train.bucketed_ddp import BucketedAllReducer
+
+    ps = [torch.nn.Parameter(torch.randn(8, 8)) for _ in range(3)]
+    BucketedAllReducer(ps, flatten_params=True)
+    assert all(p._edl_direct_grad for p in ps)
+
+    monkeypatch.setenv("EDL_DIRECT_GRAD", "0")
+    BucketedAllReducer(ps, flatten_params=True)
+    assert not any(p._edl_direct_grad for p in ps)

@@ -134,3 +134,30 @@ def test_fit_a_line_elastic_data_plane(coord_server, tmp_path):
     counts = [int(m) for m in re.findall(r"consumed (\d+) records", logs)]
     assert sum(counts) == 128, counts
     assert all(c > 0 for c in counts), counts
+
+
+def test_resnet_elastic_data_plane(coord_server, tmp_path):
+    """ResNet trained from FILES through the elastic data plane at world 2
+    (leader-balanced Reader feeding deterministic record->image synthesis;
+    reference: DALI file pipeline + utils/data_server.py balancing)."""
+    data = tmp_path / "data"
+    data.mkdir()
+    for i in range(4):
+        rows = ["%d rec-%d-%d" % ((i * 20 + j) % 10, i, j) for j in range(20)]
+        (data / ("part%d.txt" % i)).write_text("\n".join(rows) + "\n")
+
+    run_edlrun(
+        coord_server, tmp_path,
+        ["-m", "edl_amd.train.train_resnet", "--model", "resnet50_vd",
+         "--num_epochs", "1", "--batch_size", "4", "--image_hw", "32",
+         "--steps_per_epoch", "3", "--data_dir", str(data),
+         "--checkpoint", str(tmp_path / "ck")],
+        timeout=240,
+    )
+    assert (tmp_path / "ck" / "checkpoint.0").is_dir()
+    logs = "".join((tmp_path / ("agent%d.log" % i)).read_text() for i in (0, 1))
+    import re
+
+    counts = [int(m) for m in re.findall(r"rank \d+: (\d+) records", logs)]
+    assert len(counts) == 2 and sum(counts) == 80, counts
+    assert "epoch 0 done" in logs
