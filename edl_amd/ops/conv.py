@@ -223,13 +223,19 @@ class Conv2dFast(nn.Conv2d):
         ):
             if not x.is_contiguous(memory_format=torch.channels_last):
                 x = x.contiguous(memory_format=torch.channels_last)
-            w_bf16 = self._cached("w_bf16", lambda: self.weight.detach().to(
-                torch.bfloat16).contiguous())
-            w3 = self._cached("w3", lambda: _repack_w3(self.weight.detach()))
+            mb = getattr(self.weight, "_edl_bf16", None)
+            if mb is not None:  # engine bucket mirror: zero-cost bf16 copy
+                w_bf16 = mb
+                wsrc = mb
+            else:
+                w_bf16 = self._cached("w_bf16", lambda: self.weight.detach().to(
+                    torch.bfloat16).contiguous())
+                wsrc = self.weight.detach()
+            w3 = self._cached("w3", lambda: _repack_w3(wsrc))
             w3rot = None
             if self.stride[0] == 1:
                 w3rot = self._cached("w3rot", lambda: _repack_w3(
-                    self.weight.detach().permute(1, 0, 2, 3).flip(2, 3)))
+                    wsrc.permute(1, 0, 2, 3).flip(2, 3)))
             # NOTE: direct-grad stays OFF for 3x3 — the [Cout,Cin,3,3]
             # remap makes the split-K epilogue atomics stride-9 scattered
             # (measured: slower end-to-end than the AccumulateGrad add it
@@ -289,11 +295,17 @@ class Conv2dFast(nn.Conv2d):
                 and c % 64 == 0
                 and self.out_channels % 64 == 0
             ):
-                w_bf16 = self._cached("w_bf16", lambda: self.weight.detach()
-                                      .view(self.out_channels, c)
-                                      .to(torch.bfloat16).contiguous())
-                wt_t = self._cached("wt_t", lambda: self.weight.detach().view(
-                    self.out_channels, c).to(torch.bfloat16).t().contiguous())
+                mb = getattr(self.weight, "_edl_bf16", None)
+                if mb is not None:  # engine bucket mirror: free bf16 view
+                    w_bf16 = mb.view(self.out_channels, c)
+                    wt_t = self._cached("wt_t", lambda: mb.view(
+                        self.out_channels, c).t().contiguous())
+                else:
+                    w_bf16 = self._cached("w_bf16", lambda: self.weight.detach()
+                                          .view(self.out_channels, c)
+                                          .to(torch.bfloat16).contiguous())
+                    wt_t = self._cached("wt_t", lambda: self.weight.detach().view(
+                        self.out_channels, c).to(torch.bfloat16).t().contiguous())
                 y2d = _Conv1x1Hip.apply(x2d, wt, w_bf16, wt_t,
                                         self._grad_tgt())
             else:

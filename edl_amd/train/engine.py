@@ -161,8 +161,33 @@ class TrainerEngine:
         from ..ops.conv import bump_weight_epoch
 
         bump_weight_epoch()  # weights changed outside an optimizer step
+        self._init_bf16_mirrors()
         edist.barrier(self.device)
         return self
+
+    def _init_bf16_mirrors(self):
+        """Per-bucket bf16 mirror of the flat fp32 params: each conv's
+        plain bf16 compute copy becomes a VIEW into the mirror, and one
+        copy_ per bucket after every optimizer step replaces ~100
+        individual per-weight cast kernels per step (the repacked
+        layouts — transposed/s-major/rotated — stay per-weight)."""
+        self._bf16_mirrors = []
+        if not (self.use_hip_ops and self.device.type == "cuda"):
+            return
+        for b in self.reducer._buckets:
+            if b.param_flat is None or b.param_flat.dtype != torch.float32:
+                continue
+            mirror = torch.empty_like(b.param_flat, dtype=torch.bfloat16)
+            self._bf16_mirrors.append((mirror, b.param_flat))
+            off = 0
+            for p in b.params:
+                p._edl_bf16 = mirror[off:off + p.numel()].view_as(p)
+                off += p.numel()
+        self._refresh_bf16_mirrors()
+
+    def _refresh_bf16_mirrors(self):
+        for mirror, flat in getattr(self, "_bf16_mirrors", ()):
+            mirror.copy_(flat, non_blocking=True)
 
     def _build_optimizer(self):
         from ..ops.sgd import FusedSGD
@@ -272,6 +297,7 @@ class TrainerEngine:
                     for b in self.reducer._buckets:
                         b.buffer.mul_(scale)
             self.opt.step()
+            self._refresh_bf16_mirrors()
         self.global_step += 1
         return loss
 
