@@ -93,3 +93,27 @@ def test_two_pod_reader_rebalance(tmp_path):
         r0.close(); r1.close()
     finally:
         s0.stop(); s1.stop()
+
+
+def test_record_image_set_deterministic():
+    """plane.RecordImageSet: equal records -> equal tensors on any rank
+    (elastic resizes must resume on identical data); labels parse from a
+    leading integer, else derive from the record hash."""
+    import torch
+
+    from edl_amd.data.plane import RecordImageSet
+
+    recs = ["3 img-a", "7 img-b", "not-an-int payload", "3 img-a"]
+    ds1 = RecordImageSet(list(recs), batch_size=2, device=torch.device("cpu"),
+                         image_shape=(3, 8, 8), num_classes=10)
+    ds2 = RecordImageSet(list(recs), batch_size=2, device=torch.device("cpu"),
+                         image_shape=(3, 8, 8), num_classes=10)
+    x1, y1 = ds1.next()
+    x2, y2 = ds2.next()
+    assert torch.equal(x1, x2) and torch.equal(y1, y2)
+    assert y1.tolist() == [3, 7]
+    xb, yb = ds1.next()
+    assert 0 <= yb[0] < 10          # hash-derived label in range
+    assert yb[1] == 3               # wraps to equal record -> equal label
+    assert torch.equal(xb[1], x1[0])  # equal record -> equal image
+    assert ds1.steps() == 2
