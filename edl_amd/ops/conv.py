@@ -37,7 +37,8 @@ _WGRAD3_MAXC = int(os.environ.get("EDL_WGRAD3_TN_MAXC", "128"))
 class _Conv1x1Hip(torch.autograd.Function):
     """y2d = x2d @ W^T via the gemm_bt MFMA kernel; dgrad reuses the same
     kernel on a (per-step cached) transposed-weight copy; wgrad runs the
-    split-K bt kernel on transpose-padded operands."""
+    direct TN split-K kernel on the native [M, C] operands (gemm_tn.hip),
+    optionally accumulating straight into the bucket-view grad."""
 
     @staticmethod
     def forward(ctx, x2d, w_param, w_bf16, wt_cached, grad_tgt):
@@ -106,8 +107,9 @@ class _Conv3x3Hip(torch.autograd.Function):
     """3x3 same-pad conv on the implicit-GEMM kernel.
 
     dgrad (stride 1) is ANOTHER 3x3 stride-1 conv with rotated/transposed
-    weights on the same kernel; stride-2 dgrad and all wgrads go through
-    torch.nn.grad (MIOpen) until the TN/transposed kernels land."""
+    weights on the same kernel; stride-2 dgrad goes through torch.nn.grad
+    (MIOpen); wgrad runs the direct TN gather kernel (Cin<=128) or the
+    shift9+split-K pipeline (deeper layers)."""
 
     @staticmethod
     def forward(ctx, x, w_param, w_bf16, w3_cached, w3rot_cached, stride,
