@@ -18,12 +18,22 @@ log = get_logger("edl.coord")
 
 
 class CoordClient:
-    def __init__(self, endpoints, job_id="", timeout=6.0):
-        """endpoints: 'host:port' or comma-separated list (first reachable wins)."""
+    def __init__(self, endpoints, job_id="", timeout=6.0, retry_s=None):
+        """endpoints: 'host:port' or comma-separated list (first reachable wins).
+
+        retry_s (or env EDL_STORE_RETRY_S): keep retrying a failed RPC for
+        this many seconds before raising — rides out a coordd restart on
+        the same endpoint (the durability etcd gave the reference; pair
+        with the server's --snapshot). Default 0: fail after one
+        reconnect attempt (failure-detection paths stay prompt)."""
+        import os
+
         if isinstance(endpoints, str):
             endpoints = [e for e in endpoints.split(",") if e]
         self._endpoints = endpoints
         self._timeout = timeout
+        self._retry_s = (float(os.environ.get("EDL_STORE_RETRY_S", "0"))
+                         if retry_s is None else float(retry_s))
         self.job_id = job_id
         self._lock = threading.Lock()
         self._sock = None
@@ -39,22 +49,34 @@ class CoordClient:
         raise EdlStoreError("cannot reach coordd at %s: %s" % (self._endpoints, last))
 
     def _call(self, req):
+        import time as _time
+
         with self._lock:
-            for attempt in (0, 1):
-                if self._sock is None:
-                    self._sock = self._connect()
+            deadline = _time.monotonic() + self._retry_s
+            attempt = 0
+            while True:
                 try:
+                    if self._sock is None:
+                        self._sock = self._connect()
                     protocol.send_msg(self._sock, req)
                     resp = protocol.recv_msg(self._sock)
                     break
-                except (ConnectionError, OSError) as e:
-                    try:
-                        self._sock.close()
-                    except OSError:
-                        pass
-                    self._sock = None
-                    if attempt:
-                        raise EdlStoreError("coordd rpc failed: %s" % e)
+                except (EdlStoreError, ConnectionError, OSError) as e:
+                    if self._sock is not None:
+                        try:
+                            self._sock.close()
+                        except OSError:
+                            pass
+                        self._sock = None
+                    attempt += 1
+                    # one immediate reconnect (a half-open socket), then
+                    # back off within the retry window (store restart)
+                    if attempt > 1:
+                        if _time.monotonic() >= deadline:
+                            if isinstance(e, EdlStoreError):
+                                raise
+                            raise EdlStoreError("coordd rpc failed: %s" % e)
+                        _time.sleep(0.3)
         if not resp.get("ok"):
             raise EdlStoreError(resp.get("err", "unknown store error"))
         return resp

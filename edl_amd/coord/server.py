@@ -18,6 +18,8 @@ path, so a lock-around-a-dict threaded server is the right amount of
 machinery.
 """
 import argparse
+import json
+import os
 import threading
 import time
 import socket
@@ -36,12 +38,36 @@ class _State:
         self.leases = {}  # id -> dict(ttl=float, deadline=float)
         self.rev = 0
         self.next_lease = 1
+        self.dirty = False  # snapshot needed (set by bump)
 
     # All methods below assume self.lock is held.
     def bump(self):
         self.rev += 1
+        self.dirty = True
         self.lock.notify_all()
         return self.rev
+
+    # ---- snapshot persistence (the durability etcd gave the reference:
+    # a store restart must not take every job down with it) ----
+    def to_snapshot(self):
+        now = time.monotonic()
+        return {
+            "kv": self.kv,
+            "leases": {str(i): {"ttl": L["ttl"]} for i, L in self.leases.items()
+                       if L["deadline"] > now},
+            "rev": self.rev,
+            "next_lease": self.next_lease,
+        }
+
+    def load_snapshot(self, d):
+        self.kv = dict(d.get("kv", {}))
+        now = time.monotonic()
+        # every surviving lease gets one full-TTL grace window: owners
+        # resume keepalives as soon as their client reconnects
+        self.leases = {int(i): {"ttl": L["ttl"], "deadline": now + L["ttl"]}
+                       for i, L in d.get("leases", {}).items()}
+        self.rev = int(d.get("rev", 0))
+        self.next_lease = int(d.get("next_lease", 1))
 
     def alive(self, lease_id):
         L = self.leases.get(lease_id)
@@ -219,9 +245,20 @@ class _TCPServer(socketserver.ThreadingTCPServer):
 class CoordServer:
     """In-process coordination store server."""
 
-    def __init__(self, host="127.0.0.1", port=0):
+    def __init__(self, host="127.0.0.1", port=0, snapshot=None):
+        """snapshot: optional path — state is persisted there (atomic
+        rename, written by the sweeper when dirty and on stop) and
+        reloaded on construction, so a store restart on the same
+        endpoint preserves keys and gives leases one TTL of grace."""
         self._srv = _TCPServer((host, port), _Handler)
         self._srv.state = _State()
+        self._snapshot = snapshot
+        if snapshot and os.path.exists(snapshot):
+            with open(snapshot) as f:
+                self._srv.state.load_snapshot(json.load(f))
+            log.info("coordd: restored %d keys, %d leases from %s",
+                     len(self._srv.state.kv), len(self._srv.state.leases),
+                     snapshot)
         self._thread = None
         self._sweeper = None
         self._stop = threading.Event()
@@ -243,19 +280,37 @@ class CoordServer:
         while not self._stop.wait(0.5):
             with st.lock:
                 st.expire_leases()
+            self._maybe_snapshot()
+
+    def _maybe_snapshot(self, force=False):
+        if not self._snapshot:
+            return
+        st = self._srv.state
+        with st.lock:
+            if not (st.dirty or force):
+                return
+            snap = st.to_snapshot()
+            st.dirty = False
+        tmp = self._snapshot + ".tmp"
+        with open(tmp, "w") as f:
+            json.dump(snap, f)
+        os.replace(tmp, self._snapshot)
 
     def stop(self):
         self._stop.set()
         self._srv.shutdown()
         self._srv.server_close()
+        self._maybe_snapshot(force=True)
 
 
 def main():
     ap = argparse.ArgumentParser(description="edl_amd coordination store server")
     ap.add_argument("--host", default="0.0.0.0")
     ap.add_argument("--port", type=int, default=2379)
+    ap.add_argument("--snapshot", default=None,
+                    help="persist state here; reload on restart")
     args = ap.parse_args()
-    srv = CoordServer(args.host, args.port).start()
+    srv = CoordServer(args.host, args.port, snapshot=args.snapshot).start()
     log.info("coordd listening on %s", srv.endpoint)
     try:
         while True:
