@@ -59,6 +59,36 @@ def piecewise_lr(base_lr, epoch, boundaries=(30, 60, 90), decay=0.1, warmup_epoc
     return base_lr * mult
 
 
+class DynamicLossScaler:
+    """Dynamic loss scaling for fp16 training (reference
+    train_with_fleet.py:318-321: mixed_precision.decorate with
+    scale_loss=128 + use_dynamic_loss_scaling). bf16 never needs it —
+    this engages only for dtype=fp16: the loss is multiplied by `value`
+    before backward, the optimizer folds 1/value into its grad scale,
+    and a step whose gradients overflowed is SKIPPED while the scale
+    backs off; `growth_interval` clean steps double it again."""
+
+    def __init__(self, init_scale=2.0 ** 15, growth_factor=2.0,
+                 backoff_factor=0.5, growth_interval=2000):
+        self.value = float(init_scale)
+        self.growth_factor = growth_factor
+        self.backoff_factor = backoff_factor
+        self.growth_interval = growth_interval
+        self._good_steps = 0
+
+    def update(self, found_inf):
+        """-> True if the step should be applied, False to skip it."""
+        if found_inf:
+            self.value = max(1.0, self.value * self.backoff_factor)
+            self._good_steps = 0
+            return False
+        self._good_steps += 1
+        if self._good_steps >= self.growth_interval:
+            self.value *= self.growth_factor
+            self._good_steps = 0
+        return True
+
+
 class TrainerEngine:
     def __init__(
         self,
@@ -91,6 +121,8 @@ class TrainerEngine:
         self.label_smoothing = label_smoothing
         self.dtype = {"bf16": torch.bfloat16, "fp16": torch.float16,
                       "fp32": torch.float32}[dtype]
+        # fp16 needs dynamic loss scaling (bf16/fp32 do not)
+        self.scaler = DynamicLossScaler() if self.dtype == torch.float16 else None
         self.channels_last = channels_last
         self.bucket_mb = bucket_mb
         self.checkpoint_dir = checkpoint_dir
@@ -284,19 +316,43 @@ class TrainerEngine:
                 logits = self.model(images)
             loss = self._loss(logits, labels, teacher_logits)
         with _mark("edl.backward"):
-            loss.backward()
+            if self.scaler is not None:
+                (loss * self.scaler.value).backward()
+            else:
+                loss.backward()
         with _mark("edl.allreduce_finalize"):
             if self.dgc is not None:
                 self.dgc.step()
             else:
                 self.reducer.finalize()
         with _mark("edl.optimizer"):
-            if not getattr(self.opt, "handles_grad_scale", False):
-                scale = self.reducer.grad_scale
+            inv_scale = 1.0
+            if self.scaler is not None:
+                # buckets already hold the all-reduced (scaled) grads, so
+                # every rank sees the same values and makes the same call
+                found_inf = any(
+                    not bool(torch.isfinite(b.buffer).all())
+                    for b in self.reducer._buckets)
+                if not self.scaler.update(found_inf):
+                    self.global_step += 1
+                    return loss  # overflow: skip the update, scale backed off
+                inv_scale = 1.0 / self.scaler.value
+            if getattr(self.opt, "handles_grad_scale", False):
+                if inv_scale != 1.0:
+                    base = self.opt.grad_scale
+                    self.opt.grad_scale = base * inv_scale
+                    try:
+                        self.opt.step()
+                    finally:
+                        self.opt.grad_scale = base
+                else:
+                    self.opt.step()
+            else:
+                scale = self.reducer.grad_scale * inv_scale
                 if scale != 1.0:
                     for b in self.reducer._buckets:
                         b.buffer.mul_(scale)
-            self.opt.step()
+                self.opt.step()
             self._refresh_bf16_mirrors()
         self.global_step += 1
         return loss
@@ -308,6 +364,10 @@ class TrainerEngine:
         graph removes per-kernel launch gaps (guide §launches-baseline).
         Returns True if capture succeeded."""
         if self.device.type != "cuda" or self._graph is not None:
+            return False
+        if self.scaler is not None:
+            # fp16 loss scaling is data-dependent control flow (inf check,
+            # step skip) — not capturable as one static graph
             return False
         want = self.graph_capture
         if want is None:

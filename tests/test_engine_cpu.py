@@ -253,3 +253,43 @@ def test_direct_grad_flags_world1_only(monkeypatch):
     monkeypatch.setenv("EDL_DIRECT_GRAD", "0")
     BucketedAllReducer(ps, flatten_params=True)
     assert not any(p._edl_direct_grad for p in ps)
+
+
+def test_dynamic_loss_scaler_policy():
+    from edl_amd.train.engine import DynamicLossScaler
+
+    s = DynamicLossScaler(init_scale=1024.0, growth_interval=3)
+    assert s.update(found_inf=False) and s.value == 1024.0
+    assert not s.update(found_inf=True)          # overflow: skip + backoff
+    assert s.value == 512.0
+    for _ in range(3):
+        assert s.update(found_inf=False)
+    assert s.value == 1024.0                     # grew after interval
+
+
+def test_fp16_overflow_skips_step():
+    """With the scaler forced huge, gradients overflow fp32 -> the update
+    is skipped, the scale backs off, params stay put; a sane scale then
+    trains normally. (CPU runs fp32 compute; the scaler logic is the
+    same one the fp16 GPU path uses.)"""
+    import torch
+
+    from edl_amd.train.engine import DynamicLossScaler, TrainerEngine
+
+    eng = TrainerEngine(model="resnet18_vd", per_device_batch=2, dtype="fp32",
+                        channels_last=False, use_hip_ops=False,
+                        graph_capture=False, num_classes=4).setup()
+    eng.scaler = DynamicLossScaler(init_scale=1e38, growth_interval=10)
+    x = torch.randn(2, 3, 32, 32)
+    y = torch.randint(0, 4, (2,))
+    p0 = [p.detach().clone() for p in eng.model.parameters()]
+    eng.train_step(x, y)
+    assert eng.scaler.value < 1e38               # backed off
+    for p, before in zip(eng.model.parameters(), p0):
+        assert torch.equal(p.detach(), before)   # step skipped
+
+    eng.scaler.value = 128.0
+    eng.train_step(x, y)
+    changed = any(not torch.equal(p.detach(), before)
+                  for p, before in zip(eng.model.parameters(), p0))
+    assert changed                               # normal scaled step applied
