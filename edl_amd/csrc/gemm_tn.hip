@@ -17,9 +17,9 @@
 //     tiles are [BK=64 k-rows x BN n-cols] slabs of A and B;
 //   * MFMA a/b fragments want 8 CONSECUTIVE k per lane at fixed n — in a
 //     [k][n] tile that is a column walk, so each fragment is assembled
-//     with 8 swizzled ds_read_u16 instead of one ds_read_b128. The
-//     extended swizzle (slot ^ (k & (SLOTS-1))) spreads the 4 k-groups of
-//     a fragment over distinct bank sets where the slot count allows.
+//     with 8 swizzled ds_read_u16 instead of one ds_read_b128. The row
+//     swizzle (tn_swz: slot ^ (k&7), plus slot bit 2 XOR k bit 3) spreads
+//     a fragment's k-groups over distinct 32-dword bank sets.
 //     Wgrad GEMMs are HBM-bound by ~6x (e.g. M=100352, 256x64 out:
 //     64 MB read vs 3.3 GFLOP), so the extra LDS traffic sits inside the
 //     HBM shadow — measured end to end before enabling by default.
@@ -126,7 +126,7 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_kernel(
 
   // ---- staging: group's GW wave(s) stage one [BK x BN] slab pair ----
   // chunk = 1 KiB = (64/SLOTS) k-rows x SLOTS slots; lane-linear dest,
-  // swizzled source (gslot = slot ^ (r & (SLOTS-1)))
+  // swizzled source (gslot = slot ^ tn_swz(r))
   auto stage = [&](int buf, int ct) {
     const long long k0 = (long long)ct * BK;
     char* abase = gbase + buf * AB;

@@ -52,6 +52,9 @@ def parse_args(argv=None):
     p.add_argument("--checkpoint_dir", default=None)
     p.add_argument("--standalone", action="store_true",
                    help="run an in-process coordination store")
+    p.add_argument("--store_snapshot", default=None,
+                   help="standalone store: persist state here so a "
+                        "restarted agent resumes the same job keyspace")
     p.add_argument("cmd", nargs=argparse.REMAINDER, help="training script and args")
     args = p.parse_args(argv)
     if args.cmd and args.cmd[0] == "--":
@@ -208,7 +211,7 @@ def main(argv=None):
     if args.standalone:
         from .coord.server import CoordServer
 
-        server = CoordServer(port=0).start()
+        server = CoordServer(port=0, snapshot=args.store_snapshot).start()
         args.store_endpoints = server.endpoint
         log.info("standalone coordination store at %s", server.endpoint)
 
