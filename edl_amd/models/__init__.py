@@ -10,6 +10,8 @@ from .resnet_vd import (
 from .resnext_wsl import resnext101_32x16d_wsl
 from .linear import FitALine
 from .ctr import WideAndDeep
+from .mnist import MnistCNN, MnistMLP, MnistSoftmaxRegression
+from .text import TextBOW, TextCNN
 
 
 def build_model(name, num_classes=1000):
@@ -22,7 +24,15 @@ def build_model(name, num_classes=1000):
         "resnet152_vd": resnet152_vd,
         "resnet200_vd": resnet200_vd,
         "resnext101_32x16d_wsl": resnext101_32x16d_wsl,
+        # distill example families (reference example/distill/mnist_distill
+        # nn_type selector + train_with_fleet.py:55-95)
+        "mnist_cnn": MnistCNN,
+        "mnist_mlp": MnistMLP,
+        "mnist_softmax": MnistSoftmaxRegression,
     }
     if name not in table:
         raise ValueError("unknown model %r (have %s)" % (name, sorted(table)))
-    return table[name](num_classes=num_classes)
+    kwargs = {"num_classes": num_classes}
+    if name.startswith("mnist") and num_classes == 1000:
+        kwargs["num_classes"] = 10
+    return table[name](**kwargs)
