@@ -45,7 +45,10 @@ class TeacherService:
         x = torch.from_numpy(np.ascontiguousarray(images)).to(self.device)
         with self._lock:
             if self.use_bf16:
-                x = x.contiguous(memory_format=torch.channels_last)
+                # channels_last only applies to 4-D image batches; text
+                # teachers feed integer token ids [B, T]
+                if x.dim() == 4:
+                    x = x.contiguous(memory_format=torch.channels_last)
                 with torch.autocast("cuda", torch.bfloat16):
                     y = self.model(x)
             else:

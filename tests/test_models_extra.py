@@ -72,3 +72,53 @@ def test_text_kd_student_learns_teacher():
     # student's hard predictions match the teacher's on the training set
     agree = (student(ids).argmax(1) == t_logits.argmax(1)).float().mean()
     assert agree > 0.9, float(agree)
+
+
+def test_text_distill_pipeline_real_teacher():
+    """NLP distill end-to-end on CPU: a served TextBOW teacher feeds soft
+    labels through the full DistillReader pipeline (reference
+    example/distill/nlp/distill.py flow) and a token-CNN student trains
+    one pass against them."""
+    import numpy as np
+
+    from edl_amd.distill.reader import DistillReader
+    from edl_amd.distill.teacher_server import TeacherServer, TeacherService
+
+    torch.manual_seed(3)
+    teacher = TextBOW(vocab_size=32, num_classes=2)
+    srv = TeacherServer(TeacherService(model=teacher,
+                                       device=torch.device("cpu")),
+                        host="127.0.0.1", port=0).start()
+    try:
+        rng = np.random.RandomState(0)
+        n = 24
+
+        def gen():
+            for i in range(n):
+                yield (rng.randint(1, 32, size=(6,)).astype(np.int64),
+                       np.int64(i % 2))
+
+        dr = DistillReader(ins=["ids", "y"], predicts=["logits"],
+                           teacher_batch_size=4, require_num=1)
+        dr.set_sample_generator(gen)
+        dr.set_fixed_teacher(["127.0.0.1:%d" % srv.port])
+        student = TextCNN(vocab_size=32, num_classes=2)
+        opt = torch.optim.SGD(student.parameters(), lr=0.1)
+        seen = 0
+        batch = []
+        for ids, y, logits in dr():
+            ref = teacher(torch.from_numpy(np.ascontiguousarray(ids))
+                          .unsqueeze(0)).detach().numpy()[0]
+            assert np.allclose(logits, ref, atol=1e-4)  # served == local
+            batch.append((ids, logits))
+            seen += 1
+            if len(batch) == 8:
+                x = torch.from_numpy(np.stack([b[0] for b in batch]))
+                t = torch.from_numpy(np.stack([b[1] for b in batch]))
+                opt.zero_grad()
+                kd_soft_cross_entropy(student(x), t * 4.0).backward()
+                opt.step()
+                batch = []
+        assert seen == n
+    finally:
+        srv.stop()
