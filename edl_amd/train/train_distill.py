@@ -44,6 +44,10 @@ def parse_args(argv=None):
     p.add_argument("--require_num", type=int, default=2)
     p.add_argument("--checkpoint", default=None)
     p.add_argument("--dtype", default="bf16")
+    p.add_argument("--image_shape", default="3x224x224",
+                   help="CxHxW of the synthetic student input (e.g. "
+                        "1x28x28 for the mnist_* students)")
+    p.add_argument("--num_classes", type=int, default=1000)
     return p.parse_args(argv)
 
 
@@ -59,11 +63,15 @@ def main(argv=None):
         checkpoint_dir=args.checkpoint,
         use_hip_ops=torch.cuda.is_available(),
         kd_alpha=args.kd_alpha,
+        num_classes=args.num_classes,
     ).setup(tenv)
     engine.model.train()
 
+    shape = tuple(int(d) for d in args.image_shape.split("x"))
     loader = SyntheticImageNet(args.batch_size, torch.device("cpu"),
-                               seed=1234 + engine.env.global_rank, pool=8)
+                               seed=1234 + engine.env.global_rank, pool=8,
+                               image_shape=shape,
+                               num_classes=args.num_classes)
 
     def batch_gen():
         for _ in range(args.steps_per_epoch):

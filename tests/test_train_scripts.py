@@ -161,3 +161,29 @@ def test_resnet_elastic_data_plane(coord_server, tmp_path):
     counts = [int(m) for m in re.findall(r"rank \d+: (\d+) records", logs)]
     assert len(counts) == 2 and sum(counts) == 80, counts
     assert "epoch 0 done" in logs
+
+
+def test_train_distill_mnist_cpu():
+    """mnist distill config (reference example/distill/mnist_distill):
+    mnist_cnn student distilled from a served mnist_mlp teacher through
+    the real train_distill CLI at world 1 on CPU."""
+    import torch
+
+    from edl_amd.distill.teacher_server import TeacherServer, TeacherService
+    from edl_amd.models import build_model
+    from edl_amd.train import train_distill
+
+    teacher = TeacherService(model=build_model("mnist_mlp"),
+                             device=torch.device("cpu"))
+    srv = TeacherServer(teacher, host="127.0.0.1", port=0).start()
+    try:
+        rc = train_distill.main([
+            "--model", "mnist_cnn", "--image_shape", "1x28x28",
+            "--num_classes", "10", "--batch_size", "8",
+            "--steps_per_epoch", "4", "--num_epochs", "1",
+            "--teachers", "127.0.0.1:%d" % srv.port, "--require_num", "1",
+            "--kd_alpha", "0.7",
+        ])
+        assert rc == 0
+    finally:
+        srv.stop()
