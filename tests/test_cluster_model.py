@@ -70,3 +70,40 @@ def test_save_load_cluster_guarded(coord_client):
     c2 = Cluster(pods=[make_pod("z")])
     assert not save_cluster(coord_client, c2, leader_guard=("/test_job/rank/nodes/0", "zzz"))
     assert load_cluster(coord_client).stage == c.stage
+
+
+def test_trainer_env_contract():
+    """The full env contract a spawned trainer sees (reference
+    utils/train_process.py:46-73 PADDLE_* names + torch env:// vars), and
+    that TrainerEnv reads it back to the same identity."""
+    from edl_amd.train.env import TrainerEnv, trainer_env_dict
+
+    job = JobEnv({"job_id": "jctr", "store_endpoints": "127.0.0.1:2379",
+                  "nproc_per_node": 2},
+                 env={"CUDA_VISIBLE_DEVICES": ""})
+    c = Cluster(pods=[make_pod("a"), make_pod("b")])
+    c.assign_ranks()
+    pod = c.pods[1]
+    tr = pod.trainers[1]  # global rank 3
+    e = trainer_env_dict(job, c, pod, tr)
+
+    assert e["PADDLE_JOB_ID"] == "jctr"
+    assert e["PADDLE_TRAINER_ID"] == e["RANK"] == "3"
+    assert e["PADDLE_TRAINER_RANK_IN_POD"] == e["LOCAL_RANK"] == "1"
+    assert e["PADDLE_TRAINERS_NUM"] == e["WORLD_SIZE"] == "4"
+    assert e["PADDLE_TRAINER_ENDPOINTS"].count(",") == 3
+    assert e["PADDLE_CURRENT_ENDPOINT"] == tr.endpoint
+    assert e["FLAGS_selected_gpus"] == "1"
+    assert e["PADDLE_ETCD_ENDPOINTS"] == "127.0.0.1:2379"
+    assert e["EDL_CLUSTER_STAGE"] == c.stage
+    assert e["EDL_POD_ID"] == "b"
+    # MASTER_* = global rank 0's endpoint
+    host, port = c.trainer_endpoints()[0].rsplit(":", 1)
+    assert e["MASTER_PORT"] == port
+
+    tenv = TrainerEnv(env=e)
+    assert tenv.global_rank == 3 and tenv.rank_in_pod == 1
+    assert tenv.world_size == 4
+    assert tenv.current_endpoint == tr.endpoint
+    assert tenv.cluster_stage == c.stage
+    assert tenv.master_port == int(port)
