@@ -164,3 +164,21 @@ def test_nop_pipeline_epoch_soak(nop_teacher):
     for _ in range(40):
         ys = [int(y) for _, y, _ in dr()]
         assert ys == list(range(21))
+
+
+def test_timeline_gating(monkeypatch):
+    """timeline() is a no-op unless DISTILL_READER_PROFILE=1 (reference
+    distill/timeline.py:45-46 env gate)."""
+    from edl_amd.distill.timeline import timeline
+
+    monkeypatch.delenv("DISTILL_READER_PROFILE", raising=False)
+    t = timeline("reader")
+    assert type(t).__name__ == "_NopTimeLine"
+    with t("noop"):
+        pass
+
+    monkeypatch.setenv("DISTILL_READER_PROFILE", "1")
+    t = timeline("reader")
+    assert type(t).__name__ == "_RealTimeLine"
+    with t("phase_a"):
+        pass
