@@ -42,7 +42,10 @@ class TeacherService:
     @torch.no_grad()
     def predict(self, images):
         """images: numpy [B, 3, H, W] float32 -> logits numpy [B, classes]."""
-        x = torch.from_numpy(np.ascontiguousarray(images)).to(self.device)
+        arr = np.ascontiguousarray(images)
+        if not arr.flags.writeable:  # proto recv buffers are read-only views
+            arr = arr.copy()
+        x = torch.from_numpy(arr).to(self.device)
         with self._lock:
             if self.use_bf16:
                 # channels_last only applies to 4-D image batches; text

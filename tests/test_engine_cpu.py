@@ -203,7 +203,8 @@ def test_bucketed_allreduce_world2_matches_fullbatch(tmp_path):
         loss = (((m(x0) - y0) ** 2).mean() + ((m(x1) - y1) ** 2).mean()) / 2
         loss.backward()
         opt.step()
-    ref = float(sum(p.sum() for p in m.parameters()))
+    with torch.no_grad():
+        ref = float(sum(p.sum() for p in m.parameters()))
     assert res["params_sum"] == pytest.approx(ref, abs=1e-4)
 
 

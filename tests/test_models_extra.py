@@ -67,7 +67,7 @@ def test_text_kd_student_learns_teacher():
         loss = kd_soft_cross_entropy(student(ids), t_logits)
         loss.backward()
         opt.step()
-        losses.append(float(loss))
+        losses.append(float(loss.detach()))
     assert losses[-1] < losses[0] * 0.6, (losses[0], losses[-1])
     # student's hard predictions match the teacher's on the training set
     agree = (student(ids).argmax(1) == t_logits.argmax(1)).float().mean()
