@@ -59,11 +59,18 @@ class KubectlClient(ClusterClient):
                   "--replicas", str(replicas))
 
 
-def reconcile(client, free_slots=None):
-    """One reconcile pass. Policy (reference doc/usage.md autoscaler):
-    give every job its min; distribute remaining capacity toward max,
-    fair-share. free_slots=None means capacity-unconstrained."""
+def reconcile(client, free_slots=None, total_slots=None,
+              max_load_desired=0.9):
+    """One reconcile pass. Policy (reference doc/usage.md autoscaler,
+    edl_controller.yaml `-max_load_desired 0.9`): give every job its min;
+    distribute remaining capacity toward max, fair-share; when the
+    cluster capacity is known (total_slots), allocate at most
+    floor(total_slots * max_load_desired) so the cluster keeps headroom
+    for non-training workloads. free_slots/total_slots=None means
+    capacity-unconstrained."""
     jobs = client.list_training_jobs()
+    if total_slots is not None:
+        free_slots = int(total_slots * max_load_desired)
     actions = []
     want = {}
     for j in jobs:
@@ -98,11 +105,16 @@ def main(argv=None):
     ap.add_argument("--namespace", default="default")
     ap.add_argument("--period", type=float, default=10.0)
     ap.add_argument("--free_slots", type=int, default=None)
+    ap.add_argument("--total_slots", type=int, default=None,
+                    help="cluster trainer-slot capacity; allocation is "
+                         "capped at total_slots * max_load_desired")
+    ap.add_argument("--max_load_desired", type=float, default=0.9)
     args = ap.parse_args(argv)
     client = KubectlClient(args.namespace)
     while True:
         try:
-            reconcile(client, args.free_slots)
+            reconcile(client, args.free_slots, args.total_slots,
+                      args.max_load_desired)
         except Exception as e:  # noqa: BLE001
             log.warning("reconcile error: %s", e)
         time.sleep(args.period)

@@ -44,3 +44,13 @@ def test_fair_share_under_capacity():
 def test_noop_when_at_target():
     fc = FakeCluster([job("a", 1, 2, 2)])
     assert reconcile(fc) == []
+
+
+def test_max_load_desired_caps_allocation():
+    """Reference `-max_load_desired 0.9`: with capacity known, never
+    allocate past the load target (headroom for other workloads)."""
+    c = FakeCluster([job("a", 1, 8, 1), job("b", 1, 8, 1)])
+    reconcile(c, total_slots=10, max_load_desired=0.9)  # budget = 9
+    total = sum(j["running"] for j in c.jobs)
+    assert total == 9, c.jobs
+    assert all(j["running"] >= 1 for j in c.jobs)
