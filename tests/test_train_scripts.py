@@ -187,3 +187,24 @@ def test_train_distill_mnist_cpu():
         assert rc == 0
     finally:
         srv.stop()
+
+
+def test_bench_distill_service_smoke():
+    """Service-distill orchestration (BASELINE config 4 shape) on CPU:
+    one teacher process + two gloo student ranks, one JSON line out."""
+    import json as _json
+
+    out = subprocess.run(
+        [sys.executable, os.path.join(REPO, "tools", "bench_distill.py"),
+         "--teacher_gpus", "", "--student_gpus", "",
+         "--steps", "2", "--warmup", "1", "--batch_size", "2",
+         "--teacher_batch_size", "2", "--teacher_model", "resnet18_vd",
+         "--student_model", "resnet18_vd"],
+        capture_output=True, text=True, timeout=280,
+        env=dict(os.environ, CUDA_VISIBLE_DEVICES=""))
+    assert out.returncode == 0, out.stdout + out.stderr
+    line = [ln for ln in out.stdout.splitlines()
+            if ln.startswith("{")][-1]
+    d = _json.loads(line)
+    assert d["mode"] == "distill_service" and d["n_students"] == 2
+    assert d["value"] > 0

@@ -8,7 +8,11 @@ Configs (reference README.md:84-85):
 Single-GPU form (gpurun): both processes share cuda:0.
 
     python tools/bench_distill.py --steps 20 --batch_size 32
-Prints one JSON line with img/s (whole job).
+Service form (BASELINE config 4, 4 teacher + 4 student GPUs):
+    python tools/bench_distill.py --teacher_gpus 0,1,2,3 --student_gpus 4,5,6,7
+(teachers run as one process per GPU; students as one torchrun rank per
+GPU; on a no-GPU box both sides fall back to CPU/gloo — the CPU smoke
+path the tests use). Prints one JSON line with img/s (whole job).
 """
 import argparse
 import json
@@ -30,7 +34,15 @@ def main():
     ap.add_argument("--teacher_batch_size", type=int, default=16)
     ap.add_argument("--require_num", type=int, default=1)
     ap.add_argument("--teacher_model", default="resnext101_32x16d_wsl")
+    ap.add_argument("--student_model", default="resnet50_vd")
+    ap.add_argument("--teacher_gpus", default=None,
+                    help="comma list: run the SERVICE config — one "
+                         "teacher process per listed GPU")
+    ap.add_argument("--student_gpus", default=None,
+                    help="comma list: torchrun one student rank per GPU")
     args = ap.parse_args()
+    if args.teacher_gpus is not None or args.student_gpus is not None:
+        return run_service(args)
 
     import torch
 
@@ -88,5 +100,73 @@ def main():
     srv.stop()
 
 
+def run_service(args):
+    """BASELINE config 4: teachers on their own GPUs (one server process
+    each), students data-parallel on the rest (reference EDL service
+    distill, README.md:84-85: 1514 img/s on 4+4 V100)."""
+    import shlex
+    import subprocess
+
+    import torch
+
+    from edl_amd.utils.net import find_free_port
+
+    tg = [g for g in (args.teacher_gpus or "").split(",") if g != ""]
+    sg = [g for g in (args.student_gpus or "").split(",") if g != ""]
+    cuda = torch.cuda.is_available()
+    if not tg:
+        tg = [""]  # one CPU teacher (smoke form)
+    if not sg:
+        sg = ["", ""]
+    ports = find_free_port(len(tg))
+    if len(tg) == 1:
+        ports = [ports]
+    procs = []
+    env0 = dict(os.environ, PYTHONPATH=REPO + os.pathsep +
+                os.environ.get("PYTHONPATH", ""))
+    for g, port in zip(tg, ports):
+        env = dict(env0, CUDA_VISIBLE_DEVICES=str(g) if cuda else "")
+        procs.append(subprocess.Popen(
+            [sys.executable, "-m", "edl_amd.distill.teacher_server",
+             "--model", args.teacher_model, "--host", "127.0.0.1",
+             "--port", str(port)], env=env))
+    teachers = ",".join("127.0.0.1:%d" % p for p in ports)
+    # wait for teachers to accept
+    from edl_amd.distill.registry import is_server_alive
+
+    deadline = time.monotonic() + 240
+    while time.monotonic() < deadline:
+        if all(is_server_alive("127.0.0.1:%d" % p) for p in ports):
+            break
+        time.sleep(1.0)
+    else:
+        raise RuntimeError("teachers did not come up")
+
+    env = dict(env0, CUDA_VISIBLE_DEVICES=",".join(sg) if cuda else "")
+    if not cuda:
+        env["EDL_FORCE_BACKEND"] = "gloo"
+    cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+           "--nproc-per-node", str(len(sg)), "--master-addr", "127.0.0.1",
+           "--master-port", str(find_free_port()),
+           os.path.join(REPO, "tools", "bench_distill_worker.py"),
+           "--steps", str(args.steps), "--warmup", str(args.warmup),
+           "--batch_size", str(args.batch_size),
+           "--teacher_batch_size", str(args.teacher_batch_size),
+           "--require_num", str(args.require_num),
+           "--student_model", args.student_model,
+           "--teachers", teachers]
+    try:
+        rc = subprocess.run(cmd, env=env).returncode
+    finally:
+        for p in procs:
+            p.terminate()
+        for p in procs:
+            try:
+                p.wait(timeout=10)
+            except subprocess.TimeoutExpired:
+                p.kill()
+    return rc
+
+
 if __name__ == "__main__":
-    main()
+    sys.exit(main() or 0)
