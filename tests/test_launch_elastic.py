@@ -257,3 +257,33 @@ def test_elastic_leader_kill(coord_server, tmp_path, agent_reaper):
     assert any(r["world"] == 1 for r in runs), runs
     assert load_job_status(c) == Status.SUCCEED
     c.close()
+
+
+def test_standalone_snapshot_job_already_succeeded(tmp_path):
+    """--standalone --store_snapshot: a restarted agent reloads the job
+    keyspace and exits early on an already-SUCCEED job (reference
+    launch.py:44-47 early exit — possible across restarts only because
+    the store state survives)."""
+    snap = str(tmp_path / "coordd.json")
+    env = dict(os.environ)
+    env.update({
+        "PYTHONPATH": REPO + os.pathsep + env.get("PYTHONPATH", ""),
+        "EDL_LEASE_TTL": "2", "EDL_LEADER_RETRY": "0.5",
+        "CUDA_VISIBLE_DEVICES": "",
+    })
+    cmd = [sys.executable, "-m", "edl_amd.launch", "--standalone",
+           "--store_snapshot", snap, "--job_id", "snapjob",
+           "--nodes_range", "1:1", "--nproc_per_node", "1",
+           "--log_dir", str(tmp_path / "logs"), FAKE]
+    r1 = subprocess.run(cmd, env=env, capture_output=True, text=True,
+                        timeout=120, cwd=REPO)
+    assert r1.returncode == 0, r1.stdout + r1.stderr
+    assert os.path.exists(snap)
+
+    t0 = time.monotonic()
+    r2 = subprocess.run(cmd, env=env, capture_output=True, text=True,
+                        timeout=120, cwd=REPO)
+    assert r2.returncode == 0, r2.stdout + r2.stderr
+    # early exit: no trainer spawned the second time
+    assert "already SUCCEED" in (r2.stdout + r2.stderr)
+    assert time.monotonic() - t0 < 30
