@@ -24,20 +24,25 @@ def _send(sock, header, payloads):
 
 
 def _recv_exact(sock, n):
-    buf = bytearray()
-    while len(buf) < n:
-        chunk = sock.recv(min(1 << 20, n - len(buf)))
-        if not chunk:
+    # bytearray + recv_into: no per-chunk concatenation, and the buffer
+    # stays WRITABLE so np.frombuffer views are writable (torch refuses
+    # read-only arrays without a copy)
+    buf = bytearray(n)
+    view = memoryview(buf)
+    got = 0
+    while got < n:
+        r = sock.recv_into(view[got:], min(1 << 20, n - got))
+        if r == 0:
             raise ConnectionError("peer closed")
-        buf.extend(chunk)
-    return bytes(buf)
+        got += r
+    return buf
 
 
 def _recv(sock):
     (n,) = _HEAD.unpack(_recv_exact(sock, 4))
     if n > MAX_FRAME:
         raise ValueError("frame too large")
-    header = json.loads(_recv_exact(sock, n).decode())
+    header = json.loads(bytes(_recv_exact(sock, n)).decode())
     payloads = []
     for _ in range(header.get("n_payloads", 0)):
         (m,) = _HEAD.unpack(_recv_exact(sock, 4))

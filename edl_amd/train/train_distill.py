@@ -107,7 +107,7 @@ def main(argv=None):
         dt = time.monotonic() - t0
         if engine.env.is_rank0:
             log.info("distill epoch %d: %d steps, %.1f img/s (whole job), loss=%.4f",
-                     epoch, steps, imgs / dt, float(loss))
+                     epoch, steps, imgs / dt, float(loss.detach()))
         engine.save_checkpoint(epoch)
     if engine.ckpt:
         engine.ckpt.wait()
