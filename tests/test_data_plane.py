@@ -117,3 +117,32 @@ def test_record_image_set_deterministic():
     assert yb[1] == 3               # wraps to equal record -> equal label
     assert torch.equal(xb[1], x1[0])  # equal record -> equal image
     assert ds1.steps() == 2
+
+
+def test_prof_summary_parses_rocpd_schema(tmp_path, capsys):
+    """tools/prof_summary.py against a synthetic rocpd sqlite db (locks
+    the rocprofv3 schema assumptions the docs cite)."""
+    import sqlite3
+    import sys
+
+    sys.path.insert(0, str(__import__("pathlib").Path(__file__).parents[1]))
+    from tools.prof_summary import main as summarize
+
+    db = tmp_path / "x_results.db"
+    con = sqlite3.connect(db)
+    con.execute("CREATE TABLE rocpd_kernel_dispatch_abc "
+                "(kernel_id INT, start INT, end INT)")
+    con.execute("CREATE TABLE rocpd_info_kernel_symbol_abc "
+                "(id INT, display_name TEXT)")
+    con.execute("INSERT INTO rocpd_info_kernel_symbol_abc VALUES (1, 'k1')")
+    con.execute("INSERT INTO rocpd_info_kernel_symbol_abc VALUES (2, 'k2')")
+    for kid, st, en in [(1, 0, 1000), (1, 2000, 4000), (2, 0, 500)]:
+        con.execute("INSERT INTO rocpd_kernel_dispatch_abc VALUES (?,?,?)",
+                    (kid, st, en))
+    con.commit()
+    con.close()
+
+    summarize(str(db))
+    out = capsys.readouterr().out
+    assert "total kernels: 3" in out
+    assert out.index("k1") < out.index("k2")  # sorted by total time
