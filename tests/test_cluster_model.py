@@ -107,3 +107,30 @@ def test_trainer_env_contract():
     assert tenv.current_endpoint == tr.endpoint
     assert tenv.cluster_stage == c.stage
     assert tenv.master_port == int(port)
+
+
+def test_master_addr_multinode_vs_local():
+    """MASTER_ADDR selection: a remote rank-0 endpoint is kept verbatim
+    (multi-node); an endpoint on THIS host's IP rendezvous on loopback
+    (container hostnames may not resolve — single-node default)."""
+    from edl_amd.train.env import trainer_env_dict
+    from edl_amd.utils.net import local_ip
+
+    job = JobEnv({"job_id": "j", "nproc_per_node": 1},
+                 env={"CUDA_VISIBLE_DEVICES": ""})
+
+    def cluster_with(host):
+        p = Pod(pod_id="a", addr=host)
+        p.trainers.append(Trainer(endpoint="%s:9100" % host, gpus=["0"],
+                                  rank_in_pod=0))
+        c = Cluster(pods=[p])
+        c.assign_ranks()
+        return c
+
+    c = cluster_with("10.9.8.7")  # not this host
+    e = trainer_env_dict(job, c, c.pods[0], c.pods[0].trainers[0])
+    assert e["MASTER_ADDR"] == "10.9.8.7"
+
+    c = cluster_with(local_ip())
+    e = trainer_env_dict(job, c, c.pods[0], c.pods[0].trainers[0])
+    assert e["MASTER_ADDR"] in ("127.0.0.1", "localhost")
