@@ -218,3 +218,31 @@ def test_client_retry_window_rides_out_restart(tmp_path):
         c.close()
         for s in replacement:
             s.stop()
+
+
+def test_snapshot_excludes_expired_leases(tmp_path):
+    """Keys whose lease already lapsed must not resurrect on restart."""
+    import time
+
+    from edl_amd.coord.client import CoordClient
+    from edl_amd.coord.server import CoordServer
+
+    snap = str(tmp_path / "c.json")
+    srv = CoordServer(port=0, snapshot=snap).start()
+    port = srv.port
+    c = CoordClient(srv.endpoint, "job")
+    lease = c.grant(0.5)
+    c.put("/job/x/nodes/ephemeral", "v", lease=lease)
+    c.put("/job/x/nodes/durable", "v")
+    time.sleep(1.2)  # lease lapses; sweeper expires + snapshots
+    c.close()
+    srv.stop()
+
+    srv2 = CoordServer(port=port, snapshot=snap).start()
+    try:
+        c2 = CoordClient(srv2.endpoint, "job")
+        assert c2.get("/job/x/nodes/ephemeral") is None
+        assert c2.get("/job/x/nodes/durable") == "v"
+        c2.close()
+    finally:
+        srv2.stop()
