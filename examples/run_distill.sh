@@ -10,10 +10,13 @@ for GPU in 0 1 2 3; do
       --port $((9292 + GPU)) --service_name resnext_teacher \
       --store_endpoints "$STORE" &
 done
-python - <<'PY' &
-from edl_amd.distill.discovery import DiscoveryServer
+STORE="$STORE" python - <<'PY' &
+import os
 import threading
-DiscoveryServer("${STORE}").start()
+
+from edl_amd.distill.discovery import DiscoveryServer
+
+DiscoveryServer(os.environ["STORE"]).start()
 threading.Event().wait()
 PY
 CUDA_VISIBLE_DEVICES=4,5,6,7 python -m edl_amd.launch \
