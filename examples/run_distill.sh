@@ -2,6 +2,8 @@
 # BASELINE config 4: ResNeXt101_32x16d_wsl teachers serving ResNet50_vd
 # students. Run teacher(s) on their GPUs, a discovery server, then students.
 set -e
+# run from anywhere: the repo root is importable
+export PYTHONPATH="$(cd "$(dirname "$0")/.." && pwd)${PYTHONPATH:+:$PYTHONPATH}"
 STORE=${STORE:-127.0.0.1:2379}
 python -m edl_amd.coord.server --port "${STORE##*:}" &
 sleep 1

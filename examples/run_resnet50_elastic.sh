@@ -2,6 +2,8 @@
 # Elastic ResNet50_vd training on one node, 1..8 GPUs (BASELINE configs 2-3).
 # Start N of these agents (e.g. in tmux panes); kill/add agents to resize.
 set -e
+# run from anywhere: the repo root is importable
+export PYTHONPATH="$(cd "$(dirname "$0")/.." && pwd)${PYTHONPATH:+:$PYTHONPATH}"
 STORE=${STORE:-127.0.0.1:2379}
 python -m edl_amd.coord.server --port "${STORE##*:}" &
 sleep 1
