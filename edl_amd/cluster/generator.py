@@ -62,24 +62,36 @@ class ClusterGenerator:
         statuses = load_pods_status(self._client)
         current = load_cluster(self._client)
 
-        # external scale-in requests (cluster/scale.py): drop named pods
+        # external scale-in requests (cluster/scale.py): drop named pods.
+        # The request is cleared only AFTER the resulting cluster publishes
+        # (or when it names no known pod): clearing first would silently
+        # drop the scale-in if we lose leadership mid-publish — the next
+        # leader must still see and apply it (reference pattern:
+        # utils/cluster_generator.py:224-250 leader-guarded txn).
         from .scale import clear_scale_request, read_scale_request
 
         req = read_scale_request(self._client)
         removed_by_request = set()
         if req:
             removed_by_request = set(req.get("remove_pods") or [])
-            if removed_by_request:
-                for pid in removed_by_request:
-                    resource.pop(pid, None)
-                clear_scale_request(self._client)
+            for pid in removed_by_request:
+                resource.pop(pid, None)
 
         if current is None:
-            return self._first_boot(resource)
+            got = self._first_boot(resource)
+            if req and got is not None:
+                clear_scale_request(self._client)
+            return got
 
         alive = [p for p in current.pods if p.pod_id in resource
                  and statuses.get(p.pod_id) != Status.FAILED]
         disappeared = len(alive) != len(current.pods)
+        # request named no current member and no registered pod -> no-op;
+        # clear it so a stale/bogus request doesn't linger forever
+        if req and not removed_by_request & (
+                {p.pod_id for p in current.pods} | set(resource)):
+            clear_scale_request(self._client)
+            req = None
 
         # candidate new pods: in resource, not in cluster, INITIAL status
         member_ids = {p.pod_id for p in current.pods}
@@ -121,8 +133,11 @@ class ClusterGenerator:
                 "published cluster stage=%s pods=%s (was %s)",
                 nxt.stage, nxt.pod_ids(), current.pod_ids(),
             )
+            if req:
+                clear_scale_request(self._client)
             return nxt
-        log.warning("lost leadership mid-publish; cluster unchanged")
+        log.warning("lost leadership mid-publish; cluster unchanged"
+                    " (scale request, if any, stays pending)")
         return current
 
     def _first_boot(self, resource):

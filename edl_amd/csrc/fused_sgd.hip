@@ -12,8 +12,12 @@
 
 extern "C" __global__ void fused_sgd_f32(
     float* __restrict__ p, const float* __restrict__ g, float* __restrict__ m,
-    const float lr, const float mu, const float wd, const float scale,
-    const long long n) {
+    const float lr_host, const float mu, const float wd, const float scale,
+    const long long n, const float* __restrict__ lr_dev) {
+  // lr comes from a device scalar when provided so a hipGraph-captured
+  // step picks up LR-schedule changes on replay (the host scalar would be
+  // frozen at capture time); one broadcast load, L2-resident.
+  const float lr = lr_dev ? *lr_dev : lr_host;
   const long long n4 = n >> 2;
   const float4* g4 = reinterpret_cast<const float4*>(g);
   float4* p4 = reinterpret_cast<float4*>(p);
@@ -49,9 +53,10 @@ extern "C" __global__ void fused_sgd_f32(
 
 extern "C" void launch_fused_sgd_f32(float* p, const float* g, float* m,
                                      float lr, float mu, float wd, float scale,
-                                     long long n, hipStream_t stream) {
+                                     long long n, const float* lr_dev,
+                                     hipStream_t stream) {
   const int block = 256;
   const int grid = elementwise_grid((n + 3) / 4, block);
   hipLaunchKernelGGL(fused_sgd_f32, dim3(grid), dim3(block), 0, stream,
-                     p, g, m, lr, mu, wd, scale, n);
+                     p, g, m, lr, mu, wd, scale, n, lr_dev);
 }

@@ -7,7 +7,8 @@
 
 // launchers from the .hip translation units
 extern "C" void launch_fused_sgd_f32(float*, const float*, float*, float, float,
-                                     float, float, long long, hipStream_t);
+                                     float, float, long long, const float*,
+                                     hipStream_t);
 extern "C" void launch_kd_ce_fwd_f32(const float*, const float*, float*, int, int,
                                      hipStream_t);
 extern "C" void launch_kd_ce_bwd_f32(const float*, const float*, float*, float, int,
@@ -66,7 +67,8 @@ hipStream_t cur_stream() {
 }
 
 void fused_sgd(torch::Tensor p, torch::Tensor g, torch::Tensor m, double lr,
-               double momentum, double weight_decay, double grad_scale) {
+               double momentum, double weight_decay, double grad_scale,
+               c10::optional<torch::Tensor> lr_dev) {
   TORCH_CHECK(p.is_cuda() && g.is_cuda() && m.is_cuda(), "fused_sgd: GPU tensors required");
   TORCH_CHECK(p.scalar_type() == torch::kFloat32 && g.scalar_type() == torch::kFloat32 &&
                   m.scalar_type() == torch::kFloat32,
@@ -74,9 +76,16 @@ void fused_sgd(torch::Tensor p, torch::Tensor g, torch::Tensor m, double lr,
   TORCH_CHECK(p.is_contiguous() && g.is_contiguous() && m.is_contiguous(),
               "fused_sgd: contiguous flat buffers required");
   TORCH_CHECK(p.numel() == g.numel() && p.numel() == m.numel(), "fused_sgd: size mismatch");
+  const float* lrp = nullptr;
+  if (lr_dev.has_value()) {
+    TORCH_CHECK(lr_dev->is_cuda() && lr_dev->scalar_type() == torch::kFloat32 &&
+                    lr_dev->numel() == 1,
+                "fused_sgd: lr_dev must be a fp32 GPU scalar");
+    lrp = lr_dev->data_ptr<float>();
+  }
   launch_fused_sgd_f32(p.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr<float>(),
                        (float)lr, (float)momentum, (float)weight_decay,
-                       (float)grad_scale, (long long)p.numel(), cur_stream());
+                       (float)grad_scale, (long long)p.numel(), lrp, cur_stream());
 }
 
 torch::Tensor kd_ce_forward(torch::Tensor s, torch::Tensor t) {
@@ -482,7 +491,10 @@ torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w3, int64_t stride) {
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_sgd", &fused_sgd,
-        "fused flat momentum-SGD update (p,g,m flat fp32; folds grad_scale)");
+        "fused flat momentum-SGD update (p,g,m flat fp32; folds grad_scale)",
+        py::arg("p"), py::arg("g"), py::arg("m"), py::arg("lr"),
+        py::arg("momentum"), py::arg("weight_decay"), py::arg("grad_scale"),
+        py::arg("lr_dev") = py::none());
   m.def("kd_ce_forward", &kd_ce_forward, "KD soft-label CE forward -> per-row loss");
   m.def("kd_ce_backward", &kd_ce_backward, "KD soft-label CE backward -> dlogits");
   m.def("bn_fwd_train", &bn_fwd_train,

@@ -17,6 +17,7 @@ ONE store-backed implementation covers both APIs):
     assignment reads (90-109)."""
 import json
 import threading
+import time as _time
 import uuid
 
 from ..coord.client import CoordClient
@@ -161,3 +162,77 @@ class DiscoveryClient:
         except EdlStoreError:
             pass
         self._client.close()
+
+
+class RedisFlavorClient:
+    """Drop-in for the reference's redis-flavor balance client
+    (distill/redis/client.py:24-147): same constructor shape
+    (endpoints, service_name, require_num, token), same surface —
+    start() -> teacher list, get_teacher_list() -> (is_update, servers),
+    get_servers(), stop() — and the same update semantics (a background
+    ~2 s heartbeat notices version bumps; is_update latches until read).
+
+    The transport is the coordination store instead of the reference's
+    epoll TCP server + redis keys: C26's registry/TTL/rebalance semantics
+    are provided by the SAME DiscoveryServer that backs the etcd-flavor
+    API (SURVEY C25/C26 — one store-backed impl, two client surfaces)."""
+
+    def __init__(self, endpoints, service_name, require_num, token=None,
+                 job_id="distill", heartbeat_s=2.0):
+        self._inner = DiscoveryClient(endpoints, service_name,
+                                      require=require_num)
+        self._heartbeat_s = heartbeat_s
+        self.teacher_list = []
+        self._is_update = False
+        self._version_seen = False
+        self._lock = threading.Lock()
+        self._stop = threading.Event()
+        self._thread = None
+
+    def _poll_once(self):
+        changed, servers = self._inner.get_servers()
+        if changed:
+            with self._lock:
+                # first assignment is the register reply, not an update
+                if self._version_seen:
+                    self._is_update = True
+                self._version_seen = True
+                self.teacher_list = list(servers)
+
+    def _heartbeat(self):
+        while not self._stop.wait(self._heartbeat_s):
+            try:
+                self._poll_once()
+            except EdlStoreError:
+                pass  # store hiccup: next heartbeat retries
+
+    def start(self, daemon=True, timeout=60.0):
+        """Register and block until the first assignment arrives (the
+        reference's _register reply); returns the teacher list."""
+        self._inner.start()
+        deadline = _time.monotonic() + timeout
+        while _time.monotonic() < deadline:
+            self._poll_once()
+            if self._version_seen:
+                break
+            _time.sleep(0.1)
+        self._thread = threading.Thread(target=self._heartbeat, daemon=daemon,
+                                        name="redis-flavor-heartbeat")
+        self._thread.start()
+        return list(self.teacher_list)
+
+    def get_teacher_list(self):
+        """-> (is_update, servers); is_update latches until read."""
+        with self._lock:
+            upd, self._is_update = self._is_update, False
+            return upd, list(self.teacher_list)
+
+    def get_servers(self):
+        with self._lock:
+            return list(self.teacher_list)
+
+    def stop(self):
+        self._stop.set()
+        if self._thread:
+            self._thread.join(timeout=5)
+        self._inner.stop()

@@ -77,6 +77,20 @@ class BNReLU2d(nn.Module):
         self.register_buffer("running_mean", torch.zeros(num_features))
         self.register_buffer("running_var", torch.ones(num_features))
         self.register_buffer("num_batches_tracked", torch.tensor(0, dtype=torch.long))
+        # batch count kept as a host int (no per-step device kernel like
+        # nn.BatchNorm2d's add_); synced into the buffer at state_dict time
+        # so checkpoints carry the real count (momentum=None consumers need
+        # it for cumulative averaging). Steps replayed inside a captured
+        # hipGraph don't re-run forward and are not counted.
+        self._batches_seen = 0
+        self.register_state_dict_pre_hook(type(self)._sync_batches_to_buffer)
+        self.register_load_state_dict_post_hook(type(self)._sync_batches_from_buffer)
+
+    def _sync_batches_to_buffer(self, *args, **kwargs):
+        self.num_batches_tracked.fill_(self._batches_seen)
+
+    def _sync_batches_from_buffer(self, *args, **kwargs):
+        self._batches_seen = int(self.num_batches_tracked.item())
 
     def _use_fused(self, x, res=None):
         import os
@@ -123,6 +137,8 @@ class BNReLU2d(nn.Module):
         return _from_2d(y2d, x.shape)
 
     def forward(self, x, res=None):
+        if self.training:
+            self._batches_seen += 1
         if self._use_fused(x, res):
             if self.training or not torch.is_grad_enabled():
                 return self._fused(x, res)

@@ -28,11 +28,26 @@ class FusedSGD:
         self._reducer = reducer
         self._buckets = None  # list of (pflat, gflat, mflat, params, offsets)
         self._use_ext = available() and torch.cuda.is_available()
+        # LR lives in a device scalar on GPU: the update kernel reads it at
+        # run time, so a hipGraph-captured step tracks set_lr() on replay
+        # instead of freezing the capture-time value (warmup/decay schedules
+        # kept working under graph capture).
+        self._lr_dev = None
+        if reducer is not None:
+            self.attach(reducer)
 
     def attach(self, reducer):
         self._reducer = reducer
         self._buckets = None
+        reducer._attached_opt = self  # rebuild() snapshots/restores momentum
         return self
+
+    def set_lr(self, lr):
+        """Set the learning rate (updates the device scalar when present)."""
+        for g in self.param_groups:
+            g["lr"] = float(lr)
+        if self._lr_dev is not None:
+            self._lr_dev.fill_(float(lr))
 
     def _materialize(self):
         if self._buckets is not None:
@@ -53,6 +68,10 @@ class FusedSGD:
             out.append({"p": b.param_flat, "g": b.buffer, "m": m,
                         "params": b.params, "offsets": offsets})
         self._buckets = out
+        if self._use_ext and out and out[0]["p"].is_cuda and self._lr_dev is None:
+            self._lr_dev = torch.full(
+                (1,), self.param_groups[0]["lr"], dtype=torch.float32,
+                device=out[0]["p"].device)
         return out
 
     @torch.no_grad()
@@ -64,7 +83,8 @@ class FusedSGD:
         lr, mu, wd = g0["lr"], g0["momentum"], g0["weight_decay"]
         for bk in self._materialize():
             if self._use_ext:
-                ext().fused_sgd(bk["p"], bk["g"], bk["m"], lr, mu, wd, self.grad_scale)
+                ext().fused_sgd(bk["p"], bk["g"], bk["m"], lr, mu, wd,
+                                self.grad_scale, self._lr_dev)
             else:
                 # torch fallback (CPU tests / bring-up): same math
                 g = bk["g"]
@@ -105,3 +125,5 @@ class FusedSGD:
             for k in ("lr", "momentum", "weight_decay"):
                 if k in sd["param_groups"][0]:
                     self.param_groups[0][k] = sd["param_groups"][0][k]
+            if self._lr_dev is not None:
+                self._lr_dev.fill_(float(self.param_groups[0]["lr"]))

@@ -105,7 +105,14 @@ class BucketedAllReducer:
 
     def rebuild(self, bucket_cap_mb):
         """Re-bucket (e.g. after an elastic resize changed the optimal
-        chunk size). Gradients are re-viewed into fresh buffers."""
+        chunk size). Gradients are re-viewed into fresh buffers. An
+        attached FusedSGD's momentum is snapshotted per-param and restored
+        into the new bucket layout, and on_rebuild callbacks (e.g. the
+        engine's bf16-mirror rebuild) fire afterwards — without this the
+        optimizer would keep updating the DEAD flat buffers and training
+        would silently stop progressing."""
+        opt = getattr(self, "_attached_opt", None)
+        opt_sd = opt.state_dict() if opt is not None else None
         for h in self._hooks:
             h.remove()
         self._hooks = []
@@ -116,6 +123,17 @@ class BucketedAllReducer:
         if self._enabled:
             for p in self._params:
                 self._hooks.append(p.register_post_accumulate_grad_hook(self._on_grad))
+        if opt is not None:
+            opt._buckets = None  # re-materialize against the new flats
+            if opt_sd is not None:
+                opt.load_state_dict(opt_sd)
+        for cb in getattr(self, "_on_rebuild", ()):
+            cb()
+
+    def register_rebuild_callback(self, cb):
+        if not hasattr(self, "_on_rebuild"):
+            self._on_rebuild = []
+        self._on_rebuild.append(cb)
 
     def _mark_direct_grads(self):
         """World 1 (no collectives, no readiness hooks needed): flag every

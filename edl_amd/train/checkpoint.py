@@ -88,6 +88,11 @@ class CheckpointManager:
             host_flat, ev2 = self._stage_to_host(flat)
             host_opt = (optimizer_state, host_flat)
             ev = ev2 or ev
+        if ev is not None:
+            # the NEXT optimizer step (compute stream) must not overwrite
+            # params/momentum while the D2H copies are still in flight —
+            # without this the non-blocking save can snapshot a torn state
+            torch.cuda.current_stream().wait_event(ev)
 
         def _write():
             if ev is not None:
@@ -108,6 +113,15 @@ class CheckpointManager:
                 }), os.path.join(tmp, "optimizer.pt"))
             with open(os.path.join(tmp, "train_status.json"), "w") as f:
                 json.dump(train_status, f)
+            # layout manifest (docs/fault_tolerance.md "paddle_edl
+            # checkpoint-format mapping"): lets tooling verify integrity
+            # without unpickling tensors
+            files = ["model.pt", "train_status.json"]
+            if host_opt is not None:
+                files.insert(1, "optimizer.pt")
+            with open(os.path.join(tmp, "checkpoint_meta.json"), "w") as f:
+                json.dump({"format": "edl_amd.v1", "version": version,
+                           "files": files, "saved_by_rank": 0}, f)
             shutil.rmtree(final, ignore_errors=True)
             os.rename(tmp, final)
             self._gc()

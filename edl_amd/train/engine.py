@@ -190,12 +190,28 @@ class TrainerEngine:
             self._resume()
         # everyone starts from rank-0's weights (fresh or resumed)
         self.reducer.broadcast_params(src=0)
+        # ... and rank-0's momentum: with non-shared checkpoint dirs the
+        # per-rank local checkpoints may disagree, and divergent momentum
+        # makes parameters drift apart permanently (params alone are synced)
+        self._broadcast_opt_state(src=0)
         from ..ops.conv import bump_weight_epoch
 
         bump_weight_epoch()  # weights changed outside an optimizer step
         self._init_bf16_mirrors()
+        # elastic re-bucketing re-homes params into new flats; the bf16
+        # compute mirrors must follow or conv kernels read dead storage
+        self.reducer.register_rebuild_callback(self._init_bf16_mirrors)
         edist.barrier(self.device)
         return self
+
+    def _broadcast_opt_state(self, src=0):
+        import torch.distributed as dist
+
+        if not (dist.is_initialized() and dist.get_world_size() > 1):
+            return
+        if hasattr(self.opt, "_materialize"):
+            for bk in self.opt._materialize():
+                dist.broadcast(bk["m"], src=src)
 
     def _init_bf16_mirrors(self):
         """Per-bucket bf16 mirror of the flat fp32 params: each conv's
@@ -246,8 +262,13 @@ class TrainerEngine:
         return piecewise_lr(lr, epoch, step_in_epoch=step_in_epoch)
 
     def set_lr(self, lr):
-        for g in self.opt.param_groups:
-            g["lr"] = lr
+        if hasattr(self.opt, "set_lr"):
+            # FusedSGD keeps LR in a device scalar the update kernel reads,
+            # so a captured graph tracks schedule changes on replay
+            self.opt.set_lr(lr)
+        else:
+            for g in self.opt.param_groups:
+                g["lr"] = lr
 
     # ---- checkpoint ----
     def _resume(self):
@@ -329,10 +350,14 @@ class TrainerEngine:
             inv_scale = 1.0
             if self.scaler is not None:
                 # buckets already hold the all-reduced (scaled) grads, so
-                # every rank sees the same values and makes the same call
-                found_inf = any(
-                    not bool(torch.isfinite(b.buffer).all())
-                    for b in self.reducer._buckets)
+                # every rank sees the same values and makes the same call.
+                # One device-side reduction over all buckets, ONE host sync
+                # (a per-bucket .all().item() loop would serialize the step
+                # with N GPU->CPU round-trips).
+                flags = torch.stack(
+                    [torch.isfinite(b.buffer).all()
+                     for b in self.reducer._buckets])
+                found_inf = not bool(flags.all().item())
                 if not self.scaler.update(found_inf):
                     self.global_step += 1
                     return loss  # overflow: skip the update, scale backed off

@@ -75,3 +75,34 @@ def test_state_roundtrip_and_adjust(coord_client):
     coord_client.put("/test_job/rank/nodes/0", "leader")
     assert save_state(coord_client, st, guard=("/test_job/rank/nodes/0", "leader"))
     assert not save_state(coord_client, st, guard=("/test_job/rank/nodes/0", "bogus"))
+
+
+def test_paddle_edl_layout_contract(tmp_path):
+    """Asserts the paddle_edl checkpoint-format mapping documented in
+    docs/fault_tolerance.md (reference doc/fault_tolerance.md:20-62):
+    incrementing `checkpoint.<N>` version dirs, atomic temp-then-rename
+    (no .tmp residue), train_status.json with epoch_no, a manifest, and
+    TrainStatus(pass_id).next() resume semantics via the fleet facade."""
+    import json
+
+    from edl_amd.train.fleet import TrainStatus
+
+    cm = CheckpointManager(str(tmp_path), keep=3)
+    for epoch in range(2):
+        v = cm.save({"w": torch.ones(2)}, {"epoch_no": epoch}, blocking=True)
+        assert v == epoch  # incrementing version numbers
+    names = sorted(os.listdir(tmp_path))
+    assert names == ["checkpoint.0", "checkpoint.1"]  # no .tmp residue
+    d = tmp_path / "checkpoint.1"
+    assert sorted(os.listdir(d)) == [
+        "checkpoint_meta.json", "model.pt", "train_status.json"]
+    meta = json.load(open(d / "checkpoint_meta.json"))
+    assert meta["format"] == "edl_amd.v1" and meta["version"] == 1
+    assert meta["saved_by_rank"] == 0
+    ts = json.load(open(d / "train_status.json"))
+    assert ts["epoch_no"] == 1
+    # resume flow: TrainStatus(pass_id).next() is the first epoch to run
+    status = TrainStatus(ts["epoch_no"])
+    assert status.next() == 2
+    # fresh-start sentinel matches the reference (-1 -> start at 0)
+    assert TrainStatus().next() == 0

@@ -182,3 +182,47 @@ def test_timeline_gating(monkeypatch):
     assert type(t).__name__ == "_RealTimeLine"
     with t("phase_a"):
         pass
+
+
+def test_redis_flavor_client_api_compat(coord_server):
+    """C26 (redis-flavor discovery): drive the store-backed tier through
+    the reference redis Client API surface (distill/redis/client.py:24-147
+    — start() -> teacher list, get_teacher_list() -> (is_update, servers)
+    latching, get_servers(), stop()) including a version bump when a
+    teacher joins — demonstrating "one impl covers both flavors"."""
+    from edl_amd.coord.client import CoordClient
+    from edl_amd.distill.discovery import DiscoveryServer, RedisFlavorClient
+    from edl_amd.distill.registry import ServerRegister
+
+    ep = coord_server.endpoint
+    store = CoordClient(ep, "distill")
+    reg1 = ServerRegister(store, "rsvc", "10.0.0.1:9000", wait_alive=False).start()
+    ds = DiscoveryServer(ep, "distill", period=0.2).start()
+
+    cli = RedisFlavorClient([ep] if isinstance(ep, str) else ep, "rsvc",
+                            require_num=2, heartbeat_s=0.2)
+    teacher_list = cli.start(timeout=15)
+    assert teacher_list == ["10.0.0.1:9000"]
+    # register reply is not an "update" (reference: _register returns the
+    # list; only later servers_change messages set is_update)
+    upd, servers = cli.get_teacher_list()
+    assert not upd and servers == ["10.0.0.1:9000"]
+
+    # a second teacher joins -> heartbeat sees the version bump
+    reg2 = ServerRegister(store, "rsvc", "10.0.0.2:9000", wait_alive=False).start()
+    deadline = time.monotonic() + 15
+    upd = False
+    while time.monotonic() < deadline and not upd:
+        upd, servers = cli.get_teacher_list()
+        time.sleep(0.1)
+    assert upd and sorted(servers) == ["10.0.0.1:9000", "10.0.0.2:9000"]
+    # latching: consumed by the read above
+    upd2, _ = cli.get_teacher_list()
+    assert not upd2
+    assert sorted(cli.get_servers()) == ["10.0.0.1:9000", "10.0.0.2:9000"]
+
+    cli.stop()
+    ds.stop()
+    reg1.stop()
+    reg2.stop()
+    store.close()

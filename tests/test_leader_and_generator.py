@@ -198,3 +198,57 @@ def test_scale_in_request_drops_pod(coord_server, coord_client):
     reg_a.stop()
     reg_b.stop()
     cb.close()
+
+
+def test_scale_request_survives_lost_leadership(coord_server, coord_client):
+    """A pending scale-in request must NOT be consumed when the guarded
+    publish fails (leadership lost mid-publish) — clearing first would
+    silently drop the scale-in (reference guarded-txn pattern:
+    utils/cluster_generator.py:224-250)."""
+    from edl_amd.cluster.scale import read_scale_request, request_scale
+    from edl_amd.coord.client import CoordClient
+
+    pod_a, pod_b = make_pod("a"), make_pod("b")
+    reg_a = ResourceRegister(coord_client, pod_a).start()
+    cb = CoordClient(coord_server.endpoint, "test_job")
+    reg_b = ResourceRegister(cb, pod_b).start()
+    save_pod_status(coord_client, "a", Status.INITIAL)
+    save_pod_status(cb, "b", Status.INITIAL)
+    rank_key = coord_client.table_key(tables.ETCD_POD_RANK, "0")
+    coord_client.put(rank_key, "a")
+    gen = ClusterGenerator(coord_client, "a", min_nodes=1, period=0.1)
+    assert gen.generate_once().pod_ids() == ["a", "b"]
+    save_pod_status(cb, "b", Status.RUNNING)
+
+    # leadership flips to b BEFORE a's generator processes the request:
+    # the guarded publish fails and the request must stay pending
+    request_scale(coord_client, remove_pods=["b"])
+    coord_client.put(rank_key, "b")
+    c = gen.generate_once()
+    assert c.pod_ids() == ["a", "b"]  # publish refused
+    assert read_scale_request(coord_client) is not None  # NOT consumed
+
+    # a regains leadership: the still-pending request now applies
+    coord_client.put(rank_key, "a")
+    c2 = gen.generate_once()
+    assert c2.pod_ids() == ["a"]
+    assert read_scale_request(coord_client) is None  # consumed after publish
+    reg_a.stop()
+    reg_b.stop()
+    cb.close()
+
+
+def test_scale_request_noop_is_cleared(coord_server, coord_client):
+    """A request naming no known pod is a no-op and must not linger."""
+    from edl_amd.cluster.scale import read_scale_request, request_scale
+
+    pod_a = make_pod("a")
+    reg_a = ResourceRegister(coord_client, pod_a).start()
+    save_pod_status(coord_client, "a", Status.INITIAL)
+    coord_client.put(coord_client.table_key(tables.ETCD_POD_RANK, "0"), "a")
+    gen = ClusterGenerator(coord_client, "a", min_nodes=1, period=0.1)
+    assert gen.generate_once().pod_ids() == ["a"]
+    request_scale(coord_client, remove_pods=["ghost"])
+    assert gen.generate_once().pod_ids() == ["a"]
+    assert read_scale_request(coord_client) is None
+    reg_a.stop()
