@@ -51,21 +51,24 @@ extern "C" __global__ void pad_nhwc_kernel(
   (void)total;
 }
 
-// GROUPED: grouped conv with 16 in/out channels per group (ResNeXt 32x16d
-// shapes). An n-tile of 64 output channels = 4 groups = ONE contiguous
-// 64-channel input block at channel base n0; the gemm runs K = 9*64
-// against a block-diagonal zero-padded weight repack (4x MFMA work on
-// zeros, but ~MFMA rate vs MIOpen's grouped path).
+// GROUPED: grouped conv (ResNeXt 32x16d shapes, 16-128 in/out channels
+// per group). `gw` is the GEMM group width — the contiguous input-channel
+// window each 64-wide n-tile consumes, at channel base (n0/gw)*gw:
+//   cpg >= 64: gw = cpg. A 64-wide n-tile's groups share exactly their
+//     own input channels -> the dense engine runs with ZERO wasted MFMA
+//     (stage 3/4 of ResNeXt101_32x16d, ~80% of its grouped FLOPs).
+//   cpg 16/32: gw = 64 with a block-diagonal zero-padded weight repack
+//     (4x/2x MFMA on zeros, still ~MFMA rate vs MIOpen's grouped path).
 template <int BM, int BN, int WAVES_M, int WAVES_N, bool GROUPED = false>
 __global__ __launch_bounds__(256, 2) void conv3x3_kernel(
     const bf16* __restrict__ XP, const bf16* __restrict__ B,
     bf16* __restrict__ C_out, const int M, const int N, const int Cin,
     const int HW_out, const int W_out, const int Hp, const int Wp,
-    const int stride_hw) {
+    const int stride_hw, const int gw = 0) {
   constexpr int BK = 64;
   constexpr int A_BYTES = BM * BK * 2;
   constexpr int B_BYTES = BN * BK * 2;
-  const int Cin_k = GROUPED ? 64 : Cin;  // channels entering the gemm K
+  const int Cin_k = GROUPED ? gw : Cin;  // channels entering the gemm K
   const int K = 9 * Cin_k;               // gemm K
   const int cb_per_s = Cin_k >> 6;       // 64-wide channel blocks per shift
   char* lds = smem;
@@ -102,7 +105,7 @@ __global__ __launch_bounds__(256, 2) void conv3x3_kernel(
     shift_elems[s] = ((s / 3) * Wp + (s % 3)) * Cin;
   }
 
-  const int a_ch_base = GROUPED ? n0 : 0;  // group-block channel base
+  const int a_ch_base = GROUPED ? (n0 / gw) * gw : 0;  // group channel base
   auto stage = [&](int buf, int kt) {
     const int s = kt / cb_per_s;
     const int cb = kt % cb_per_s;
@@ -202,14 +205,15 @@ extern "C" void launch_pad_nhwc(const void* x, void* xp, int Nimg, int H, int W,
 extern "C" void launch_conv3x3_grouped(const void* xp, const void* w3g, void* y,
                                        int M, int Cout, int Cin, int HW_out,
                                        int W_out, int Hp, int Wp, int stride,
-                                       hipStream_t s) {
-  // BN must be 64 (= 4 groups of 16); w3g is [Cout, 9*64] block-diagonal
+                                       int gw, hipStream_t s) {
+  // BN = 64; w3g is [Cout, 9*gw] (gw = gemm group width: cpg when >= 64,
+  // else 64 with a block-diagonal repack)
   constexpr int BM = 256, BN = 64;
   const int grid = ((M + BM - 1) / BM) * (Cout / BN);
   const int lds_bytes = 2 * (BM * 64 * 2 + BN * 64 * 2);
   hipLaunchKernelGGL((conv3x3_kernel<BM, BN, 4, 1, true>), dim3(grid),
                      dim3(256), lds_bytes, s, (const bf16*)xp, (const bf16*)w3g,
-                     (bf16*)y, M, Cout, Cin, HW_out, W_out, Hp, Wp, stride);
+                     (bf16*)y, M, Cout, Cin, HW_out, W_out, Hp, Wp, stride, gw);
 }
 
 extern "C" void launch_conv3x3(const void* xp, const void* w3, void* y, int M,

@@ -39,7 +39,8 @@ extern "C" void launch_pad_nhwc(const void*, void*, int, int, int, int, int, int
 extern "C" void launch_conv3x3(const void*, const void*, void*, int, int, int,
                                int, int, int, int, int, hipStream_t);
 extern "C" void launch_conv3x3_grouped(const void*, const void*, void*, int, int,
-                                       int, int, int, int, int, int, hipStream_t);
+                                       int, int, int, int, int, int, int,
+                                       hipStream_t);
 extern "C" void launch_transpose_pad(const void*, void*, int, int, int,
                                      hipStream_t);
 extern "C" void launch_gemm_bt_splitk(const void*, const void*, float*, int, int,
@@ -259,8 +260,9 @@ torch::Tensor gemm_bt(torch::Tensor a, torch::Tensor b) {
 
 torch::Tensor conv3x3_grouped_fwd(torch::Tensor x, torch::Tensor w3g,
                                   int64_t stride) {
-  // grouped 3x3 (16 in/out channels per group): w3g [Cout, 9*64]
-  // block-diagonal repack; 4 groups per 64-wide n-tile.
+  // grouped 3x3: w3g [Cout, 9*gw] where gw = the gemm group width —
+  // channels-per-group when >= 64 (exact, zero wasted MFMA), else 64
+  // with a block-diagonal zero-padded repack (conv.py _repack_w3_grouped)
   TORCH_CHECK(x.is_cuda() && x.dim() == 4 &&
                   x.scalar_type() == torch::kBFloat16 &&
                   x.is_contiguous(torch::MemoryFormat::ChannelsLast),
@@ -268,7 +270,9 @@ torch::Tensor conv3x3_grouped_fwd(torch::Tensor x, torch::Tensor w3g,
   const int Nimg = (int)x.size(0), Cin = (int)x.size(1);
   const int H = (int)x.size(2), W = (int)x.size(3);
   const int Cout = (int)w3g.size(0);
-  TORCH_CHECK(w3g.size(1) == 9 * 64 && w3g.is_contiguous(), "conv3x3g: w3g");
+  TORCH_CHECK(w3g.size(1) % 9 == 0 && w3g.is_contiguous(), "conv3x3g: w3g");
+  const int gw = (int)(w3g.size(1) / 9);
+  TORCH_CHECK(gw % 64 == 0 && Cin % gw == 0, "conv3x3g: gw % 64, Cin % gw");
   TORCH_CHECK(Cin % 64 == 0 && Cout % 64 == 0 && Cin == Cout,
               "conv3x3g: C % 64, equal in/out");
   const int Hp = H + 2, Wp = W + 2;
@@ -280,7 +284,8 @@ torch::Tensor conv3x3_grouped_fwd(torch::Tensor x, torch::Tensor w3g,
   launch_pad_nhwc(x.data_ptr(), xp.data_ptr(), Nimg, H, W, Hp, Wp, Cin, s);
   auto y = torch::empty({M, Cout}, x.options());
   launch_conv3x3_grouped(xp.data_ptr(), w3g.data_ptr(), y.data_ptr(), (int)M,
-                         Cout, Cin, Hout * Wout, Wout, Hp, Wp, (int)stride, s);
+                         Cout, Cin, Hout * Wout, Wout, Hp, Wp, (int)stride,
+                         gw, s);
   return y;
 }
 

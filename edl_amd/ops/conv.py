@@ -84,6 +84,10 @@ class _Conv1x1Hip(torch.autograd.Function):
 
 _CONV3X3 = os.environ.get("EDL_CONV3X3", "hip")
 
+# Grouped-conv routing: groups with >= this many channels-per-group run
+# the in-repo grouped kernel (teacher fwd); below it, MIOpen.
+_GROUPED_MINC = int(os.environ.get("EDL_CONV3X3_GROUPED_MINC", "32"))
+
 # Weight-derived tensors (bf16 casts, transposed/repacked layouts) are
 # immutable within one optimizer step; FusedSGD.step() bumps this epoch and
 # Conv2dFast caches per-epoch, removing ~100 small cast/copy kernels per
@@ -168,19 +172,28 @@ class _Conv3x3Hip(torch.autograd.Function):
 
 
 def _repack_w3_grouped(weight):
-    """Grouped 3x3 (16 in/out ch per group) -> block-diagonal [Cout, 9*64]:
-    the 4 groups of each 64-wide output tile share one contiguous 64-channel
-    input block; out-channel n's 16 true input channels sit at lane block
-    (n//16)%4 and the rest are zeros (4x MFMA work on zeros ~ still far
-    above MIOpen's grouped path)."""
-    co, cpg, _, _ = weight.shape  # [Cout, 16, 3, 3]
-    assert cpg == 16
+    """Grouped 3x3 weight [Cout, cpg, 3, 3] -> s-major [Cout, 9*gw] bf16
+    for the grouped implicit-GEMM kernel (conv3x3.hip GROUPED mode).
+
+    cpg >= 64 (ResNeXt101_32x16d stages 3/4 — ~80% of its grouped FLOPs):
+    gw = cpg, a plain s-major repack — each 64-wide n-tile consumes
+    exactly its own group's input channels, ZERO wasted MFMA.
+
+    cpg 16/32 (stages 1/2): gw = 64 block-diagonal — the 64//cpg groups
+    of each 64-wide output tile share one contiguous 64-channel input
+    block; out-channel n's cpg true input channels sit at lane block
+    (n//cpg) % (64//cpg), the rest are zeros (4x/2x MFMA on zeros, still
+    ~MFMA rate vs MIOpen's grouped path)."""
+    co, cpg = weight.shape[0], weight.shape[1]
+    if cpg >= 64:
+        return _repack_w3(weight.detach())
     w = weight.detach().to(torch.bfloat16)
+    gpt = 64 // cpg  # groups per 64-wide tile
     out = torch.zeros(co, 3, 3, 64, dtype=torch.bfloat16, device=w.device)
-    pos = (torch.arange(co, device=w.device) // 16) % 4
-    for p in range(4):
+    pos = (torch.arange(co, device=w.device) // cpg) % gpt
+    for p in range(gpt):
         m = pos == p
-        out[m, :, :, p * 16:(p + 1) * 16] = w[m].permute(0, 2, 3, 1)
+        out[m, :, :, p * cpg:(p + 1) * cpg] = w[m].permute(0, 2, 3, 1)
     return out.reshape(co, 9 * 64).contiguous()
 
 
@@ -245,17 +258,21 @@ class Conv2dFast(nn.Conv2d):
             return _Conv3x3Hip.apply(x, self.weight, w_bf16, w3, w3rot,
                                      self.stride[0], None)
         if (
-            # Measured SLOWER than MIOpen's grouped path on the ResNeXt
-            # teacher (distill 486 -> 415 img/s): the 4x block-diagonal
-            # zero work does not pay at these widths. Opt-in for A/B.
-            os.environ.get("EDL_CONV3X3_GROUPED") == "hip"
+            # Grouped teacher 3x3s. cpg >= _GROUPED_MINC runs the in-repo
+            # grouped implicit-GEMM kernel (cpg >= 64: exact, zero wasted
+            # MFMA — ResNeXt stage 3/4; cpg 16/32: block-diagonal with
+            # 4x/2x zero work). Default MINC=32: the r1-measured loss was
+            # the cpg=16-only 4x-waste path (distill 486 -> 415 img/s);
+            # EDL_CONV3X3_GROUPED_MINC to A/B (16 = all, 9999 = none).
+            _CONV3X3 == "hip"
             and x.is_cuda
             and available()
             and x.dtype == torch.bfloat16
             and self.kernel_size == (3, 3)
             and self.groups > 1
             and self.in_channels == self.out_channels
-            and self.in_channels // self.groups == 16
+            and self.in_channels // self.groups >= _GROUPED_MINC
+            and self.in_channels // self.groups in (16, 32, 64, 128)
             and self.in_channels % 64 == 0
             and self.padding == (1, 1)
             and self.stride[0] == self.stride[1]

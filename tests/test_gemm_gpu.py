@@ -134,18 +134,21 @@ def test_gemm_bt_splitk_numerics(M, N, K, splitk):
 
 
 @pytest.mark.parametrize("stride", [1, 2])
-def test_conv3x3_grouped_eval_numerics(stride, monkeypatch):
-    """Grouped 3x3 (ResNeXt 16ch/group) inference fast path vs F.conv2d.
-    Opt-in path (EDL_CONV3X3_GROUPED=hip): measured slower than MIOpen on
-    the teacher shapes, kept correct for future tile work."""
+@pytest.mark.parametrize("cpg", [16, 32, 64, 128])
+def test_conv3x3_grouped_eval_numerics(stride, cpg, monkeypatch):
+    """Grouped 3x3 inference fast path vs F.conv2d at every ResNeXt
+    channels-per-group width: 16/32 run the block-diagonal repack, 64/128
+    the exact per-group dense engine (zero wasted MFMA)."""
     import torch.nn.functional as F
 
+    import edl_amd.ops.conv as conv_mod
     from edl_amd.ops.conv import Conv2dFast
 
-    monkeypatch.setenv("EDL_CONV3X3_GROUPED", "hip")
+    monkeypatch.setattr(conv_mod, "_GROUPED_MINC", 16)  # route every width
 
     torch.manual_seed(5)
-    C, groups = 128, 8  # 16 ch/group
+    C = max(128, cpg * 2)
+    groups = C // cpg
     conv = Conv2dFast(C, C, 3, stride=stride, padding=1, groups=groups,
                       bias=False).cuda().to(torch.bfloat16)
     x = torch.randn(2, C, 14, 14, device="cuda").to(torch.bfloat16)
