@@ -197,7 +197,9 @@ __global__ void bn_apply_kernel(
 
 // B1: per-channel reductions s1 = sum(dy_eff), s2 = sum(dy_eff * xhat),
 // where dy_eff applies the fused ReLU mask (bitmask from the fwd apply).
-template <bool RELU>
+// ST = independent row streams (4 or 8): more outstanding loads per
+// thread for latency hiding at small grids.
+template <bool RELU, int ST = 4>
 __global__ void bn_bwd_reduce_kernel(
     const bf16* __restrict__ dy, const unsigned char* __restrict__ mask,
     const bf16* __restrict__ x, const float* __restrict__ mean,
@@ -223,10 +225,10 @@ __global__ void bn_bwd_reduce_kernel(
   }
   float s1[8] = {0}, s2[8] = {0};
   long long r = row0;
-  // 4 row streams x 3 tensors = 12 outstanding 16-B loads per thread
-  for (; r + 3 * rstride < M; r += 4 * rstride) {
+  // ST row streams x 3 tensors = up to 3*ST outstanding 16-B loads
+  for (; r + (ST - 1) * rstride < M; r += ST * rstride) {
 #pragma unroll
-    for (int u = 0; u < 2; ++u) {
+    for (int u = 0; u < ST / 2; ++u) {
       const long long r0 = r + 2 * u * rstride, r1 = r + (2 * u + 1) * rstride;
       const long long e0 = r0 * C + c0, e1 = r1 * C + c0;
       F8 g0 = load8(dy + e0), g1 = load8(dy + e1);
@@ -271,6 +273,44 @@ __global__ void bn_bwd_reduce_kernel(
   __syncthreads();
   float* out = sums + (long long)blockIdx.x * 2 * C;  // [grid, 2C] partials
   for (int i = threadIdx.x; i < 2 * C; i += blockDim.x) out[i] = lsum[i];
+}
+
+// B1b-v2: strip-parallel partial reduction — 256 threads = 64 channel
+// indices x 4 strips over the block range (4x the parallelism of v1,
+// which ran only 2C threads total: HALF A WAVE at C=64 doing the whole
+// latency-bound serial partial read).
+extern "C" __global__ void bn_bwd_finalize_v2_kernel(
+    const float* __restrict__ partial, const int nblocks,
+    float* __restrict__ sums, const int C, float* __restrict__ db_acc,
+    float* __restrict__ dg_acc) {
+  __shared__ float acc[4][64];
+  const int lane = threadIdx.x & 63;
+  const int strip = threadIdx.x >> 6;
+  const int i = blockIdx.x * 64 + lane;
+  const long long st = 2 * C;
+  float sa[8] = {0};
+  if (i < 2 * C) {
+    int b = strip;
+    for (; b + 28 < nblocks; b += 32) {
+#pragma unroll
+      for (int u = 0; u < 8; ++u)
+        sa[u] += partial[(long long)(b + 4 * u) * st + i];
+    }
+    for (; b < nblocks; b += 4) sa[0] += partial[(long long)b * st + i];
+  }
+  acc[strip][lane] = ((sa[0] + sa[1]) + (sa[2] + sa[3])) +
+                     ((sa[4] + sa[5]) + (sa[6] + sa[7]));
+  __syncthreads();
+  if (strip == 0 && i < 2 * C) {
+    const float v = (acc[0][lane] + acc[1][lane]) +
+                    (acc[2][lane] + acc[3][lane]);
+    sums[i] = v;
+    if (i < C) {
+      if (db_acc) db_acc[i] += v;
+    } else if (dg_acc) {
+      dg_acc[i - C] += v;
+    }
+  }
 }
 
 // B1b: reduce bwd partials over blocks -> sums[2C] (= [dbeta; dgamma]).
@@ -349,24 +389,38 @@ __global__ void bn_bwd_dx_kernel(
 
 // ---------------- launchers ----------------
 
-extern "C" int bn_stats_grid(long long M, int C) {
-  // cover the tensor with enough blocks for bandwidth, but keep the
-  // partial buffer (grid x 2C fp32) around <= 1 MiB
+static long long env_ll(const char* name, long long dflt) {
+  const char* e = getenv(name);
+  return e ? atoll(e) : dflt;
+}
+
+static int bn_grid_common(long long M, int C, long long cap_env) {
   const int c8 = C >> 3;
   long long rows_per_block = 256 / c8 > 0 ? 256 / c8 : 1;
   long long want = (M + rows_per_block - 1) / rows_per_block;
-  // cap bounds the finalize kernels' serial partial-read loop (they were
-  // latency-bound at ~12-14 us with 512-block partials; ~5 us at 192).
-  // EDL_BN_GRID_CAP overrides for A/B (192 = measured best tradeoff).
-  static const long long cap_env = []() {
-    const char* e = getenv("EDL_BN_GRID_CAP");
-    return e ? atoll(e) : 192LL;
-  }();
   long long cap = 131072 / (2 * (long long)C);
   if (cap > cap_env) cap = cap_env;
   if (cap < 8) cap = 8;
   long long g = want < cap ? want : cap;
   return (int)(g > 0 ? g : 1);
+}
+
+extern "C" int bn_stats_grid(long long M, int C) {
+  // cover the tensor with enough blocks for bandwidth, but keep the
+  // partial buffer (grid x 2C fp32) around <= 1 MiB.
+  // cap bounds the finalize kernels' serial partial-read loop (they were
+  // latency-bound at ~12-14 us with 512-block partials; ~5 us at 192).
+  // EDL_BN_GRID_CAP overrides for A/B (192 = measured best tradeoff).
+  return bn_grid_common(M, C, env_ll("EDL_BN_GRID_CAP", 192));
+}
+
+extern "C" int bn_bwd_grid(long long M, int C) {
+  // the BWD reduce grid is decoupled from the fwd one: with the
+  // strip-parallel finalize_v2 the partial-read cost no longer binds the
+  // grid, so the reduce can run wider for bandwidth.
+  return bn_grid_common(M, C,
+                        env_ll("EDL_BN_BWD_GRID_CAP",
+                               env_ll("EDL_BN_GRID_CAP", 192)));
 }
 
 extern "C" void launch_bn_stats(const void* x, float* partial, int grid,
@@ -389,6 +443,12 @@ extern "C" void launch_bn_finalize(const float* partial, int nblocks,
 extern "C" void launch_bn_bwd_finalize(const float* partial, int nblocks,
                                        float* sums, int C, float* db_acc,
                                        float* dg_acc, hipStream_t s) {
+  if (env_ll("EDL_BN_FIN_V2", 1)) {
+    hipLaunchKernelGGL(bn_bwd_finalize_v2_kernel, dim3((2 * C + 63) / 64),
+                       dim3(256), 0, s, partial, nblocks, sums, C, db_acc,
+                       dg_acc);
+    return;
+  }
   hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((2 * C + 255) / 256), dim3(256),
                      0, s, partial, nblocks, sums, C, db_acc, dg_acc);
 }
@@ -431,14 +491,19 @@ extern "C" void launch_bn_bwd_reduce(const void* dy, const unsigned char* mask,
                                      const float* mean, const float* invstd,
                                      float* partial, int grid, long long M, int C,
                                      bool relu, hipStream_t s) {
-  if (relu)
-    hipLaunchKernelGGL((bn_bwd_reduce_kernel<true>), dim3(grid), dim3(256), 0, s,
-                       (const bf16*)dy, mask, (const bf16*)x, mean, invstd,
-                       partial, M, C);
-  else
-    hipLaunchKernelGGL((bn_bwd_reduce_kernel<false>), dim3(grid), dim3(256), 0, s,
-                       (const bf16*)dy, nullptr, (const bf16*)x, mean, invstd,
-                       partial, M, C);
+  const bool st8 = env_ll("EDL_BN_BWD_STREAMS", 4) >= 8;
+#define RCASE(R, ST)                                                        \
+  hipLaunchKernelGGL((bn_bwd_reduce_kernel<R, ST>), dim3(grid), dim3(256), \
+                     0, s, (const bf16*)dy, (R) ? mask : nullptr,          \
+                     (const bf16*)x, mean, invstd, partial, M, C)
+  if (relu) {
+    if (st8) RCASE(true, 8);
+    else RCASE(true, 4);
+  } else {
+    if (st8) RCASE(false, 8);
+    else RCASE(false, 4);
+  }
+#undef RCASE
 }
 
 extern "C" void launch_bn_bwd_dx(const void* dy, const unsigned char* mask,

@@ -18,6 +18,7 @@ extern "C" void launch_kd_ce_fwd_bf16(const void*, const void*, float*, int, int
 extern "C" void launch_kd_ce_bwd_bf16(const void*, const void*, void*, float, int,
                                       int, hipStream_t);
 extern "C" int bn_stats_grid(long long, int);
+extern "C" int bn_bwd_grid(long long, int);
 extern "C" void launch_bn_stats(const void*, float*, int, long long, int,
                                 hipStream_t);
 extern "C" void launch_bn_finalize(const float*, int, const float*, const float*,
@@ -199,7 +200,7 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, c10::optional<torch::Tensor>
   check_bn_inputs(x, C);
   const long long M = x.numel() / C;
   auto opts = gamma.options().dtype(torch::kFloat32);
-  const int grid = bn_stats_grid(M, (int)C);
+  const int grid = bn_bwd_grid(M, (int)C);
   auto partial = torch::empty({grid, 2 * C}, opts);
   auto sums = torch::empty({2, C}, opts);
   auto dx = torch::empty_like(x);

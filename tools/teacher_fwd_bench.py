@@ -32,16 +32,18 @@ def main():
     from edl_amd.models import build_model
 
     torch.manual_seed(7)
-    m = build_model(args.model).cuda().to(torch.bfloat16)
+    # same prep as distill/teacher_server.py: fp32 params, eval,
+    # channels_last, bf16 autocast forward
+    m = build_model(args.model).cuda()
     m = m.to(memory_format=torch.channels_last).eval()
-    x = torch.randn(args.batch, 3, 224, 224, device="cuda").to(torch.bfloat16)
+    x = torch.randn(args.batch, 3, 224, 224, device="cuda")
     x = x.contiguous(memory_format=torch.channels_last)
 
     results = {}
     ref = None
     for minc in (9999, 128, 64, 32, 16):
         conv_mod._GROUPED_MINC = minc
-        with torch.no_grad():
+        with torch.no_grad(), torch.autocast("cuda", torch.bfloat16):
             for _ in range(args.warmup):
                 y = m(x)
             torch.cuda.synchronize()
