@@ -53,7 +53,8 @@ __device__ __forceinline__ void store8(bf16* p, const F8& x) {
 // atomics at all (single-stage atomics measured 17-40 us/dispatch fixed
 // cost: same-word serialization at ~88 adds/us for small C, grid x 2C
 // traffic for large C).
-extern "C" __global__ void bn_stats_kernel(
+template <int ST>
+__global__ void bn_stats_kernel(
     const bf16* __restrict__ x, float* __restrict__ partial,  // [grid, 2C]
     const long long M, const int C) {
   __shared__ float lsum[2 * 2048];
@@ -70,19 +71,20 @@ extern "C" __global__ void bn_stats_kernel(
 
   float s[8] = {0}, q[8] = {0};
   long long r = row0;
-  // 4 independent row streams: the grid is capped (finalize reads the
+  // ST independent row streams: the grid is capped (finalize reads the
   // partials serially), so per-thread in-flight bytes must cover HBM
   // latency — 1 stream measured ~1.2 TB/s, latency-bound
-  for (; r + 3 * rstride < M; r += 4 * rstride) {
-    F8 v0 = load8(x + r * C + c0);
-    F8 v1 = load8(x + (r + rstride) * C + c0);
-    F8 v2 = load8(x + (r + 2 * rstride) * C + c0);
-    F8 v3 = load8(x + (r + 3 * rstride) * C + c0);
+  for (; r + (ST - 1) * rstride < M; r += ST * rstride) {
+    F8 v[ST];
 #pragma unroll
-    for (int i = 0; i < 8; ++i) {
-      s[i] += (v0.v[i] + v1.v[i]) + (v2.v[i] + v3.v[i]);
-      q[i] = fmaf(v0.v[i], v0.v[i], fmaf(v1.v[i], v1.v[i],
-                  fmaf(v2.v[i], v2.v[i], fmaf(v3.v[i], v3.v[i], q[i]))));
+    for (int u = 0; u < ST; ++u) v[u] = load8(x + (r + u * rstride) * C + c0);
+#pragma unroll
+    for (int u = 0; u < ST; ++u) {
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        s[i] += v[u].v[i];
+        q[i] = fmaf(v[u].v[i], v[u].v[i], q[i]);
+      }
     }
   }
   for (; r < M; r += rstride) {
@@ -101,6 +103,62 @@ extern "C" __global__ void bn_stats_kernel(
   __syncthreads();
   float* out = partial + (long long)blockIdx.x * 2 * C;
   for (int i = threadIdx.x; i < 2 * C; i += blockDim.x) out[i] = lsum[i];
+}
+
+// K2-v2: strip-parallel fwd finalize — 256 threads = 64 channels x 4
+// strips over the block range (v1 ran C threads total: at C=64 a single
+// wave serially read every partial, latency-bound).
+extern "C" __global__ void bn_finalize_v2_kernel(
+    const float* __restrict__ partial, const int nblocks,
+    const float* __restrict__ gamma,
+    const float* __restrict__ beta, float* __restrict__ mean_out,
+    float* __restrict__ invstd_out, float* __restrict__ scale_out,
+    float* __restrict__ shift_out, float* __restrict__ running_mean,
+    float* __restrict__ running_var, const float momentum, const float eps,
+    const long long M, const int C) {
+  __shared__ float accs[4][64];
+  __shared__ float accq[4][64];
+  const int lane = threadIdx.x & 63;
+  const int strip = threadIdx.x >> 6;
+  const int c = blockIdx.x * 64 + lane;
+  const long long st = 2 * C;
+  float sa[4] = {0}, qa[4] = {0};
+  if (c < C) {
+    int b = strip;
+    for (; b + 12 < nblocks; b += 16) {
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        sa[u] += partial[(long long)(b + 4 * u) * st + c];
+        qa[u] += partial[(long long)(b + 4 * u) * st + C + c];
+      }
+    }
+    for (; b < nblocks; b += 4) {
+      sa[0] += partial[(long long)b * st + c];
+      qa[0] += partial[(long long)b * st + C + c];
+    }
+  }
+  accs[strip][lane] = (sa[0] + sa[1]) + (sa[2] + sa[3]);
+  accq[strip][lane] = (qa[0] + qa[1]) + (qa[2] + qa[3]);
+  __syncthreads();
+  if (strip != 0 || c >= C) return;
+  const float s = (accs[0][lane] + accs[1][lane]) +
+                  (accs[2][lane] + accs[3][lane]);
+  const float q = (accq[0][lane] + accq[1][lane]) +
+                  (accq[2][lane] + accq[3][lane]);
+  const float inv_m = 1.0f / (float)M;
+  const float mean = s * inv_m;
+  const float var = fmaxf(q * inv_m - mean * mean, 0.0f);
+  const float invstd = rsqrtf(var + eps);
+  mean_out[c] = mean;
+  invstd_out[c] = invstd;
+  const float g = gamma[c];
+  scale_out[c] = g * invstd;
+  shift_out[c] = beta[c] - g * invstd * mean;
+  if (running_mean != nullptr) {
+    const float ub = (M > 1) ? var * (float)M / (float)(M - 1) : var;
+    running_mean[c] = fmaf(momentum, mean - running_mean[c], running_mean[c]);
+    running_var[c] = fmaf(momentum, ub - running_var[c], running_var[c]);
+  }
 }
 
 // K2: reduce partials over blocks, finalize mean/invstd, update running
@@ -425,8 +483,12 @@ extern "C" int bn_bwd_grid(long long M, int C) {
 
 extern "C" void launch_bn_stats(const void* x, float* partial, int grid,
                                 long long M, int C, hipStream_t s) {
-  hipLaunchKernelGGL(bn_stats_kernel, dim3(grid), dim3(256), 0, s,
-                     (const bf16*)x, partial, M, C);
+  if (env_ll("EDL_BN_STATS_STREAMS", 4) >= 8)
+    hipLaunchKernelGGL((bn_stats_kernel<8>), dim3(grid), dim3(256), 0, s,
+                       (const bf16*)x, partial, M, C);
+  else
+    hipLaunchKernelGGL((bn_stats_kernel<4>), dim3(grid), dim3(256), 0, s,
+                       (const bf16*)x, partial, M, C);
 }
 
 extern "C" void launch_bn_finalize(const float* partial, int nblocks,
@@ -435,6 +497,12 @@ extern "C" void launch_bn_finalize(const float* partial, int nblocks,
                                    float* scale, float* shift, float* rmean,
                                    float* rvar, float momentum, float eps,
                                    long long M, int C, hipStream_t s) {
+  if (env_ll("EDL_BN_FIN_V2", 1)) {
+    hipLaunchKernelGGL(bn_finalize_v2_kernel, dim3((C + 63) / 64), dim3(256),
+                       0, s, partial, nblocks, gamma, beta, mean, invstd,
+                       scale, shift, rmean, rvar, momentum, eps, M, C);
+    return;
+  }
   hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 255) / 256), dim3(256), 0, s,
                      partial, nblocks, gamma, beta, mean, invstd, scale, shift,
                      rmean, rvar, momentum, eps, M, C);

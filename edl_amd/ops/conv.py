@@ -85,8 +85,12 @@ class _Conv1x1Hip(torch.autograd.Function):
 _CONV3X3 = os.environ.get("EDL_CONV3X3", "hip")
 
 # Grouped-conv routing: groups with >= this many channels-per-group run
-# the in-repo grouped kernel (teacher fwd); below it, MIOpen.
-_GROUPED_MINC = int(os.environ.get("EDL_CONV3X3_GROUPED_MINC", "32"))
+# the in-repo grouped kernel (teacher fwd); below it, MIOpen. Measured
+# r2 (gpurun_out/r2c3/teacher_fwd_b16.log): routing EVERYTHING (16) wins
+# — ResNeXt101 teacher fwd 13.4 -> 5.5 ms/bs16 (2.4x vs all-MIOpen);
+# even the 4x-zero cpg=16 block-diagonal beats MIOpen once the rest of
+# the net runs in-repo kernels.
+_GROUPED_MINC = int(os.environ.get("EDL_CONV3X3_GROUPED_MINC", "16"))
 
 # Weight-derived tensors (bf16 casts, transposed/repacked layouts) are
 # immutable within one optimizer step; FusedSGD.step() bumps this epoch and
