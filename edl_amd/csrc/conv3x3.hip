@@ -193,6 +193,193 @@ __global__ __launch_bounds__(256, 2) void conv3x3_kernel(
   }
 }
 
+// ---- stride-2 3x3 same-pad DGRAD via parity decomposition ----
+//
+// dx[n,h,w,:] of parity class p=(h&1,w&1) receives only taps with
+// dy≡(h+1)&1, dxx≡(w+1)&1 — classes have 1/2/2/4 taps — so each class is
+// an implicit GEMM over K = T*Cout against the zero-padded dY:
+//   dx[m, ci] = sum_t sum_co dYp[arow(m) + shift_t, co] * W[co, ci, tap_t]
+// Exact work (no zero-stuffed transposed-conv upsampling; MIOpen's
+// igemm_bwd path replaced — VERDICT r1 #4). Weights come repacked as
+// wcat [Cin, 9*Cout], classes' tap slabs contiguous (conv.py
+// _repack_w3_s2dgrad), col_base selecting the class.
+template <int BM, int BN, int WAVES_M, int WAVES_N>
+__global__ __launch_bounds__(256, 2) void conv3x3s2_dgrad_kernel(
+    const bf16* __restrict__ DYP, const bf16* __restrict__ B,
+    bf16* __restrict__ DX, const int M, const int N, const int Cout,
+    const int Wc, const int HWc, const int Hop, const int Wop, const int H,
+    const int W, const int ph, const int pw, const int T, const int col_base,
+    const int sh0, const int sh1, const int sh2, const int sh3) {
+  constexpr int BK = 64;
+  constexpr int A_BYTES = BM * BK * 2;
+  constexpr int B_BYTES = BN * BK * 2;
+  const int K = T * Cout;       // this class's gemm K
+  const int Krow = 9 * Cout;    // B row stride (all classes)
+  const int cb_per_t = Cout >> 6;
+  char* lds = smem;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+
+  const int tiles_n = N / BN;
+  const int bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  const int m0 = (bid / tiles_n) * BM;
+  const int n0 = (bid % tiles_n) * BN;
+  const int wm = (wave / WAVES_N) * 64;
+  const int wn = (wave % WAVES_N) * 64;
+
+  constexpr int A_CHUNKS = A_BYTES / 1024;
+  constexpr int B_CHUNKS = B_BYTES / 1024;
+  long long arow[A_CHUNKS / 4];
+#pragma unroll
+  for (int i = 0; i < A_CHUNKS / 4; ++i) {
+    const int ch = wave * (A_CHUNKS / 4) + i;
+    const int r = ch * 8 + (lane >> 3);
+    long long m = m0 + r;
+    if (m >= M) m = M - 1;  // clamped dup row; C-write guarded
+    const int n_img = (int)(m / HWc);
+    const int rem = (int)(m % HWc);
+    const int i_r = rem / Wc;
+    const int j_c = rem % Wc;
+    arow[i] = (((long long)n_img * Hop + i_r) * Wop + j_c) * Cout;
+  }
+  const int shift_elems[4] = {sh0, sh1, sh2, sh3};
+
+  auto stage = [&](int buf, int kt) {
+    const int s = kt / cb_per_t;
+    const int cb = kt % cb_per_t;
+    char* abase = lds + buf * (A_BYTES + B_BYTES);
+    char* bbase = abase + A_BYTES;
+#pragma unroll
+    for (int i = 0; i < A_CHUNKS / 4; ++i) {
+      const int ch = wave * (A_CHUNKS / 4) + i;
+      const int r = ch * 8 + (lane >> 3);
+      const int gslot = (lane & 7) ^ (r & 7);
+      const bf16* src = DYP + arow[i] + shift_elems[s] + cb * 64 + gslot * 8;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)(abase + ch * 1024), 16, 0, 0);
+    }
+    const int k0 = col_base + kt * BK;
+#pragma unroll
+    for (int i = 0; i < B_CHUNKS / 4; ++i) {
+      const int ch = wave * (B_CHUNKS / 4) + i;
+      const int r = ch * 8 + (lane >> 3);
+      const int gslot = (lane & 7) ^ (r & 7);
+      const bf16* src = B + (long long)(n0 + r) * Krow + k0 + gslot * 8;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)(bbase + ch * 1024), 16, 0, 0);
+    }
+  };
+
+  auto read_a = [&](int buf, int mf, int kk) -> bf16x8 {
+    const char* abase = lds + buf * (A_BYTES + B_BYTES);
+    const int r = wm + mf * 16 + (lane & 15);
+    const int c = kk * 4 + (lane >> 4);
+    return *(const __attribute__((address_space(3))) bf16x8*)(
+        (const __attribute__((address_space(3))) char*)(abase) + r * 128 +
+        ((c ^ (r & 7)) << 4));
+  };
+  auto read_b = [&](int buf, int nf, int kk) -> bf16x8 {
+    const char* bbase = lds + buf * (A_BYTES + B_BYTES) + A_BYTES;
+    const int r = wn + nf * 16 + (lane & 15);
+    const int c = kk * 4 + (lane >> 4);
+    return *(const __attribute__((address_space(3))) bf16x8*)(
+        (const __attribute__((address_space(3))) char*)(bbase) + r * 128 +
+        ((c ^ (r & 7)) << 4));
+  };
+
+  f32x4 acc[4][4] = {};
+  const int KT = K / BK;
+  stage(0, 0);
+  __syncthreads();
+  for (int kt = 0; kt < KT; ++kt) {
+    const int cur = kt & 1;
+    if (kt + 1 < KT) stage(cur ^ 1, kt + 1);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      bf16x8 a[4], b[4];
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf) a[mf] = read_a(cur, mf, kk);
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) b[nf] = read_b(cur, nf, kk);
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < 4; ++nf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[mf], b[nf], acc[mf][nf], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  const int cn = lane & 15;
+  const int r4 = (lane >> 4) * 4;
+#pragma unroll
+  for (int mf = 0; mf < 4; ++mf) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int m = m0 + wm + mf * 16 + r4 + reg;
+      if (m < M) {
+        const int n_img = m / HWc;
+        const int rem = m % HWc;
+        const int h = 2 * (rem / Wc) + ph;
+        const int w = 2 * (rem % Wc) + pw;
+        bf16* crow = DX + (((long long)n_img * H + h) * W + w) * N + n0 + wn + cn;
+#pragma unroll
+        for (int nf = 0; nf < 4; ++nf)
+          crow[nf * 16] = __float2bfloat16(acc[mf][nf][reg]);
+      }
+    }
+  }
+}
+
+extern "C" void launch_conv3x3s2_dgrad(const void* dyp, const void* wcat,
+                                       void* dx, int Nimg, int H, int W,
+                                       int Cin, int Cout, int Hop, int Wop,
+                                       hipStream_t s) {
+  // classes (ph,pw) with tap lists per the conv.py repack ordering
+  const int col_base[4] = {0, Cout, 3 * Cout, 5 * Cout};
+  const int dys[2][2] = {{1, 1}, {0, 2}};   // dys[ph][...]; ph=0 -> {1}
+  const int ndy[2] = {1, 2};
+  for (int ph = 0; ph < 2; ++ph) {
+    for (int pw = 0; pw < 2; ++pw) {
+      const int cls = ph * 2 + pw;
+      const int Hc = (H - ph + 1) / 2, Wc = (W - pw + 1) / 2;
+      const int M = Nimg * Hc * Wc;
+      if (M <= 0) continue;
+      int shifts[4] = {0, 0, 0, 0};
+      int t = 0;
+      for (int a = 0; a < ndy[ph]; ++a)
+        for (int b = 0; b < ndy[pw]; ++b) {
+          const int dy = dys[ph][a], dxx = dys[pw][b];
+          const int r = (ph + 3 - dy) / 2, c = (pw + 3 - dxx) / 2;
+          shifts[t++] = (r * Wop + c) * Cout;
+        }
+      const int T = t;
+      if (Cin % 128 == 0) {
+        constexpr int BM = 128, BN = 128;
+        const int grid = ((M + BM - 1) / BM) * (Cin / BN);
+        const int lds_bytes = 2 * (BM * 64 * 2 + BN * 64 * 2);
+        hipLaunchKernelGGL((conv3x3s2_dgrad_kernel<BM, BN, 2, 2>), dim3(grid),
+                           dim3(256), lds_bytes, s, (const bf16*)dyp,
+                           (const bf16*)wcat, (bf16*)dx, M, Cin, Cout, Wc,
+                           Hc * Wc, Hop, Wop, H, W, ph, pw, T, col_base[cls],
+                           shifts[0], shifts[1], shifts[2], shifts[3]);
+      } else {
+        constexpr int BM = 256, BN = 64;
+        const int grid = ((M + BM - 1) / BM) * (Cin / BN);
+        const int lds_bytes = 2 * (BM * 64 * 2 + BN * 64 * 2);
+        hipLaunchKernelGGL((conv3x3s2_dgrad_kernel<BM, BN, 4, 1>), dim3(grid),
+                           dim3(256), lds_bytes, s, (const bf16*)dyp,
+                           (const bf16*)wcat, (bf16*)dx, M, Cin, Cout, Wc,
+                           Hc * Wc, Hop, Wop, H, W, ph, pw, T, col_base[cls],
+                           shifts[0], shifts[1], shifts[2], shifts[3]);
+      }
+    }
+  }
+}
+
 extern "C" void launch_pad_nhwc(const void* x, void* xp, int Nimg, int H, int W,
                                 int Hp, int Wp, int C, hipStream_t s) {
   const long long per_img = (long long)Hp * Wp * (C >> 3);

@@ -42,6 +42,9 @@ extern "C" void launch_conv3x3(const void*, const void*, void*, int, int, int,
 extern "C" void launch_conv3x3_grouped(const void*, const void*, void*, int, int,
                                        int, int, int, int, int, int, int,
                                        hipStream_t);
+extern "C" void launch_conv3x3s2_dgrad(const void*, const void*, void*, int,
+                                       int, int, int, int, int, int,
+                                       hipStream_t);
 extern "C" void launch_transpose_pad(const void*, void*, int, int, int,
                                      hipStream_t);
 extern "C" void launch_gemm_bt_splitk(const void*, const void*, float*, int, int,
@@ -290,6 +293,33 @@ torch::Tensor conv3x3_grouped_fwd(torch::Tensor x, torch::Tensor w3g,
   return y;
 }
 
+torch::Tensor conv3x3s2_dgrad(torch::Tensor dy, torch::Tensor wcat,
+                              int64_t H, int64_t W) {
+  // stride-2 3x3 same-pad dgrad (parity decomposition, conv3x3.hip).
+  // dy: [N, Cout, Ho, Wo] channels_last bf16; wcat: [Cin, 9*Cout]
+  // (conv.py _repack_w3_s2dgrad). Returns dx2d [N*H*W, Cin] bf16.
+  TORCH_CHECK(dy.is_cuda() && dy.dim() == 4 &&
+                  dy.scalar_type() == torch::kBFloat16 &&
+                  dy.is_contiguous(torch::MemoryFormat::ChannelsLast),
+              "s2dgrad: 4-D channels_last bf16 dy");
+  const int Nimg = (int)dy.size(0), Cout = (int)dy.size(1);
+  const int Ho = (int)dy.size(2), Wo = (int)dy.size(3);
+  const int Cin = (int)(wcat.size(0));
+  TORCH_CHECK(wcat.is_contiguous() && wcat.size(1) == 9 * Cout,
+              "s2dgrad: wcat [Cin, 9*Cout]");
+  TORCH_CHECK(Cin % 64 == 0 && Cout % 64 == 0, "s2dgrad: C % 64");
+  TORCH_CHECK((H + 1) / 2 == Ho && (W + 1) / 2 == Wo, "s2dgrad: shape");
+  const int Hop = Ho + 2, Wop = Wo + 2;
+  auto s = cur_stream();
+  auto dyp = torch::empty({(long long)Nimg * Hop * Wop * Cout}, dy.options());
+  launch_pad_nhwc(dy.data_ptr(), dyp.data_ptr(), Nimg, Ho, Wo, Hop, Wop, Cout,
+                  s);
+  auto dx = torch::empty({(long long)Nimg * H * W, Cin}, dy.options());
+  launch_conv3x3s2_dgrad(dyp.data_ptr(), wcat.data_ptr(), dx.data_ptr(), Nimg,
+                         (int)H, (int)W, Cin, Cout, Hop, Wop, s);
+  return dx;
+}
+
 torch::Tensor avgpool2x2_fwd(torch::Tensor x) {
   // x: 4-D channels_last bf16; 2x2 stride-2 ceil_mode pool
   TORCH_CHECK(x.is_cuda() && x.dim() == 4 &&
@@ -534,6 +564,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("avgpool2x2_fwd", &avgpool2x2_fwd, "2x2/s2 ceil avg pool (NHWC bf16)");
   m.def("conv3x3_grouped_fwd", &conv3x3_grouped_fwd,
         "grouped (16ch/group) 3x3 conv fwd -> y2d [M, Cout]");
+  m.def("conv3x3s2_dgrad", &conv3x3s2_dgrad,
+        "stride-2 3x3 same-pad dgrad (parity implicit GEMM) -> dx2d [N*H*W, Cin]");
   m.def("avgpool2x2_bwd", &avgpool2x2_bwd, "2x2/s2 ceil avg pool backward");
   m.attr("_arch") = "gfx950";
 }

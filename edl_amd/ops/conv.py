@@ -124,7 +124,7 @@ class _Conv3x3Hip(torch.autograd.Function):
                 grad_tgt):
         ctx.save_for_backward(x, w_bf16)
         ctx.stride = stride
-        ctx.w3rot = w3rot_cached
+        ctx.w3rot = w3rot_cached  # s1: rotated fwd repack; s2: s2dgrad wcat
         ctx.w_dtype = w_param.dtype
         ctx.grad_tgt = grad_tgt
         y2d = ext().conv3x3_fwd(x, w3_cached, stride)
@@ -139,9 +139,14 @@ class _Conv3x3Hip(torch.autograd.Function):
         x, weight = ctx.saved_tensors
         stride = ctx.stride
         dy = dy.contiguous(memory_format=torch.channels_last)
+        n, _, h, w = x.shape
         if stride == 1:
             dx2d = ext().conv3x3_fwd(dy.to(torch.bfloat16), ctx.w3rot, 1)
-            n, _, h, w = x.shape
+            dx = dx2d.view(n, h, w, x.shape[1]).permute(0, 3, 1, 2)
+        elif ctx.w3rot is not None:
+            # stride 2: parity-decomposed implicit-GEMM dgrad (4 class
+            # launches, exact work) — replaces MIOpen's igemm_bwd
+            dx2d = ext().conv3x3s2_dgrad(dy.to(torch.bfloat16), ctx.w3rot, h, w)
             dx = dx2d.view(n, h, w, x.shape[1]).permute(0, 3, 1, 2)
         else:
             dx = torch.nn.grad.conv2d_input(
@@ -173,6 +178,22 @@ class _Conv3x3Hip(torch.autograd.Function):
         if dw.dtype != ctx.w_dtype:
             dw = dw.to(ctx.w_dtype)
         return dx, dw, None, None, None, None, None
+
+
+_S2D_TAP_ORDER = [(1, 1), (1, 0), (1, 2), (0, 1), (2, 1),
+                  (0, 0), (0, 2), (2, 0), (2, 2)]
+
+
+def _repack_w3_s2dgrad(weight):
+    """[Cout, Cin, 3, 3] -> wcat [Cin, 9*Cout] bf16 for the stride-2 dgrad
+    parity kernel (conv3x3.hip conv3x3s2_dgrad_kernel): parity classes'
+    tap slabs contiguous — class (h&1,w&1) uses taps with dy=(h+1)&1,
+    dxx=(w+1)&1; column bases [0, Co, 3Co, 5Co]."""
+    co, ci = weight.shape[0], weight.shape[1]
+    w = weight.detach()
+    slabs = torch.stack([w[:, :, dy, dx] for dy, dx in _S2D_TAP_ORDER])
+    return (slabs.permute(2, 0, 1).reshape(ci, 9 * co)
+            .to(torch.bfloat16).contiguous())
 
 
 def _repack_w3_grouped(weight):
@@ -251,10 +272,12 @@ class Conv2dFast(nn.Conv2d):
                     torch.bfloat16).contiguous())
                 wsrc = self.weight.detach()
             w3 = self._cached("w3", lambda: _repack_w3(wsrc))
-            w3rot = None
             if self.stride[0] == 1:
                 w3rot = self._cached("w3rot", lambda: _repack_w3(
                     wsrc.permute(1, 0, 2, 3).flip(2, 3)))
+            else:
+                w3rot = self._cached("w3s2d",
+                                     lambda: _repack_w3_s2dgrad(wsrc))
             # NOTE: direct-grad stays OFF for 3x3 — the [Cout,Cin,3,3]
             # remap makes the split-K epilogue atomics stride-9 scattered
             # (measured: slower end-to-end than the AccumulateGrad add it

@@ -258,3 +258,57 @@ def test_gemm_tn3x3_accumulate_out_layout():
     ref = native.view(co, 3, 3, ci).permute(0, 3, 1, 2) + 0.25
     assert torch.allclose(out, ref, atol=1e-3, rtol=1e-3), \
         (out - ref).abs().max().item()
+
+
+@pytest.mark.parametrize("shape", [(2, 128, 128, 14), (2, 256, 256, 28),
+                                   (1, 512, 512, 14), (2, 64, 128, 13)])
+def test_conv3x3_s2_dgrad_numerics(shape):
+    """Stride-2 3x3 dgrad parity kernel vs torch.nn.grad.conv2d_input
+    (fp32 reference). Covers even and odd H/W and the Cin%128 launch
+    split (replaces MIOpen's igemm_bwd — VERDICT r1 #4)."""
+    import torch.nn.functional as F
+
+    from edl_amd import ops
+    from edl_amd.ops.conv import _repack_w3_s2dgrad
+
+    n, ci, co, hw = shape
+    torch.manual_seed(11)
+    w = (torch.randn(co, ci, 3, 3, device="cuda") * 0.2)
+    x_shape = [n, ci, hw, hw]
+    ho = (hw + 1) // 2
+    dy = (torch.randn(n, co, ho, ho, device="cuda") * 0.5).to(torch.bfloat16)
+    dy = dy.contiguous(memory_format=torch.channels_last)
+
+    wcat = _repack_w3_s2dgrad(w)
+    dx2d = ops.ext().conv3x3s2_dgrad(dy, wcat, hw, hw)
+    dx = dx2d.view(n, hw, hw, ci).permute(0, 3, 1, 2)
+
+    ref = torch.nn.grad.conv2d_input(
+        x_shape, w.float(), dy.float(), stride=(2, 2), padding=(1, 1))
+    err = (dx.float() - ref).abs()
+    scale = ref.abs().mean().clamp(min=0.05)
+    assert (err / scale).max() < 0.1, (err.max().item(), scale.item())
+
+
+def test_conv3x3_s2_full_backward_matches_miopen():
+    """End-to-end Conv2dFast stride-2 backward (hip dgrad + wgrad) vs the
+    fp32 autograd reference."""
+    from edl_amd.ops.conv import Conv2dFast
+
+    torch.manual_seed(13)
+    conv = Conv2dFast(128, 128, 3, stride=2, padding=1, bias=False).cuda()
+    x = (torch.randn(2, 128, 28, 28, device="cuda") * 0.5).to(torch.bfloat16)
+    x = x.contiguous(memory_format=torch.channels_last).requires_grad_(True)
+    y = conv(x)
+    g = torch.randn_like(y)
+    y.backward(g)
+
+    xr = x.detach().float().requires_grad_(True)
+    wr = conv.weight.detach().float().requires_grad_(True)
+    yr = torch.nn.functional.conv2d(xr, wr, stride=2, padding=1)
+    yr.backward(g.float())
+
+    for got, ref in ((x.grad.float(), xr.grad), (conv.weight.grad.float(), wr.grad)):
+        err = (got - ref).abs()
+        scale = ref.abs().mean().clamp(min=0.02)
+        assert (err / scale).max() < 0.12, (err.max().item(), scale.item())
