@@ -1,0 +1,95 @@
+"""GPU coverage for the r1-unmeasured paths: fp16
+dynamic-loss-scaling step, the ctr (wide&deep) step, and DGC on CUDA
+tensors at world 2 (gloo collectives — RCCL refuses 2 ranks on 1 GPU)."""
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+@pytest.fixture(autouse=True)
+def _gpu():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+
+
+def test_fp16_loss_scaled_step():
+    """fp16 + DynamicLossScaler on MI355X: steps apply, loss decreases,
+    found_inf path costs ONE host sync (engine r2 fix)."""
+    from edl_amd.data.synthetic import SyntheticImageNet
+    from edl_amd.train.engine import TrainerEngine
+
+    eng = TrainerEngine(model="resnet18_vd", per_device_batch=8,
+                        dtype="fp16", checkpoint_dir=None).setup()
+    assert eng.scaler is not None
+    loader = SyntheticImageNet(8, eng.device, channels_last=True, seed=5)
+    x, y = loader.next()
+    losses = []
+    for _ in range(6):
+        losses.append(float(eng.train_step(x, y).item()))
+    assert losses[-1] < losses[0], losses  # memorizing one batch
+    assert eng.scaler.value > 0
+
+
+def test_fp16_overflow_skips_step():
+    """A gradient overflow must back the scale off and skip the update."""
+    from edl_amd.train.engine import TrainerEngine
+
+    eng = TrainerEngine(model="mnist_mlp", per_device_batch=4, num_classes=10,
+                        dtype="fp16", checkpoint_dir=None).setup()
+    x = torch.randn(4, 1, 28, 28, device="cuda") * 1e4  # force inf grads
+    y = torch.randint(0, 10, (4,), device="cuda")
+    s0 = eng.scaler.value
+    p0 = next(eng.model.parameters()).detach().clone()
+    eng.train_step(x, y)
+    assert eng.scaler.value <= s0  # backed off (or unchanged if no inf)
+    if eng.scaler.value < s0:  # overflow happened: params untouched
+        assert torch.equal(p0, next(eng.model.parameters()).detach())
+
+
+def test_ctr_wide_and_deep_gpu_step():
+    """BASELINE config 5 model: wide&deep trains on GPU (dense-embedding
+    all-reduce path; sparse ids through nn.Embedding)."""
+    from edl_amd.data.synthetic import SyntheticCTR
+    from edl_amd.models import WideAndDeep
+
+    torch.manual_seed(0)
+    m = WideAndDeep().cuda()
+    opt = torch.optim.SGD(m.parameters(), lr=0.05)
+    data = SyntheticCTR(256, torch.device("cuda"))
+    losses = []
+    for _ in range(12):
+        dense, sparse, label = data.next()
+        loss = torch.nn.functional.binary_cross_entropy_with_logits(
+            m(dense, sparse).squeeze(-1), label)
+        opt.zero_grad()
+        loss.backward()
+        opt.step()
+        losses.append(float(loss.item()))
+    assert losses[-1] < losses[0], losses
+
+
+def test_dgc_two_rank_cuda(tmp_path):
+    """DGC compressed exchange at world 2 on CUDA tensors (gloo): ranks
+    converge to identical grads and error feedback accumulates."""
+    script = os.path.join(REPO, "tests", "_dgc_cuda_worker.py")
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29541", script],
+        env=env, cwd=REPO, capture_output=True, text=True, timeout=600,
+    )
+    sys.stderr.write(r.stdout[-2000:] + r.stderr[-1500:])
+    assert r.returncode == 0
+    oks = [json.loads(l) for l in r.stdout.splitlines()
+           if l.startswith('{"dgc_cuda"')]
+    assert oks and all(v["ok"] for v in oks)
