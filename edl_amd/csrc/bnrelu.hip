@@ -53,17 +53,27 @@ __device__ __forceinline__ void store8(bf16* p, const F8& x) {
 // atomics at all (single-stage atomics measured 17-40 us/dispatch fixed
 // cost: same-word serialization at ~88 adds/us for small C, grid x 2C
 // traffic for large C).
-template <int ST>
+template <int ST, bool CONTIG = false>
 __global__ void bn_stats_kernel(
     const bf16* __restrict__ x, float* __restrict__ partial,  // [grid, 2C]
     const long long M, const int C) {
   __shared__ float lsum[2 * 2048];
   const int c8 = C >> 3;  // channel-octet count
   const int tpr = c8;     // threads per row-slice (each owns 8 channels)
-  const int tid = blockIdx.x * blockDim.x + threadIdx.x;
-  const int lane_c = (int)(tid % tpr);   // which channel octet
-  const long long row0 = tid / tpr;
-  const long long rstride = ((long long)gridDim.x * blockDim.x) / tpr;
+  long long row0, rstride, row_end;
+  if (CONTIG) {
+    const long long rows_blk = (M + gridDim.x - 1) / gridDim.x;
+    const long long blk0 = blockIdx.x * rows_blk;
+    row0 = blk0 + threadIdx.x / tpr;
+    rstride = blockDim.x / tpr;
+    row_end = blk0 + rows_blk < M ? blk0 + rows_blk : M;
+  } else {
+    const int tid = blockIdx.x * blockDim.x + threadIdx.x;
+    row0 = tid / tpr;
+    rstride = ((long long)gridDim.x * blockDim.x) / tpr;
+    row_end = M;
+  }
+  const int lane_c = (int)(threadIdx.x % tpr);   // which channel octet
   const int c0 = lane_c * 8;
 
   for (int i = threadIdx.x; i < 2 * C; i += blockDim.x) lsum[i] = 0.0f;
@@ -74,7 +84,7 @@ __global__ void bn_stats_kernel(
   // ST independent row streams: the grid is capped (finalize reads the
   // partials serially), so per-thread in-flight bytes must cover HBM
   // latency — 1 stream measured ~1.2 TB/s, latency-bound
-  for (; r + (ST - 1) * rstride < M; r += ST * rstride) {
+  for (; r + (ST - 1) * rstride < row_end; r += ST * rstride) {
     F8 v[ST];
 #pragma unroll
     for (int u = 0; u < ST; ++u) v[u] = load8(x + (r + u * rstride) * C + c0);
@@ -87,7 +97,7 @@ __global__ void bn_stats_kernel(
       }
     }
   }
-  for (; r < M; r += rstride) {
+  for (; r < row_end; r += rstride) {
     F8 v = load8(x + r * C + c0);
 #pragma unroll
     for (int i = 0; i < 8; ++i) {
@@ -257,7 +267,10 @@ __global__ void bn_apply_kernel(
 // where dy_eff applies the fused ReLU mask (bitmask from the fwd apply).
 // ST = independent row streams (4 or 8): more outstanding loads per
 // thread for latency hiding at small grids.
-template <bool RELU, int ST = 4>
+// CONTIG: each block owns a CONTIGUOUS row range instead of striding the
+// whole tensor by grid*rows — one streaming region per block (DRAM-page
+// and L2-friendly) instead of grid interleaved streams.
+template <bool RELU, int ST = 4, bool CONTIG = false>
 __global__ void bn_bwd_reduce_kernel(
     const bf16* __restrict__ dy, const unsigned char* __restrict__ mask,
     const bf16* __restrict__ x, const float* __restrict__ mean,
@@ -266,10 +279,20 @@ __global__ void bn_bwd_reduce_kernel(
   __shared__ float lsum[2 * 2048];
   const int c8 = C >> 3;
   const int tpr = c8;
-  const int tid = blockIdx.x * blockDim.x + threadIdx.x;
-  const int lane_c = (int)(tid % tpr);
-  const long long row0 = tid / tpr;
-  const long long rstride = ((long long)gridDim.x * blockDim.x) / tpr;
+  long long row0, rstride, row_end;
+  if (CONTIG) {
+    const long long rows_blk = (M + gridDim.x - 1) / gridDim.x;
+    const long long blk0 = blockIdx.x * rows_blk;
+    row0 = blk0 + threadIdx.x / tpr;
+    rstride = blockDim.x / tpr;
+    row_end = blk0 + rows_blk < M ? blk0 + rows_blk : M;
+  } else {
+    const int tid = blockIdx.x * blockDim.x + threadIdx.x;
+    row0 = tid / tpr;
+    rstride = ((long long)gridDim.x * blockDim.x) / tpr;
+    row_end = M;
+  }
+  const int lane_c = (int)(threadIdx.x % tpr);
   const int c0 = lane_c * 8;
 
   for (int i = threadIdx.x; i < 2 * C; i += blockDim.x) lsum[i] = 0.0f;
@@ -284,7 +307,7 @@ __global__ void bn_bwd_reduce_kernel(
   float s1[8] = {0}, s2[8] = {0};
   long long r = row0;
   // ST row streams x 3 tensors = up to 3*ST outstanding 16-B loads
-  for (; r + (ST - 1) * rstride < M; r += ST * rstride) {
+  for (; r + (ST - 1) * rstride < row_end; r += ST * rstride) {
 #pragma unroll
     for (int u = 0; u < ST / 2; ++u) {
       const long long r0 = r + 2 * u * rstride, r1 = r + (2 * u + 1) * rstride;
@@ -308,7 +331,7 @@ __global__ void bn_bwd_reduce_kernel(
       }
     }
   }
-  for (; r < M; r += rstride) {
+  for (; r < row_end; r += rstride) {
     const long long eoff = r * C + c0;
     F8 g = load8(dy + eoff);
     F8 xv = load8(x + eoff);
@@ -483,12 +506,22 @@ extern "C" int bn_bwd_grid(long long M, int C) {
 
 extern "C" void launch_bn_stats(const void* x, float* partial, int grid,
                                 long long M, int C, hipStream_t s) {
-  if (env_ll("EDL_BN_STATS_STREAMS", 4) >= 8)
+  const bool st8 = env_ll("EDL_BN_STATS_STREAMS", 4) >= 8;
+  const bool contig = env_ll("EDL_BN_CONTIG", 0) != 0;
+  if (contig) {
+    if (st8)
+      hipLaunchKernelGGL((bn_stats_kernel<8, true>), dim3(grid), dim3(256), 0,
+                         s, (const bf16*)x, partial, M, C);
+    else
+      hipLaunchKernelGGL((bn_stats_kernel<4, true>), dim3(grid), dim3(256), 0,
+                         s, (const bf16*)x, partial, M, C);
+  } else if (st8) {
     hipLaunchKernelGGL((bn_stats_kernel<8>), dim3(grid), dim3(256), 0, s,
                        (const bf16*)x, partial, M, C);
-  else
+  } else {
     hipLaunchKernelGGL((bn_stats_kernel<4>), dim3(grid), dim3(256), 0, s,
                        (const bf16*)x, partial, M, C);
+  }
 }
 
 extern "C" void launch_bn_finalize(const float* partial, int nblocks,
@@ -560,16 +593,19 @@ extern "C" void launch_bn_bwd_reduce(const void* dy, const unsigned char* mask,
                                      float* partial, int grid, long long M, int C,
                                      bool relu, hipStream_t s) {
   const bool st8 = env_ll("EDL_BN_BWD_STREAMS", 4) >= 8;
-#define RCASE(R, ST)                                                        \
-  hipLaunchKernelGGL((bn_bwd_reduce_kernel<R, ST>), dim3(grid), dim3(256), \
-                     0, s, (const bf16*)dy, (R) ? mask : nullptr,          \
+  const bool contig = env_ll("EDL_BN_CONTIG", 0) != 0;
+#define RCASE(R, ST, CG)                                                       \
+  hipLaunchKernelGGL((bn_bwd_reduce_kernel<R, ST, CG>), dim3(grid), dim3(256), \
+                     0, s, (const bf16*)dy, (R) ? mask : nullptr,              \
                      (const bf16*)x, mean, invstd, partial, M, C)
   if (relu) {
-    if (st8) RCASE(true, 8);
-    else RCASE(true, 4);
+    if (contig) { if (st8) RCASE(true, 8, true); else RCASE(true, 4, true); }
+    else if (st8) RCASE(true, 8, false);
+    else RCASE(true, 4, false);
   } else {
-    if (st8) RCASE(false, 8);
-    else RCASE(false, 4);
+    if (contig) { if (st8) RCASE(false, 8, true); else RCASE(false, 4, true); }
+    else if (st8) RCASE(false, 8, false);
+    else RCASE(false, 4, false);
   }
 #undef RCASE
 }

@@ -53,8 +53,13 @@ def main():
     e = ops.ext()
     torch.manual_seed(3)
 
-    variants = list(itertools.product(["192", "384", "768"], ["4", "8"], ["1"]))
-    variants.insert(0, ("192", "4", "0"))  # r1 baseline: cap192/st4/fin-v1
+    # (cap, streams, fin_v2, contig)
+    variants = [("192", "4", "0", "0"),  # r1 baseline
+                ("192", "4", "1", "0"),  # r2 shipped default
+                ("192", "4", "1", "1"),  # contiguous block rows
+                ("384", "4", "1", "1"),
+                ("768", "4", "1", "1"),
+                ("192", "8", "1", "1")]
 
     tensors = {}
     for (M, C, relu, add, w) in SHAPES:
@@ -68,11 +73,13 @@ def main():
         tensors[(M, C, relu, add)] = (dy, mask, x, mean, invstd, gamma)
 
     summary = {}
-    for cap, st, fv2 in variants:
+    for cap, st, fv2, contig in variants:
         os.environ["EDL_BN_BWD_GRID_CAP"] = cap
         os.environ["EDL_BN_BWD_STREAMS"] = st
         os.environ["EDL_BN_FIN_V2"] = fv2
-        key = "cap%s_st%s_fin%s" % (cap, st, "v2" if fv2 == "1" else "v1")
+        os.environ["EDL_BN_CONTIG"] = contig
+        key = "cap%s_st%s_fin%s%s" % (cap, st, "v2" if fv2 == "1" else "v1",
+                                      "_contig" if contig == "1" else "")
         total_us = 0.0
         rows = []
         for (M, C, relu, add, w) in SHAPES:
