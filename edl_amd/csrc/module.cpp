@@ -45,6 +45,11 @@ extern "C" void launch_conv3x3_grouped(const void*, const void*, void*, int, int
 extern "C" void launch_conv3x3s2_dgrad(const void*, const void*, void*, int,
                                        int, int, int, int, int, int,
                                        hipStream_t);
+extern "C" void launch_conv3x3_small(const void*, const void*, void*, int, int,
+                                     int, int, int, int, int, int,
+                                     hipStream_t);
+extern "C" void launch_pad_nhwc_cpad(const void*, void*, int, int, int, int,
+                                     int, int, int, hipStream_t);
 extern "C" void launch_transpose_pad(const void*, void*, int, int, int,
                                      hipStream_t);
 extern "C" void launch_gemm_bt_splitk(const void*, const void*, float*, int, int,
@@ -290,6 +295,45 @@ torch::Tensor conv3x3_grouped_fwd(torch::Tensor x, torch::Tensor w3g,
   launch_conv3x3_grouped(xp.data_ptr(), w3g.data_ptr(), y.data_ptr(), (int)M,
                          Cout, Cin, Hout * Wout, Wout, Hp, Wp, (int)stride,
                          gw, s);
+  return y;
+}
+
+torch::Tensor conv3x3_small_fwd(torch::Tensor x, torch::Tensor w3s,
+                                int64_t cout_real, int64_t cpt,
+                                int64_t stride) {
+  // stem conv (small channels): x [N, Cin, H, W] channels_last bf16 with
+  // Cin <= cpt (cpt in {16,32,64}); w3s [64, TAPS_PAD*cpt] zero-padded
+  // repack (conv.py _repack_w3_small). Returns y2d [M, cout_real].
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 &&
+                  x.scalar_type() == torch::kBFloat16 &&
+                  x.is_contiguous(torch::MemoryFormat::ChannelsLast),
+              "conv3x3s: 4-D channels_last bf16");
+  TORCH_CHECK(stride == 1 || stride == 2, "conv3x3s: stride 1 or 2");
+  TORCH_CHECK(cpt == 16 || cpt == 32 || cpt == 64, "conv3x3s: cpt 16/32/64");
+  const int Nimg = (int)x.size(0), Cin = (int)x.size(1);
+  const int H = (int)x.size(2), W = (int)x.size(3);
+  TORCH_CHECK(Cin <= cpt, "conv3x3s: Cin <= cpt");
+  const int taps_pad = cpt == 16 ? 12 : (cpt == 32 ? 10 : 9);
+  TORCH_CHECK(w3s.is_contiguous() && w3s.size(0) == 64 &&
+                  w3s.size(1) == taps_pad * cpt,
+              "conv3x3s: w3s [64, taps_pad*cpt]");
+  TORCH_CHECK(cout_real >= 1 && cout_real <= 64, "conv3x3s: cout <= 64");
+  const int Hp = H + 2, Wp = W + 2;
+  const int Hout = (H - 1) / (int)stride + 1;
+  const int Wout = (W - 1) / (int)stride + 1;
+  const long long M = (long long)Nimg * Hout * Wout;
+  auto s = cur_stream();
+  auto xp = torch::empty({(long long)Nimg * Hp * Wp * cpt}, x.options());
+  if (Cin == cpt) {
+    launch_pad_nhwc(x.data_ptr(), xp.data_ptr(), Nimg, H, W, Hp, Wp, Cin, s);
+  } else {
+    launch_pad_nhwc_cpad(x.data_ptr(), xp.data_ptr(), Nimg, H, W, Hp, Wp,
+                         Cin, (int)cpt, s);
+  }
+  auto y = torch::empty({M, cout_real}, x.options());
+  launch_conv3x3_small(xp.data_ptr(), w3s.data_ptr(), y.data_ptr(), (int)M,
+                       (int)cout_real, (int)cpt, Hout * Wout, Wout, Hp, Wp,
+                       (int)stride, s);
   return y;
 }
 
@@ -566,6 +610,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "grouped (16ch/group) 3x3 conv fwd -> y2d [M, Cout]");
   m.def("conv3x3s2_dgrad", &conv3x3s2_dgrad,
         "stride-2 3x3 same-pad dgrad (parity implicit GEMM) -> dx2d [N*H*W, Cin]");
+  m.def("conv3x3_small_fwd", &conv3x3_small_fwd,
+        "small-channel 3x3 conv (deep stem) -> y2d [M, cout_real]");
   m.def("avgpool2x2_bwd", &avgpool2x2_bwd, "2x2/s2 ceil avg pool backward");
   m.attr("_arch") = "gfx950";
 }

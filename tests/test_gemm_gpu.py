@@ -312,3 +312,70 @@ def test_conv3x3_s2_full_backward_matches_miopen():
         err = (got - ref).abs()
         scale = ref.abs().mean().clamp(min=0.02)
         assert (err / scale).max() < 0.12, (err.max().item(), scale.item())
+
+
+@pytest.mark.parametrize("shape", [
+    (2, 3, 32, 2, 64),    # stem conv0: 3->32 s2
+    (2, 32, 32, 1, 33),   # stem conv1: 32->32 s1 (odd HW)
+    (2, 32, 64, 1, 28),   # stem conv2: 32->64 s1
+])
+def test_conv3x3_small_fwd_numerics(shape):
+    """Deep-stem small-channel conv fwd vs F.conv2d fp32."""
+    import torch.nn.functional as F
+
+    from edl_amd.ops.conv import Conv2dFast
+
+    n, ci, co, stride, hw = shape
+    torch.manual_seed(21)
+    conv = Conv2dFast(ci, co, 3, stride=stride, padding=1, bias=False
+                      ).cuda().to(torch.bfloat16)
+    x = (torch.randn(n, ci, hw, hw, device="cuda")).to(torch.bfloat16)
+    x = x.contiguous(memory_format=torch.channels_last)
+    with torch.no_grad():
+        y = conv(x)
+        ref = F.conv2d(x.float(), conv.weight.float(), stride=stride, padding=1)
+    assert y.shape == ref.shape
+    err = (y.float() - ref).abs()
+    scale = ref.abs().mean().clamp(min=0.2)
+    assert (err / scale).max() < 0.1, (err.max().item(), scale.item())
+
+
+@pytest.mark.parametrize("shape", [(2, 32, 32, 30), (2, 32, 64, 14)])
+def test_conv3x3_small_backward_numerics(shape):
+    """Stem conv backward: in-repo dgrad + unfold-GEMM wgrad vs fp32."""
+    import torch.nn.functional as F
+
+    from edl_amd.ops.conv import Conv2dFast
+
+    n, ci, co, hw = shape
+    torch.manual_seed(22)
+    conv = Conv2dFast(ci, co, 3, padding=1, bias=False).cuda().to(torch.bfloat16)
+    x = (torch.randn(n, ci, hw, hw, device="cuda") * 0.5).to(torch.bfloat16)
+    x = x.contiguous(memory_format=torch.channels_last).requires_grad_(True)
+    y = conv(x)
+    g = torch.randn_like(y).contiguous(memory_format=torch.channels_last)
+    y.backward(g)
+
+    xr = x.detach().float().requires_grad_(True)
+    wr = conv.weight.detach().float().requires_grad_(True)
+    yr = F.conv2d(xr, wr, padding=1)
+    yr.backward(g.float())
+    for got, ref in ((x.grad.float(), xr.grad),
+                     (conv.weight.grad.float(), wr.grad)):
+        err = (got - ref).abs()
+        scale = ref.abs().mean().clamp(min=0.02)
+        assert (err / scale).max() < 0.15, (err.max().item(), scale.item())
+
+
+def test_stem_conv0_wgrad_no_dx():
+    """conv0 (3ch input, requires_grad False): weight grad flows, no dx."""
+    from edl_amd.ops.conv import Conv2dFast
+
+    conv = Conv2dFast(3, 32, 3, stride=2, padding=1, bias=False
+                      ).cuda().to(torch.bfloat16)
+    x = torch.randn(2, 3, 32, 32, device="cuda").to(torch.bfloat16)
+    x = x.contiguous(memory_format=torch.channels_last)
+    y = conv(x)
+    y.float().sum().backward()
+    assert conv.weight.grad is not None
+    assert torch.isfinite(conv.weight.grad.float()).all()

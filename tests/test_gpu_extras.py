@@ -68,7 +68,7 @@ def test_ctr_wide_and_deep_gpu_step():
     for _ in range(12):
         dense, sparse, label = data.next()
         loss = torch.nn.functional.binary_cross_entropy_with_logits(
-            m(dense, sparse).squeeze(-1), label)
+            m(dense, sparse).view(-1), label.view(-1))
         opt.zero_grad()
         loss.backward()
         opt.step()
