@@ -47,11 +47,18 @@ def ext():
     return m
 
 
-def swap_module_ops(model):
-    """Swap hot torch modules for HIP-fused equivalents in-place.
+def assert_hip_ops_available(model=None):
+    """Fail LOUDLY if the HIP extension is missing on a GPU host.
 
-    Currently: validates the extension is loadable on GPU (loud failure).
-    Fused BN+ReLU module swaps land here as kernels arrive."""
+    The fused modules (Conv2dFast, BNReLU2d, FusedSGD, avgpool, KD-CE)
+    are baked into the model classes themselves — nothing is swapped at
+    runtime; this is the engine's guard against silently training on
+    torch fallbacks (the old name swap_module_ops
+    implied a swap that never happened)."""
     if torch.cuda.is_available():
         ext()  # raises if the HIP extension is missing
     return model
+
+
+# backward-compat alias (older scripts)
+swap_module_ops = assert_hip_ops_available

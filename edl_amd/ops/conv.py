@@ -238,12 +238,20 @@ class _Conv3x3SmallHip(torch.autograd.Function):
             cpt = _small_cpt(co)
             dx2d = ext().conv3x3_small_fwd(dyb, ctx.w3srot, ci, cpt, 1)
             dx = dx2d.view(n, h, w, ci).permute(0, 3, 1, 2)
-        # wgrad: dW[co, ci*9] = dY2d^T @ unfold(x) — one library GEMM
-        # (bf16 operands, fp32 accumulate inside hipBLASLt)
-        dy2d = dyb.permute(0, 2, 3, 1).reshape(-1, co)
-        xu = F.unfold(x, 3, padding=1, stride=ctx.stride)
-        xu2d = xu.permute(0, 2, 1).reshape(-1, ci * 9)  # [M, ci*9] bf16
-        dw = (dy2d.t() @ xu2d).float().view(co, ci, 3, 3)
+        # wgrad. Default: torch.nn.grad.conv2d_weight (MIOpen igemm_wrw,
+        # ~51 us/call steady). The unfold+hipBLASLt route is kept opt-in
+        # (EDL_STEM_WGRAD=unfold) but MEASURED NEGATIVE: hipBLASLt runs
+        # the [32, 401k] @ [401k, 288] reduction GEMM at ~918 us — no
+        # split-K pick at tiny M,N (gpurun_out/r2c7, bench 3221 -> 2566).
+        if os.environ.get("EDL_STEM_WGRAD", "miopen") == "unfold":
+            dy2d = dyb.permute(0, 2, 3, 1).reshape(-1, co)
+            xu = F.unfold(x, 3, padding=1, stride=ctx.stride)
+            xu2d = xu.permute(0, 2, 1).reshape(-1, ci * 9)  # [M, ci*9]
+            dw = (dy2d.t() @ xu2d).float().view(co, ci, 3, 3)
+        else:
+            dw = torch.nn.grad.conv2d_weight(
+                x, ctx.w_shape, dyb, stride=(ctx.stride, ctx.stride),
+                padding=(1, 1))
         if dw.dtype != ctx.w_dtype:
             dw = dw.to(ctx.w_dtype)
         return dx, dw, None, None, None, None
@@ -359,7 +367,8 @@ class Conv2dFast(nn.Conv2d):
             and os.environ.get("EDL_CONV3X3_SMALL", "1") == "1"
             and x.is_cuda
             and available()
-            and x.dtype == torch.bfloat16
+            and (x.dtype == torch.bfloat16
+                 or (x.dtype == torch.float32 and torch.is_autocast_enabled()))
             and self.kernel_size == (3, 3)
             and self.groups == 1
             and self.padding == (1, 1)
@@ -370,6 +379,8 @@ class Conv2dFast(nn.Conv2d):
             and self.out_channels <= 64
             and (self.in_channels % 64 != 0 or self.out_channels % 64 != 0)
         ):
+            if x.dtype != torch.bfloat16:
+                x = x.to(torch.bfloat16)  # the cast autocast would do
             if not x.is_contiguous(memory_format=torch.channels_last):
                 x = x.contiguous(memory_format=torch.channels_last)
             cpt = _small_cpt(self.in_channels)

@@ -158,7 +158,7 @@ class TrainerEngine:
         if self.use_hip_ops and self.device.type == "cuda":
             from .. import ops
 
-            ops.swap_module_ops(self.model)
+            ops.assert_hip_ops_available(self.model)
         if self.recompute and hasattr(self.model, "stages"):
             import torch.utils.checkpoint as ckpt_mod
 

@@ -22,6 +22,7 @@ def main():
         "stage": env.cluster_stage,
         "job_stage": env.job_stage,
         "pid": os.getpid(),
+        "endpoints": ",".join(env.trainer_endpoints or []),
     }
     print("fake_trainer start: %s" % rec, flush=True)
     if out:

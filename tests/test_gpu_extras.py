@@ -64,9 +64,9 @@ def test_ctr_wide_and_deep_gpu_step():
     m = WideAndDeep().cuda()
     opt = torch.optim.SGD(m.parameters(), lr=0.05)
     data = SyntheticCTR(256, torch.device("cuda"))
+    dense, sparse, label = data.next()  # fixed batch: loss must memorize
     losses = []
     for _ in range(12):
-        dense, sparse, label = data.next()
         loss = torch.nn.functional.binary_cross_entropy_with_logits(
             m(dense, sparse).view(-1), label.view(-1))
         opt.zero_grad()
