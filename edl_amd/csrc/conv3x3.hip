@@ -59,12 +59,18 @@ extern "C" __global__ void pad_nhwc_kernel(
 //     (stage 3/4 of ResNeXt101_32x16d, ~80% of its grouped FLOPs).
 //   cpg 16/32: gw = 64 with a block-diagonal zero-padded weight repack
 //     (4x/2x MFMA on zeros, still ~MFMA rate vs MIOpen's grouped path).
+// SPLITK (grid.y > 1, non-grouped): the deep-K small-M shapes (ResNet
+// stage 3/4: 52-98 tiles on a 256-CU chip, K up to 4608) slice the
+// K-range over grid.y and atomically fold fp32 partials into C_part
+// (zeroed by the caller; cast to bf16 afterwards) — measured 56 us/call
+// at 52 blocks before, CU-starved.
 template <int BM, int BN, int WAVES_M, int WAVES_N, bool GROUPED = false>
 __global__ __launch_bounds__(256, 2) void conv3x3_kernel(
     const bf16* __restrict__ XP, const bf16* __restrict__ B,
     bf16* __restrict__ C_out, const int M, const int N, const int Cin,
     const int HW_out, const int W_out, const int Hp, const int Wp,
-    const int stride_hw, const int gw = 0) {
+    const int stride_hw, const int gw = 0,
+    float* __restrict__ C_part = nullptr) {
   constexpr int BK = 64;
   constexpr int A_BYTES = BM * BK * 2;
   constexpr int B_BYTES = BN * BK * 2;
@@ -154,11 +160,18 @@ __global__ __launch_bounds__(256, 2) void conv3x3_kernel(
 
   f32x4 acc[4][4] = {};
   const int KT = K / BK;
-  stage(0, 0);
+  int kt0 = 0, kt1 = KT;
+  if (!GROUPED && gridDim.y > 1) {
+    const int per = (KT + gridDim.y - 1) / gridDim.y;
+    kt0 = blockIdx.y * per;
+    kt1 = kt0 + per < KT ? kt0 + per : KT;
+    if (kt0 >= kt1) return;
+  }
+  stage(kt0 & 1, kt0);
   __syncthreads();
-  for (int kt = 0; kt < KT; ++kt) {
+  for (int kt = kt0; kt < kt1; ++kt) {
     const int cur = kt & 1;
-    if (kt + 1 < KT) stage(cur ^ 1, kt + 1);
+    if (kt + 1 < kt1) stage(cur ^ 1, kt + 1);
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
       bf16x8 a[4], b[4];
@@ -184,10 +197,17 @@ __global__ __launch_bounds__(256, 2) void conv3x3_kernel(
     for (int reg = 0; reg < 4; ++reg) {
       const int m = m0 + wm + mf * 16 + r4 + reg;
       if (m < M) {
-        bf16* crow = C_out + (long long)m * N + n0 + wn + cn;
+        if (!GROUPED && C_part != nullptr) {
+          float* prow = C_part + (long long)m * N + n0 + wn + cn;
 #pragma unroll
-        for (int nf = 0; nf < 4; ++nf)
-          crow[nf * 16] = __float2bfloat16(acc[mf][nf][reg]);
+          for (int nf = 0; nf < 4; ++nf)
+            atomicAdd(&prow[nf * 16], acc[mf][nf][reg]);
+        } else {
+          bf16* crow = C_out + (long long)m * N + n0 + wn + cn;
+#pragma unroll
+          for (int nf = 0; nf < 4; ++nf)
+            crow[nf * 16] = __float2bfloat16(acc[mf][nf][reg]);
+        }
       }
     }
   }
@@ -591,22 +611,40 @@ extern "C" void launch_conv3x3_grouped(const void* xp, const void* w3g, void* y,
                      (bf16*)y, M, Cout, Cin, HW_out, W_out, Hp, Wp, stride, gw);
 }
 
+extern "C" int conv3x3_pick_splitk(int M, int Cout, int Cin) {
+  // split only CU-starved deep-K launches (stage 3/4 at bs32: 52-98
+  // tiles, K 2304-4608); aim for ~384-512 total blocks
+  const int bm = Cout % 128 == 0 ? 128 : 256;
+  const int bn = bm == 128 ? 128 : 64;
+  const int tiles = ((M + bm - 1) / bm) * (Cout / bn);
+  const int KT = 9 * Cin / 64;
+  if (tiles >= 224 || KT < 8) return 1;
+  int sk = 384 / tiles;
+  if (sk > KT / 4) sk = KT / 4;  // keep >= 4 k-steps per slice
+  return sk < 1 ? 1 : sk;
+}
+
 extern "C" void launch_conv3x3(const void* xp, const void* w3, void* y, int M,
                                int Cout, int Cin, int HW_out, int W_out, int Hp,
-                               int Wp, int stride, hipStream_t s) {
+                               int Wp, int stride, float* cpart, int splitk,
+                               hipStream_t s) {
+  if (splitk < 1) splitk = 1;
+  if (cpart == nullptr) splitk = 1;
   if (Cout % 128 == 0) {
     constexpr int BM = 128, BN = 128;
     const int grid = ((M + BM - 1) / BM) * (Cout / BN);
     const int lds_bytes = 2 * (BM * 64 * 2 + BN * 64 * 2);
-    hipLaunchKernelGGL((conv3x3_kernel<BM, BN, 2, 2>), dim3(grid), dim3(256),
-                       lds_bytes, s, (const bf16*)xp, (const bf16*)w3, (bf16*)y,
-                       M, Cout, Cin, HW_out, W_out, Hp, Wp, stride);
+    hipLaunchKernelGGL((conv3x3_kernel<BM, BN, 2, 2>), dim3(grid, splitk),
+                       dim3(256), lds_bytes, s, (const bf16*)xp,
+                       (const bf16*)w3, (bf16*)y, M, Cout, Cin, HW_out, W_out,
+                       Hp, Wp, stride, 0, splitk > 1 ? cpart : nullptr);
   } else {
     constexpr int BM = 256, BN = 64;
     const int grid = ((M + BM - 1) / BM) * (Cout / BN);
     const int lds_bytes = 2 * (BM * 64 * 2 + BN * 64 * 2);
-    hipLaunchKernelGGL((conv3x3_kernel<BM, BN, 4, 1>), dim3(grid), dim3(256),
-                       lds_bytes, s, (const bf16*)xp, (const bf16*)w3, (bf16*)y,
-                       M, Cout, Cin, HW_out, W_out, Hp, Wp, stride);
+    hipLaunchKernelGGL((conv3x3_kernel<BM, BN, 4, 1>), dim3(grid, splitk),
+                       dim3(256), lds_bytes, s, (const bf16*)xp,
+                       (const bf16*)w3, (bf16*)y, M, Cout, Cin, HW_out, W_out,
+                       Hp, Wp, stride, 0, splitk > 1 ? cpart : nullptr);
   }
 }

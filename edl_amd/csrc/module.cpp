@@ -38,7 +38,9 @@ extern "C" void launch_gemm_bt(const void*, const void*, void*, int, int, int,
 extern "C" void launch_pad_nhwc(const void*, void*, int, int, int, int, int, int,
                                 hipStream_t);
 extern "C" void launch_conv3x3(const void*, const void*, void*, int, int, int,
-                               int, int, int, int, int, hipStream_t);
+                               int, int, int, int, int, float*, int,
+                               hipStream_t);
+extern "C" int conv3x3_pick_splitk(int, int, int);
 extern "C" void launch_conv3x3_grouped(const void*, const void*, void*, int, int,
                                        int, int, int, int, int, int, int,
                                        hipStream_t);
@@ -561,9 +563,18 @@ torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w3, int64_t stride) {
   auto s = cur_stream();
   auto xp = torch::empty({(long long)Nimg * Hp * Wp * Cin}, x.options());
   launch_pad_nhwc(x.data_ptr(), xp.data_ptr(), Nimg, H, W, Hp, Wp, Cin, s);
+  const int splitk = conv3x3_pick_splitk((int)M, Cout, Cin);
+  if (splitk > 1) {
+    // CU-starved deep-K shape: fp32 split-K partials + cast
+    auto part = torch::zeros({M, Cout}, x.options().dtype(torch::kFloat32));
+    launch_conv3x3(xp.data_ptr(), w3.data_ptr(), nullptr, (int)M, Cout, Cin,
+                   Hout * Wout, Wout, Hp, Wp, (int)stride,
+                   part.data_ptr<float>(), splitk, s);
+    return part.to(torch::kBFloat16);
+  }
   auto y = torch::empty({M, Cout}, x.options());
   launch_conv3x3(xp.data_ptr(), w3.data_ptr(), y.data_ptr(), (int)M, Cout, Cin,
-                 Hout * Wout, Wout, Hp, Wp, (int)stride, s);
+                 Hout * Wout, Wout, Hp, Wp, (int)stride, nullptr, 1, s);
   return y;
 }
 
