@@ -29,7 +29,7 @@ extern "C" __global__ void avgpool2x2_fwd_kernel(
       for (int dw = 0; dw < wn; ++dw) {
         const bf16* src = x + n * img_in +
                           (((long long)(h0 + dh) * W) + (w0 + dw)) * C + oct * 8;
-        const uint4 raw = *reinterpret_cast<const uint4*>(src);
+        const uint4 raw = *reinterpret_cast<const uint4*>(__builtin_assume_aligned(src, 16));
         const ushort* u = reinterpret_cast<const ushort*>(&raw);
 #pragma unroll
         for (int k = 0; k < 8; ++k) {
@@ -45,7 +45,7 @@ extern "C" __global__ void avgpool2x2_fwd_kernel(
     for (int k = 0; k < 8; ++k)
       ou[k] = (ushort)__hip_bfloat16_raw(__float2bfloat16(acc[k] * inv)).x;
     bf16* dst = y + (long long)n * Ho * Wo * C + pix * C + oct * 8;
-    *reinterpret_cast<uint4*>(dst) = out;
+    *reinterpret_cast<uint4*>(__builtin_assume_aligned(dst, 16)) = out;
   }
   (void)total;
 }
@@ -67,7 +67,7 @@ extern "C" __global__ void avgpool2x2_bwd_kernel(
     const float inv = 1.0f / (float)(hn * wn);
     const bf16* src = dy + ((long long)n * Ho * Wo + (long long)ho * Wo + wo) * C
                       + oct * 8;
-    const uint4 raw = *reinterpret_cast<const uint4*>(src);
+    const uint4 raw = *reinterpret_cast<const uint4*>(__builtin_assume_aligned(src, 16));
     const ushort* u = reinterpret_cast<const ushort*>(&raw);
     uint4 out;
     ushort* ou = reinterpret_cast<ushort*>(&out);
@@ -78,7 +78,7 @@ extern "C" __global__ void avgpool2x2_bwd_kernel(
       ou[k] = (ushort)__hip_bfloat16_raw(__float2bfloat16(cv.f * inv)).x;
     }
     bf16* dst = dx + (long long)n * H * W * C + pix * C + oct * 8;
-    *reinterpret_cast<uint4*>(dst) = out;
+    *reinterpret_cast<uint4*>(__builtin_assume_aligned(dst, 16)) = out;
   }
 }
 

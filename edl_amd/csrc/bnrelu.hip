@@ -22,8 +22,12 @@ struct F8 {
 };
 
 __device__ __forceinline__ F8 load8(const bf16* p) {
-  // 16-byte vector load of 8 bf16
-  const uint4 raw = *reinterpret_cast<const uint4*>(p);
+  // 16-byte vector load of 8 bf16. assume_aligned is LOAD-BEARING: from
+  // bf16 pointer arithmetic LLVM infers align 2 and lowers the uint4
+  // load as ushort+dwordx2+dword+ushort pieces (4 VMEM ops, seen in the
+  // .s of the r2 reduce loop — the BN kernels ran 5x off roofline).
+  const uint4 raw = *reinterpret_cast<const uint4*>(
+      __builtin_assume_aligned(p, 16));
   F8 o;
   const ushort* u = reinterpret_cast<const ushort*>(&raw);
 #pragma unroll
@@ -42,7 +46,7 @@ __device__ __forceinline__ void store8(bf16* p, const F8& x) {
   for (int i = 0; i < 8; ++i) {
     u[i] = (ushort)(__hip_bfloat16_raw(__float2bfloat16(x.v[i])).x);
   }
-  *reinterpret_cast<uint4*>(p) = raw;
+  *reinterpret_cast<uint4*>(__builtin_assume_aligned(p, 16)) = raw;
 }
 
 // ---------------- forward ----------------

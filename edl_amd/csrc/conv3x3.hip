@@ -42,9 +42,10 @@ extern "C" __global__ void pad_nhwc_kernel(
     const int h = hp - 1, w = wp - 1;
     if (h >= 0 && h < H && w >= 0 && w < W) {
       const bf16* src = x + (((long long)n * H + h) * W + w) * C + oct * 8;
-      *reinterpret_cast<uint4*>(dst) = *reinterpret_cast<const uint4*>(src);
+      *reinterpret_cast<uint4*>(__builtin_assume_aligned(dst, 16)) =
+          *reinterpret_cast<const uint4*>(__builtin_assume_aligned(src, 16));
     } else {
-      *reinterpret_cast<uint4*>(dst) = uint4{0, 0, 0, 0};
+      *reinterpret_cast<uint4*>(__builtin_assume_aligned(dst, 16)) = uint4{0, 0, 0, 0};
     }
   }
   (void)HW_in;

@@ -26,7 +26,7 @@ extern "C" __global__ void transpose_pad_kernel(
     if (m < M && tc0 + lc < C) {
       const bf16* src = in + (long long)m * C + tc0 + lc;
       *reinterpret_cast<uint4*>(&tile[half * 32 + lr][lc]) =
-          *reinterpret_cast<const uint4*>(src);
+          *reinterpret_cast<const uint4*>(__builtin_assume_aligned(src, 16));
     } else {
       uint4 z = {0, 0, 0, 0};
       *reinterpret_cast<uint4*>(&tile[half * 32 + lr][lc]) = z;
@@ -41,7 +41,7 @@ extern "C" __global__ void transpose_pad_kernel(
 #pragma unroll
       for (int i = 0; i < 8; ++i) v[i] = tile[lc + i][half * 32 + lr];
       bf16* dst = out + (long long)c * Mp + tm0 + lc;
-      *reinterpret_cast<uint4*>(dst) = *reinterpret_cast<const uint4*>(v);
+      *reinterpret_cast<uint4*>(__builtin_assume_aligned(dst, 16)) = *reinterpret_cast<const uint4*>(v);
     }
   }
 }
@@ -78,7 +78,8 @@ extern "C" __global__ void shift9_transpose_kernel(
       const int h = rem / W_out, w = rem % W_out;
       const long long prow =
           ((long long)n_img * Hp + h * stride_hw) * Wp + w * stride_hw;
-      v = *reinterpret_cast<const uint4*>(xp + prow * C + shift + tc0 + lc);
+      v = *reinterpret_cast<const uint4*>(
+          __builtin_assume_aligned(xp + prow * C + shift + tc0 + lc, 16));
     }
     *reinterpret_cast<uint4*>(&tile[half * 32 + lr][lc]) = v;
   }
@@ -90,7 +91,7 @@ extern "C" __global__ void shift9_transpose_kernel(
 #pragma unroll
       for (int i = 0; i < 8; ++i) v[i] = tile[lc + i][half * 32 + lr];
       bf16* dst = out + ((long long)s * C + c) * Mp + tm0 + lc;
-      *reinterpret_cast<uint4*>(dst) = *reinterpret_cast<const uint4*>(v);
+      *reinterpret_cast<uint4*>(__builtin_assume_aligned(dst, 16)) = *reinterpret_cast<const uint4*>(v);
     }
   }
 }
