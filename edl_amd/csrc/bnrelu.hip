@@ -511,7 +511,7 @@ extern "C" int bn_bwd_grid(long long M, int C) {
 extern "C" void launch_bn_stats(const void* x, float* partial, int grid,
                                 long long M, int C, hipStream_t s) {
   const bool st8 = env_ll("EDL_BN_STATS_STREAMS", 4) >= 8;
-  const bool contig = env_ll("EDL_BN_CONTIG", 1) != 0;
+  const bool contig = env_ll("EDL_BN_CONTIG", 0) != 0;
   if (contig) {
     if (st8)
       hipLaunchKernelGGL((bn_stats_kernel<8, true>), dim3(grid), dim3(256), 0,
@@ -597,7 +597,7 @@ extern "C" void launch_bn_bwd_reduce(const void* dy, const unsigned char* mask,
                                      float* partial, int grid, long long M, int C,
                                      bool relu, hipStream_t s) {
   const bool st8 = env_ll("EDL_BN_BWD_STREAMS", 4) >= 8;
-  const bool contig = env_ll("EDL_BN_CONTIG", 1) != 0;
+  const bool contig = env_ll("EDL_BN_CONTIG", 0) != 0;
 #define RCASE(R, ST, CG)                                                       \
   hipLaunchKernelGGL((bn_bwd_reduce_kernel<R, ST, CG>), dim3(grid), dim3(256), \
                      0, s, (const bf16*)dy, (R) ? mask : nullptr,              \

@@ -515,13 +515,15 @@ torch::Tensor gemm_tn3x3_splitk(torch::Tensor dy2d, torch::Tensor x,
   int perm = 0;
   if (out.has_value()) {
     // direct-grad mode: accumulate straight into the conv weight's
-    // bucket-view gradient, laid out [Cout, Cin, 3, 3] (the kernel
-    // epilogue remaps from its native s-major [Cout, 9*Cin])
+    // bucket-view gradient. A 2-D [Cout, 9*Cin] out (channels-last
+    // bucket storage) takes the kernel's NATIVE s-major epilogue —
+    // coalesced; a [Cout, Cin, 3, 3] out takes the perm remap
+    // (stride-9 scatter, kept for the standard layout).
     c = *out;
     TORCH_CHECK(c.is_contiguous() && c.scalar_type() == torch::kFloat32 &&
                     c.numel() == (int64_t)Cout * Cin * 9,
-                "gemm_tn3x3_splitk: out fp32 contiguous [Cout,Cin,3,3]");
-    perm = Cin;
+                "gemm_tn3x3_splitk: out fp32 contiguous");
+    perm = (c.dim() == 2 && c.size(1) == 9 * Cin) ? 0 : Cin;
   } else {
     c = torch::zeros({Cout, 9 * Cin}, x.options().dtype(torch::kFloat32));
   }

@@ -340,6 +340,8 @@ class Conv2dFast(nn.Conv2d):
         ):
             if not x.is_contiguous(memory_format=torch.channels_last):
                 x = x.contiguous(memory_format=torch.channels_last)
+            co, ci = self.out_channels, self.in_channels
+            cl = getattr(self.weight, "_edl_phys_shape", None) is not None
             mb = getattr(self.weight, "_edl_bf16", None)
             if mb is not None:  # engine bucket mirror: zero-cost bf16 copy
                 w_bf16 = mb
@@ -348,19 +350,31 @@ class Conv2dFast(nn.Conv2d):
                 w_bf16 = self._cached("w_bf16", lambda: self.weight.detach().to(
                     torch.bfloat16).contiguous())
                 wsrc = self.weight.detach()
-            w3 = self._cached("w3", lambda: _repack_w3(wsrc))
+            if cl and mb is not None:
+                # channels-last bucket: s-major w3 IS the mirror (a view)
+                w3 = mb.permute(0, 2, 3, 1).reshape(co, 9 * ci)
+            else:
+                w3 = self._cached("w3", lambda: _repack_w3(wsrc))
             if self.stride[0] == 1:
                 w3rot = self._cached("w3rot", lambda: _repack_w3(
                     wsrc.permute(1, 0, 2, 3).flip(2, 3)))
             else:
                 w3rot = self._cached("w3s2d",
                                      lambda: _repack_w3_s2dgrad(wsrc))
-            # NOTE: direct-grad stays OFF for 3x3 — the [Cout,Cin,3,3]
-            # remap makes the split-K epilogue atomics stride-9 scattered
-            # (measured: slower end-to-end than the AccumulateGrad add it
-            # saves). 1x1/BN targets are layout-native and stay direct.
+            # direct-grad (world 1): ONLY with channels-last bucket
+            # storage — the wgrad's s-major [Cout, 9Cin] epilogue then IS
+            # the grad-bucket layout (coalesced accumulate). With the
+            # standard layout the stride-9 scatter measured slower than
+            # the AccumulateGrad add it saves (round 1).
+            grad_tgt = None
+            w = self.weight
+            if (cl and getattr(w, "_edl_direct_grad", False)
+                    and w.grad is not None and w.grad.dtype == torch.float32
+                    and ci <= _WGRAD3_MAXC
+                    and torch.is_grad_enabled() and self.training):
+                grad_tgt = w.grad.permute(0, 2, 3, 1).reshape(co, 9 * ci)
             return _Conv3x3Hip.apply(x, self.weight, w_bf16, w3, w3rot,
-                                     self.stride[0], None)
+                                     self.stride[0], grad_tgt)
         if (
             # deep-stem 3x3s (3->32->32->64): small-channel kernel
             _CONV3X3 == "hip"

@@ -13,6 +13,7 @@ checkpoints survive re-bucketing across elastic resizes."""
 import torch
 
 from . import available, ext
+from ..train.bucketed_ddp import _view_like
 
 
 class FusedSGD:
@@ -104,7 +105,7 @@ class FusedSGD:
         idx = 0
         for bk in self._materialize():
             for (off, n), p in zip(bk["offsets"], bk["params"]):
-                state[idx] = {"momentum_buffer": bk["m"][off:off + n].view_as(p).clone()}
+                state[idx] = {"momentum_buffer": _view_like(bk["m"][off:off + n], p).clone()}
                 idx += 1
         groups = [{k: v for k, v in self.param_groups[0].items() if k != "params"}]
         groups[0]["params"] = list(range(idx))
@@ -117,7 +118,7 @@ class FusedSGD:
             for (off, n), p in zip(bk["offsets"], bk["params"]):
                 ent = state.get(idx, state.get(str(idx)))
                 if ent is not None and "momentum_buffer" in ent and ent["momentum_buffer"] is not None:
-                    bk["m"][off:off + n].view_as(p).copy_(
+                    _view_like(bk["m"][off:off + n], p).copy_(
                         ent["momentum_buffer"].to(bk["m"].device)
                     )
                 idx += 1

@@ -28,6 +28,22 @@ import torch
 import torch.distributed as dist
 
 
+def _view_like(flat_slice, p):
+    """View a flat bucket slice with the param's LOGICAL shape.
+
+    Plain params: a contiguous view. Params flagged `_edl_phys_shape` /
+    `_edl_phys_perm` (the engine marks conv3x3 weights channels-last:
+    phys [Cout, 3, 3, Cin], perm (0, 3, 1, 2)) get the physical layout
+    in the bucket and a permuted logical view — the per-step s-major w3
+    repack then becomes a zero-cost reshape of the bf16 mirror, and the
+    3x3 wgrad's s-major epilogue writes the grad bucket COALESCED
+    (round-1's measured-negative stride-9 scatter disappears)."""
+    phys = getattr(p, "_edl_phys_shape", None)
+    if phys is None:
+        return flat_slice.view_as(p)
+    return flat_slice.view(*phys).permute(*p._edl_phys_perm)
+
+
 class _Bucket:
     __slots__ = ("params", "buffer", "ready", "work", "grads", "param_flat")
 
@@ -89,12 +105,12 @@ class BucketedAllReducer:
             grads = []
             off = 0
             for p in ps:
-                g = buf[off:off + p.numel()].view_as(p)
+                g = _view_like(buf[off:off + p.numel()], p)
                 p.grad = g
                 grads.append(g)
                 if pflat is not None:
                     with torch.no_grad():
-                        pv = pflat[off:off + p.numel()].view_as(p)
+                        pv = _view_like(pflat[off:off + p.numel()], p)
                         pv.copy_(p.data)
                         p.data = pv
                 off += p.numel()

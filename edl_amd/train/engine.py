@@ -159,6 +159,7 @@ class TrainerEngine:
             from .. import ops
 
             ops.assert_hip_ops_available(self.model)
+            self._mark_channels_last_conv_weights()
         if self.recompute and hasattr(self.model, "stages"):
             import torch.utils.checkpoint as ckpt_mod
 
@@ -204,6 +205,24 @@ class TrainerEngine:
         edist.barrier(self.device)
         return self
 
+    def _mark_channels_last_conv_weights(self):
+        """Flag dense 3x3 conv weights for channels-last BUCKET storage
+        ([Cout, 3, 3, Cin] physical, logical view unchanged): the s-major
+        w3 repack becomes a zero-cost view of the bf16 mirror (one less
+        full-weight copy per conv per step) and the 3x3 wgrad's s-major
+        epilogue can accumulate the grad bucket COALESCED (round-1's
+        stride-9 scatter measured-negative disappears)."""
+        from ..ops.conv import Conv2dFast
+
+        for m in self.model.modules():
+            if (isinstance(m, Conv2dFast) and m.kernel_size == (3, 3)
+                    and m.groups == 1 and m.bias is None
+                    and m.in_channels % 64 == 0 and m.out_channels % 64 == 0
+                    and m.padding == (1, 1)):
+                w = m.weight
+                w._edl_phys_shape = (w.shape[0], 3, 3, w.shape[1])
+                w._edl_phys_perm = (0, 3, 1, 2)
+
     def _broadcast_opt_state(self, src=0):
         import torch.distributed as dist
 
@@ -228,8 +247,9 @@ class TrainerEngine:
             mirror = torch.empty_like(b.param_flat, dtype=torch.bfloat16)
             self._bf16_mirrors.append((mirror, b.param_flat))
             off = 0
+            from .bucketed_ddp import _view_like
             for p in b.params:
-                p._edl_bf16 = mirror[off:off + p.numel()].view_as(p)
+                p._edl_bf16 = _view_like(mirror[off:off + p.numel()], p)
                 off += p.numel()
         self._refresh_bf16_mirrors()
 
