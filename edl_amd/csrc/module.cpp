@@ -52,6 +52,14 @@ extern "C" void launch_conv3x3_small(const void*, const void*, void*, int, int,
                                      hipStream_t);
 extern "C" void launch_pad_nhwc_cpad(const void*, void*, int, int, int, int,
                                      int, int, int, hipStream_t);
+extern "C" void launch_repack_dgrad_w3(const void*, void*, int, int, int,
+                                       hipStream_t);
+extern "C" void launch_maxpool3x3s2_fwd(const void*, void*, unsigned char*,
+                                        int, int, int, int, int, int,
+                                        hipStream_t);
+extern "C" void launch_maxpool3x3s2_bwd(const void*, const unsigned char*,
+                                        void*, int, int, int, int, int, int,
+                                        hipStream_t);
 extern "C" void launch_transpose_pad(const void*, void*, int, int, int,
                                      hipStream_t);
 extern "C" void launch_gemm_bt_splitk(const void*, const void*, float*, int, int,
@@ -366,6 +374,59 @@ torch::Tensor conv3x3s2_dgrad(torch::Tensor dy, torch::Tensor wcat,
   return dx;
 }
 
+torch::Tensor repack_dgrad_w3(torch::Tensor w_smaj, int64_t ci,
+                              int64_t mode) {
+  // w_smaj: [Co, 9*Ci] bf16 contiguous (the channels-last mirror's
+  // s-major view) -> [Ci, 9*Co]: mode 0 = rotated stride-1 dgrad
+  // weights, mode 1 = the stride-2 parity-class wcat. One kernel
+  // instead of flip + permute-copy + cast per conv per step.
+  TORCH_CHECK(w_smaj.is_cuda() && w_smaj.scalar_type() == torch::kBFloat16 &&
+                  w_smaj.is_contiguous() && w_smaj.dim() == 2 &&
+                  w_smaj.size(1) == 9 * ci,
+              "repack_dgrad_w3: [Co, 9*Ci] bf16");
+  const int Co = (int)w_smaj.size(0);
+  TORCH_CHECK(Co % 64 == 0 && ci % 64 == 0, "repack_dgrad_w3: C % 64");
+  auto out = torch::empty({ci, (long long)9 * Co}, w_smaj.options());
+  launch_repack_dgrad_w3(w_smaj.data_ptr(), out.data_ptr(), Co, (int)ci,
+                         (int)mode, cur_stream());
+  return out;
+}
+
+std::vector<torch::Tensor> maxpool3x3s2_fwd(torch::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 &&
+                  x.scalar_type() == torch::kBFloat16 &&
+                  x.is_contiguous(torch::MemoryFormat::ChannelsLast),
+              "maxpool3x3s2: 4-D channels_last bf16");
+  const int N = (int)x.size(0), C = (int)x.size(1);
+  const int H = (int)x.size(2), W = (int)x.size(3);
+  TORCH_CHECK(C % 8 == 0, "maxpool3x3s2: C % 8");
+  const int Ho = (H + 1) / 2, Wo = (W + 1) / 2;
+  auto y = torch::empty({N, C, Ho, Wo},
+                        x.options().memory_format(torch::MemoryFormat::ChannelsLast));
+  auto idx = torch::empty({(long long)N * Ho * Wo * C},
+                          x.options().dtype(torch::kUInt8));
+  launch_maxpool3x3s2_fwd(x.data_ptr(), y.data_ptr(),
+                          idx.data_ptr<unsigned char>(), N, H, W, Ho, Wo, C,
+                          cur_stream());
+  return {y, idx};
+}
+
+torch::Tensor maxpool3x3s2_bwd(torch::Tensor dy, torch::Tensor idx,
+                               int64_t H, int64_t W) {
+  TORCH_CHECK(dy.is_cuda() && dy.dim() == 4 &&
+                  dy.scalar_type() == torch::kBFloat16 &&
+                  dy.is_contiguous(torch::MemoryFormat::ChannelsLast),
+              "maxpool3x3s2_bwd: 4-D channels_last bf16");
+  const int N = (int)dy.size(0), C = (int)dy.size(1);
+  const int Ho = (int)dy.size(2), Wo = (int)dy.size(3);
+  auto dx = torch::empty({N, C, (long long)H, (long long)W},
+                         dy.options().memory_format(torch::MemoryFormat::ChannelsLast));
+  launch_maxpool3x3s2_bwd(dy.data_ptr(), idx.data_ptr<unsigned char>(),
+                          dx.data_ptr(), N, (int)H, (int)W, Ho, Wo, C,
+                          cur_stream());
+  return dx;
+}
+
 torch::Tensor avgpool2x2_fwd(torch::Tensor x) {
   // x: 4-D channels_last bf16; 2x2 stride-2 ceil_mode pool
   TORCH_CHECK(x.is_cuda() && x.dim() == 4 &&
@@ -625,6 +686,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "stride-2 3x3 same-pad dgrad (parity implicit GEMM) -> dx2d [N*H*W, Cin]");
   m.def("conv3x3_small_fwd", &conv3x3_small_fwd,
         "small-channel 3x3 conv (deep stem) -> y2d [M, cout_real]");
+  m.def("repack_dgrad_w3", &repack_dgrad_w3,
+        "dgrad weight repack [Co,9Ci]->[Ci,9Co] (mode 0 rot / 1 s2 order)");
+  m.def("maxpool3x3s2_fwd", &maxpool3x3s2_fwd,
+        "3x3/s2/p1 max pool NHWC bf16 -> (y, tap idx)");
+  m.def("maxpool3x3s2_bwd", &maxpool3x3s2_bwd,
+        "3x3/s2/p1 max pool backward (bounded gather via tap idx)");
   m.def("avgpool2x2_bwd", &avgpool2x2_bwd, "2x2/s2 ceil avg pool backward");
   m.attr("_arch") = "gfx950";
 }

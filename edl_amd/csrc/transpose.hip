@@ -104,3 +104,65 @@ extern "C" void launch_shift9_transpose(const void* xp, void* out, int M, int C,
                      (const bf16*)xp, (bf16*)out, M, C, Mp, HW_out, W_out, Hp,
                      Wp, stride);
 }
+
+// One-pass dgrad-weight repack from the channels-last mirror:
+//   out[ci][s][co] = in[co][tap_map(s)][ci]
+// in: [Co, 9, Ci] bf16 (the channels-last bucket mirror viewed s-major);
+// out: [Ci, 9, Co] bf16. mode 0: tap_map(s) = 8-s (the 180-degree
+// rotation + Cin<->Cout transpose of the stride-1 dgrad weights — was a
+// flip kernel + permute copy + cast per conv per step); mode 1: the
+// stride-2 dgrad parity-class tap order (conv.py _S2D_TAP_ORDER).
+__constant__ int S2D_ORDER[9] = {4, 3, 5, 1, 7, 0, 2, 6, 8};
+// _S2D_TAP_ORDER[j] = (dy,dx) -> tap index dy*3+dx:
+// [(1,1)=4,(1,0)=3,(1,2)=5,(0,1)=1,(2,1)=7,(0,0)=0,(0,2)=2,(2,0)=6,(2,2)=8]
+
+extern "C" __global__ void repack_dgrad_w3_kernel(
+    const __hip_bfloat16* __restrict__ in, __hip_bfloat16* __restrict__ out,
+    const int Co, const int Ci, const int mode) {
+  __shared__ __hip_bfloat16 tile[64][64 + 8];
+  // grid: x = (Ci/64)*(Co/64) tiles, y = tap s
+  const int s = blockIdx.y;
+  const int tap = mode == 0 ? 8 - s : S2D_ORDER[s];
+  const int tiles_co = Co >> 6;
+  const int ci0 = (blockIdx.x / tiles_co) << 6;
+  const int co0 = (blockIdx.x % tiles_co) << 6;
+  const int lane = threadIdx.x & 63;
+  const int quad = threadIdx.x >> 6;
+  // load: rows = co (strided), cols = ci contiguous, 16 B per thread
+  {
+    const int c8 = lane & 7;        // which 8-ci chunk
+    const int rr = (lane >> 3) + quad * 8;  // row 0..31; two passes
+    for (int half = 0; half < 2; ++half) {
+      const int r = rr + half * 32;
+      const __hip_bfloat16* src =
+          in + ((long long)(co0 + r) * 9 + tap) * Ci + ci0 + c8 * 8;
+      *reinterpret_cast<uint4*>(
+          __builtin_assume_aligned(&tile[r][c8 * 8], 16)) =
+          *reinterpret_cast<const uint4*>(__builtin_assume_aligned(src, 16));
+    }
+  }
+  __syncthreads();
+  // store: rows = ci (strided), cols = co contiguous
+  {
+    const int c8 = lane & 7;
+    const int rr = (lane >> 3) + quad * 8;
+    for (int half = 0; half < 2; ++half) {
+      const int r = rr + half * 32;  // ci row
+      __hip_bfloat16* dst =
+          out + ((long long)(ci0 + r) * 9 + s) * Co + co0 + c8 * 8;
+      uint4 v;
+      __hip_bfloat16* pv = reinterpret_cast<__hip_bfloat16*>(&v);
+#pragma unroll
+      for (int k = 0; k < 8; ++k) pv[k] = tile[c8 * 8 + k][r];
+      *reinterpret_cast<uint4*>(__builtin_assume_aligned(dst, 16)) = v;
+    }
+  }
+}
+
+extern "C" void launch_repack_dgrad_w3(const void* in, void* out, int Co,
+                                       int Ci, int mode, hipStream_t s) {
+  const dim3 grid((Ci >> 6) * (Co >> 6), 9);
+  hipLaunchKernelGGL(repack_dgrad_w3_kernel, grid, dim3(256), 0, s,
+                     (const __hip_bfloat16*)in, (__hip_bfloat16*)out, Co, Ci,
+                     mode);
+}

@@ -34,6 +34,10 @@ class TeacherService:
         )
         self.model = model or build_model(model_name, num_classes=num_classes)
         self.model.eval().to(self.device)
+        for p in self.model.parameters():
+            # frozen: lets the conv layer cache its repacked weights
+            # forever instead of per (student) weight epoch
+            p.requires_grad_(False)
         self.use_bf16 = self.device.type == "cuda"
         if self.use_bf16:
             self.model.to(memory_format=torch.channels_last)

@@ -15,7 +15,7 @@ import torch.nn as nn
 
 from ..ops.bnrelu import BNAddReLU2d, BNReLU2d
 from ..ops.conv import Conv2dFast
-from ..ops.pool import AvgPool2x2
+from ..ops.pool import AvgPool2x2, MaxPool3x3s2
 
 
 class ConvBN(nn.Module):
@@ -105,7 +105,7 @@ class ResNetVd(nn.Module):
             ConvBN(3, 32, 3, stride=2),
             ConvBN(32, 32, 3),
             ConvBN(32, 64, 3),
-            nn.MaxPool2d(3, 2, padding=1),
+            MaxPool3x3s2(),
         )
         planes = [64, 128, 256, 512]
         cin = 64

@@ -9,6 +9,7 @@ import torch.nn as nn
 
 from ..ops.bnrelu import BNAddReLU2d, BNReLU2d
 from ..ops.conv import Conv2dFast
+from ..ops.pool import MaxPool3x3s2
 
 
 class ResNeXtBottleneck(nn.Module):
@@ -44,7 +45,7 @@ class ResNeXtWSL(nn.Module):
         super().__init__()
         self.conv1 = Conv2dFast(3, 64, 7, stride=2, padding=3, bias=False)
         self.bn1 = BNReLU2d(64)
-        self.maxpool = nn.MaxPool2d(3, 2, padding=1)
+        self.maxpool = MaxPool3x3s2()
         cin = 64
         stages = []
         for bi, (p, d) in enumerate(zip([64, 128, 256, 512], depths)):

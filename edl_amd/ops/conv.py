@@ -314,9 +314,14 @@ class Conv2dFast(nn.Conv2d):
         return None
 
     def _cached(self, key, fn):
+        # frozen weights (teacher serving: requires_grad False) cache
+        # FOREVER — the global weight epoch is bumped by the STUDENT's
+        # optimizer and would otherwise re-repack every teacher conv's
+        # weights each student step (~400 MB of copies for ResNeXt101)
+        epoch = _weight_epoch if self.weight.requires_grad else -1
         cache = getattr(self, "_w_cache", None)
-        if cache is None or cache[0] != _weight_epoch:
-            cache = (_weight_epoch, {})
+        if cache is None or cache[0] != epoch:
+            cache = (epoch, {})
             self._w_cache = cache
         d = cache[1]
         if key not in d:
@@ -355,7 +360,13 @@ class Conv2dFast(nn.Conv2d):
                 w3 = mb.permute(0, 2, 3, 1).reshape(co, 9 * ci)
             else:
                 w3 = self._cached("w3", lambda: _repack_w3(wsrc))
-            if self.stride[0] == 1:
+            if cl and mb is not None:
+                # one-pass kernel repack from the s-major mirror view
+                mode = 0 if self.stride[0] == 1 else 1
+                w3rot = self._cached(
+                    "w3rot" if mode == 0 else "w3s2d",
+                    lambda: ext().repack_dgrad_w3(w3, ci, mode))
+            elif self.stride[0] == 1:
                 w3rot = self._cached("w3rot", lambda: _repack_w3(
                     wsrc.permute(1, 0, 2, 3).flip(2, 3)))
             else:

@@ -379,3 +379,26 @@ def test_stem_conv0_wgrad_no_dx():
     y.float().sum().backward()
     assert conv.weight.grad is not None
     assert torch.isfinite(conv.weight.grad.float()).all()
+
+
+@pytest.mark.parametrize("mode", [0, 1])
+@pytest.mark.parametrize("co,ci", [(64, 64), (128, 64), (256, 128)])
+def test_repack_dgrad_w3_kernel(mode, co, ci):
+    """One-pass dgrad weight repack vs the torch reference chain."""
+    from edl_amd import ops
+    from edl_amd.ops.conv import _repack_w3, _repack_w3_s2dgrad
+
+    torch.manual_seed(17)
+    w = torch.randn(co, ci, 3, 3, device="cuda")
+    w_smaj = w.permute(0, 2, 3, 1).reshape(co, 9 * ci).to(
+        torch.bfloat16).contiguous()
+    got = ops.ext().repack_dgrad_w3(w_smaj, ci, mode)
+    wb = w.to(torch.bfloat16).float().to(torch.bfloat16)  # match rounding
+    if mode == 0:
+        ref = _repack_w3(w.to(torch.bfloat16).float()
+                         .permute(1, 0, 2, 3).flip(2, 3))
+    else:
+        ref = _repack_w3_s2dgrad(w.to(torch.bfloat16).float())
+    assert got.shape == ref.shape
+    assert torch.equal(got.float(), ref.float()), \
+        (got.float() - ref.float()).abs().max().item()
