@@ -370,8 +370,21 @@ extern "C" __global__ void pad_nhwc_cpad_kernel(
     const int h = hp - 1, w = wp - 1;
     const bool in = h >= 0 && h < H && w >= 0 && w < W;
     const bf16* src = x + (((long long)n * H + h) * W + w) * Creal;
-    for (int c = 0; c < Cpad; ++c)
-      dst[c] = (in && c < Creal) ? src[c] : __float2bfloat16(0.0f);
+    // build the padded row in registers, store 16 B octets (the scalar
+    // per-channel store loop measured ~0.7 TB/s)
+    for (int o = 0; o < Cpad; o += 8) {
+      uint4 raw = {0, 0, 0, 0};
+      ushort* u = reinterpret_cast<ushort*>(&raw);
+      if (in) {
+#pragma unroll 8
+        for (int k = 0; k < 8; ++k) {
+          const int c = o + k;
+          if (c < Creal)
+            u[k] = __hip_bfloat16_raw(src[c]).x;
+        }
+      }
+      *reinterpret_cast<uint4*>(__builtin_assume_aligned(dst + o, 16)) = raw;
+    }
   }
 }
 
