@@ -38,6 +38,8 @@ def spawn_agent(store_ep, job_id, idx, nodes_range, log_dir, gpu=None):
         "EDL_LEADER_RETRY": "0.5",
         "EDL_STEP_MARKER": MARKER,
         "CUDA_VISIBLE_DEVICES": str(gpu) if gpu is not None else "",
+        **({"EDL_FORCE_BACKEND": "gloo"}
+           if os.environ.get("EDL_SHARE_GPU0") == "1" else {}),
     })
     return subprocess.Popen(
         [sys.executable, "-m", "edl_amd.launch",
@@ -81,7 +83,12 @@ def main():
     ap.add_argument("--drop-to", type=int, default=1)
     ap.add_argument("--rejoin", action="store_true")
     ap.add_argument("--use-gpus", action="store_true")
+    ap.add_argument("--share-gpu0", action="store_true",
+                    help="all agents on cuda:0 with gloo collectives "
+                         "(1-GPU box: RCCL refuses duplicate devices)")
     args = ap.parse_args()
+    if args.share_gpu0:
+        os.environ["EDL_SHARE_GPU0"] = "1"
 
     if os.path.exists(MARKER):
         os.remove(MARKER)
@@ -93,7 +100,7 @@ def main():
     try:
         for i in range(args.start):
             agents.append(spawn_agent(srv.endpoint, job, i, rng, log_dir,
-                                      gpu=i if args.use_gpus else None))
+                                      gpu=0 if args.share_gpu0 else (i if args.use_gpus else None)))
         wait_world_steps(args.start, args.start, 300)
         time.sleep(1.0)  # steady state
 
@@ -115,7 +122,7 @@ def main():
             for i in range(args.drop_to, args.start):
                 agents.append(spawn_agent(srv.endpoint, job, 100 + i, rng,
                                           log_dir,
-                                          gpu=i if args.use_gpus else None))
+                                          gpu=0 if args.share_gpu0 else (i if args.use_gpus else None)))
             t_back = wait_world_steps(args.start, args.start, 300,
                                       after_ts=t_join)
             result["resize_up_s"] = round(t_back - t_join, 2)
