@@ -63,7 +63,17 @@ __device__ __forceinline__ int tn_swz(int r) {
 // read — the same glds staging as the plain kernel with a computed row
 // base. This removes the shift9_transpose + transpose_pad(dy)
 // materializations from conv3x3 wgrad entirely.
-template <int BN1, int BN2, int KW, bool G3 = false>
+// TR: fragment reads via ds_read_b64_tr_b16 (gfx950 hardware transpose
+// read). The u16 column-walk costs 128 ds_read_u16 + 64 v_perm per
+// 32-MFMA block — the kernel measured 55% issue-stall (PMC r2c14). With
+// TR the LDS image is subtiled per 16-col n-block as two 512-element
+// regions (region r holds taps k%8 in [4r, 4r+4)):
+//   elem(k, n) = nb*1024 + (k&4)*128 + (k>>3)*64 + (k&3)*16 + (n&15)
+// and ONE tr read per 4-k half delivers each lane its column — 32 reads,
+// zero packing VALU. Staging stays glds lane-linear: chunk ch = region
+// (nb = ch>>1, r = ch&1), lane l holds k = (l>>3)*8 + 4r + ((l>>1)&3),
+// n = nb*16 + (l&1)*8.
+template <int BN1, int BN2, int KW, bool G3 = false, bool TR = false>
 __global__ __launch_bounds__(256, 2) void gemm_tn_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B,
     float* __restrict__ C, const int N1, const int N2, const int K,
@@ -127,6 +137,14 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_kernel(
   // ---- staging: group's GW wave(s) stage one [BK x BN] slab pair ----
   // chunk = 1 KiB = (64/SLOTS) k-rows x SLOTS slots; lane-linear dest,
   // swizzled source (gslot = slot ^ tn_swz(r))
+  // TR source mapping: chunk ch, lane l -> (k, n-offset) of the subtiled
+  // image (see template comment)
+  auto tr_kn = [&](int ch, int& k, int& noff) {
+    const int nb = ch >> 1, reg = ch & 1;
+    k = (lane >> 3) * 8 + reg * 4 + ((lane >> 1) & 3);
+    noff = nb * 16 + (lane & 1) * 8;
+  };
+
   auto stage = [&](int buf, int ct) {
     const long long k0 = (long long)ct * BK;
     char* abase = gbase + buf * AB;
@@ -135,9 +153,16 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_kernel(
 #pragma unroll
     for (int i = 0; i < ACH / GW; ++i) {
       const int ch = wg * (ACH / GW) + i;
-      const int r = ch * (64 / SLOTS_A) + lane / SLOTS_A;
-      const int gslot = (lane % SLOTS_A) ^ tn_swz(r);
-      const bf16* src = A + (k0 + r) * N1 + t1 + gslot * 8;
+      const bf16* src;
+      if constexpr (TR) {
+        int k, noff;
+        tr_kn(ch, k, noff);
+        src = A + (k0 + k) * N1 + t1 + noff;
+      } else {
+        const int r = ch * (64 / SLOTS_A) + lane / SLOTS_A;
+        const int gslot = (lane % SLOTS_A) ^ tn_swz(r);
+        src = A + (k0 + r) * N1 + t1 + gslot * 8;
+      }
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) void*)src,
           (__attribute__((address_space(3))) void*)(abase + ch * 1024), 16, 0, 0);
@@ -145,9 +170,16 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_kernel(
 #pragma unroll
     for (int i = 0; i < BCH / GW; ++i) {
       const int ch = wg * (BCH / GW) + i;
-      const int r = ch * (64 / SLOTS_B) + lane / SLOTS_B;
-      const int gslot = (lane % SLOTS_B) ^ tn_swz(r);
-      const bf16* src = b_row(k0 + r) + gslot * 8;
+      const bf16* src;
+      if constexpr (TR) {
+        int k, noff;
+        tr_kn(ch, k, noff);
+        src = b_row(k0 + k) + noff;
+      } else {
+        const int r = ch * (64 / SLOTS_B) + lane / SLOTS_B;
+        const int gslot = (lane % SLOTS_B) ^ tn_swz(r);
+        src = b_row(k0 + r) + gslot * 8;
+      }
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) void*)src,
           (__attribute__((address_space(3))) void*)(bbase + ch * 1024), 16, 0, 0);
@@ -164,10 +196,15 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_kernel(
 #pragma unroll
     for (int i = 0; i < ACH / GW; ++i) {
       const int ch = wg * (ACH / GW) + i;
-      const int r = ch * (64 / SLOTS_A) + lane / SLOTS_A;
-      const int gslot = (lane % SLOTS_A) ^ tn_swz(r);
+      int r, noff;
+      if constexpr (TR) {
+        tr_kn(ch, r, noff);
+      } else {
+        r = ch * (64 / SLOTS_A) + lane / SLOTS_A;
+        noff = ((lane % SLOTS_A) ^ tn_swz(r)) * 8;
+      }
       bf16x8 v = {};
-      if (k0 + r < K) v = *(const bf16x8*)(A + (k0 + r) * N1 + t1 + gslot * 8);
+      if (k0 + r < K) v = *(const bf16x8*)(A + (k0 + r) * N1 + t1 + noff);
       *(__attribute__((address_space(3))) bf16x8*)(
           (__attribute__((address_space(3))) char*)(abase) + ch * 1024 +
           lane * 16) = v;
@@ -175,10 +212,15 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_kernel(
 #pragma unroll
     for (int i = 0; i < BCH / GW; ++i) {
       const int ch = wg * (BCH / GW) + i;
-      const int r = ch * (64 / SLOTS_B) + lane / SLOTS_B;
-      const int gslot = (lane % SLOTS_B) ^ tn_swz(r);
+      int r, noff;
+      if constexpr (TR) {
+        tr_kn(ch, r, noff);
+      } else {
+        r = ch * (64 / SLOTS_B) + lane / SLOTS_B;
+        noff = ((lane % SLOTS_B) ^ tn_swz(r)) * 8;
+      }
       bf16x8 v = {};
-      if (k0 + r < K) v = *(const bf16x8*)(b_row(k0 + r) + gslot * 8);
+      if (k0 + r < K) v = *(const bf16x8*)(b_row(k0 + r) + noff);
       *(__attribute__((address_space(3))) bf16x8*)(
           (__attribute__((address_space(3))) char*)(bbase) + ch * 1024 +
           lane * 16) = v;
@@ -186,36 +228,45 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_kernel(
   };
 
   // ---- transposed fragment reads: 8 swizzled u16 column-walk ----
+  // All addressing is hoisted 32-BIT integer offsets against one
+  // addrspace(3) base: the lambda-local generic-pointer form cost a
+  // v_mad_u64/v_lshl_add_u64 chain PER ELEMENT (PMC: 55% of wave cycles
+  // were issue stalls). rb is a multiple of 8, so tn_swz(rb+j) =
+  // j ^ ((rb>>3)&1)<<2 — the per-j XOR folds to one constant.
+  const __attribute__((address_space(3))) char* lds3 =
+      (const __attribute__((address_space(3))) char*)smem;
+  const unsigned gbase_off = (unsigned)(grp * 2 * AB);
   auto read_a = [&](int buf, int mf, int kk) -> bf16x8 {
-    const __attribute__((address_space(3))) char* abase =
-        (const __attribute__((address_space(3))) char*)(gbase + buf * AB);
     const int n = wm + mf * 16 + (lane & 15);
-    const int slot = n >> 3, boff = (n & 7) * 2;
+    const int slot = n >> 3;
     const int rb = kk * 32 + (lane >> 4) * 8;
+    const unsigned s0 = (unsigned)(slot ^ (((rb >> 3) & 1) << 2));
+    const unsigned base = gbase_off + (unsigned)buf * AB +
+                          (unsigned)rb * (SLOTS_A * 16) +
+                          (unsigned)((n & 7) * 2);
     bf16x8 v;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      const int r = rb + j;
-      v[j] = *(const __attribute__((address_space(3))) __bf16*)(
-          abase + r * (SLOTS_A * 16) + ((slot ^ tn_swz(r)) << 4) +
-          boff);
+      const unsigned off = base + (unsigned)j * (SLOTS_A * 16) +
+                           ((s0 ^ (unsigned)j) << 4);
+      v[j] = *(const __attribute__((address_space(3))) __bf16*)(lds3 + off);
     }
     return v;
   };
   auto read_b = [&](int buf, int nf, int kk) -> bf16x8 {
-    const __attribute__((address_space(3))) char* bbase =
-        (const __attribute__((address_space(3))) char*)(gbase + buf * AB +
-                                                        A_BYTES);
     const int n = wn + nf * 16 + (lane & 15);
-    const int slot = n >> 3, boff = (n & 7) * 2;
+    const int slot = n >> 3;
     const int rb = kk * 32 + (lane >> 4) * 8;
+    const unsigned s0 = (unsigned)(slot ^ (((rb >> 3) & 1) << 2));
+    const unsigned base = gbase_off + (unsigned)buf * AB + A_BYTES +
+                          (unsigned)rb * (SLOTS_B * 16) +
+                          (unsigned)((n & 7) * 2);
     bf16x8 v;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      const int r = rb + j;
-      v[j] = *(const __attribute__((address_space(3))) __bf16*)(
-          bbase + r * (SLOTS_B * 16) + ((slot ^ tn_swz(r)) << 4) +
-          boff);
+      const unsigned off = base + (unsigned)j * (SLOTS_B * 16) +
+                           ((s0 ^ (unsigned)j) << 4);
+      v[j] = *(const __attribute__((address_space(3))) __bf16*)(lds3 + off);
     }
     return v;
   };
@@ -230,6 +281,18 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_kernel(
       stage_tail(buf, ct);
   };
 
+  // TR per-lane base byte addresses (buf/kk/frag offsets added per read;
+  // all uniform parts folded here — the read is ONE ds_read_b64_tr_b16)
+  const unsigned tr_a_base =
+      gbase_off + (unsigned)((wm >> 4) * 2048) + (unsigned)lane * 8;
+  const unsigned tr_b_base =
+      gbase_off + A_BYTES + (unsigned)((wn >> 4) * 2048) + (unsigned)lane * 8;
+  auto tr_read = [&](unsigned addr) -> unsigned long long {
+    unsigned long long d;
+    asm volatile("ds_read_b64_tr_b16 %0, %1" : "=v"(d) : "v"(addr));
+    return d;
+  };
+
   do_stage(0, 0);
   __syncthreads();  // implicit vmcnt(0)+lgkmcnt(0) drains glds/ds stores
 
@@ -240,10 +303,38 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_kernel(
 #pragma unroll
       for (int kk = 0; kk < 2; ++kk) {
         bf16x8 a[4], b[4];
+        if constexpr (TR) {
+          union Q { unsigned long long q[2]; bf16x8 v; };
+          Q qa[4], qb[4];
+          const unsigned ab =
+              tr_a_base + (unsigned)cur * AB + (unsigned)(kk * 512);
+          const unsigned bb =
+              tr_b_base + (unsigned)cur * AB + (unsigned)(kk * 512);
 #pragma unroll
-        for (int mf = 0; mf < 4; ++mf) a[mf] = read_a(cur, mf, kk);
+          for (int mf = 0; mf < 4; ++mf) {
+            qa[mf].q[0] = tr_read(ab + mf * 2048);
+            qa[mf].q[1] = tr_read(ab + mf * 2048 + 1024);
+          }
 #pragma unroll
-        for (int nf = 0; nf < 4; ++nf) b[nf] = read_b(cur, nf, kk);
+          for (int nf = 0; nf < 4; ++nf) {
+            qb[nf].q[0] = tr_read(bb + nf * 2048);
+            qb[nf].q[1] = tr_read(bb + nf * 2048 + 1024);
+          }
+          // hipcc does not count asm LDS reads: drain them, then fence
+          // the MFMAs below the wait (guide rule 18)
+          asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+          __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+          for (int f = 0; f < 4; ++f) {
+            a[f] = qa[f].v;
+            b[f] = qb[f].v;
+          }
+        } else {
+#pragma unroll
+          for (int mf = 0; mf < 4; ++mf) a[mf] = read_a(cur, mf, kk);
+#pragma unroll
+          for (int nf = 0; nf < 4; ++nf) b[nf] = read_b(cur, nf, kk);
+        }
 #pragma unroll
         for (int mf = 0; mf < 4; ++mf)
 #pragma unroll
@@ -318,6 +409,14 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_kernel(
   }
 }
 
+static bool tn_use_tr() {
+  static const bool v = []() {
+    const char* e = getenv("EDL_TN_TR");
+    return !e || atoi(e) != 0;  // default ON (A/B: EDL_TN_TR=0)
+  }();
+  return v;
+}
+
 extern "C" void launch_gemm_tn_splitk(const void* A, const void* B, float* C,
                                       int N1, int N2, int K, int splitk,
                                       hipStream_t s) {
@@ -326,18 +425,15 @@ extern "C" void launch_gemm_tn_splitk(const void* A, const void* B, float* C,
   const dim3 grid((N1 / BN1) * (N2 / BN2), splitk);
   const int KW = 4 / ((BN1 / 64) * (BN2 / 64));
   const int lds = KW * 2 * (64 * BN1 * 2 + 64 * BN2 * 2);
-  if (b1 && b2)
-    hipLaunchKernelGGL((gemm_tn_kernel<128, 128, 1>), grid, dim3(256), lds, s,
-                       (const bf16*)A, (const bf16*)B, C, N1, N2, K);
-  else if (b1)
-    hipLaunchKernelGGL((gemm_tn_kernel<128, 64, 2>), grid, dim3(256), lds, s,
-                       (const bf16*)A, (const bf16*)B, C, N1, N2, K);
-  else if (b2)
-    hipLaunchKernelGGL((gemm_tn_kernel<64, 128, 2>), grid, dim3(256), lds, s,
-                       (const bf16*)A, (const bf16*)B, C, N1, N2, K);
-  else
-    hipLaunchKernelGGL((gemm_tn_kernel<64, 64, 4>), grid, dim3(256), lds, s,
-                       (const bf16*)A, (const bf16*)B, C, N1, N2, K);
+#define TN_LAUNCH(B1, B2, W)                                                  do {                                                                          if (tn_use_tr())                                                              hipLaunchKernelGGL((gemm_tn_kernel<B1, B2, W, false, true>), grid,                             dim3(256), lds, s, (const bf16*)A, (const bf16*)B,                          C, N1, N2, K);                                         else                                                                          hipLaunchKernelGGL((gemm_tn_kernel<B1, B2, W, false, false>), grid,                            dim3(256), lds, s, (const bf16*)A, (const bf16*)B,                          C, N1, N2, K);                                       } while (0)
+  if (b1 && b2) TN_LAUNCH(128, 128, 1);
+  else if (b1) TN_LAUNCH(128, 64, 2);
+  else if (b2) TN_LAUNCH(64, 128, 2);
+  else  // KW=4 spills 28 B scratch under TR (16 live u64 reads + acc)
+    hipLaunchKernelGGL((gemm_tn_kernel<64, 64, 4, false, false>), grid,
+                       dim3(256), lds, s, (const bf16*)A, (const bf16*)B, C,
+                       N1, N2, K);
+#undef TN_LAUNCH
 }
 
 extern "C" void launch_gemm_tn3x3_splitk(const void* dy, const void* xpad,
@@ -353,20 +449,13 @@ extern "C" void launch_gemm_tn3x3_splitk(const void* dy, const void* xpad,
   const dim3 grid((Cout / BN1) * (N2 / BN2), splitk);
   const int lds = (4 / ((BN1 / 64) * (BN2 / 64))) * 2 *
                   (64 * BN1 * 2 + 64 * BN2 * 2);
-  if (b1 && b2)
-    hipLaunchKernelGGL((gemm_tn_kernel<128, 128, 1, true>), grid, dim3(256),
-                       lds, s, (const bf16*)dy, (const bf16*)xpad, C, Cout, N2,
-                       M, Ho, Wo, Hp, Wp, Cin, stride, perm);
-  else if (b1)
-    hipLaunchKernelGGL((gemm_tn_kernel<128, 64, 2, true>), grid, dim3(256),
-                       lds, s, (const bf16*)dy, (const bf16*)xpad, C, Cout, N2,
-                       M, Ho, Wo, Hp, Wp, Cin, stride, perm);
-  else if (b2)
-    hipLaunchKernelGGL((gemm_tn_kernel<64, 128, 2, true>), grid, dim3(256),
-                       lds, s, (const bf16*)dy, (const bf16*)xpad, C, Cout, N2,
-                       M, Ho, Wo, Hp, Wp, Cin, stride, perm);
+#define TN3_LAUNCH(B1, B2, W)                                                   do {                                                                            if (tn_use_tr())                                                                hipLaunchKernelGGL((gemm_tn_kernel<B1, B2, W, true, true>), grid,                                dim3(256), lds, s, (const bf16*)dy,                                           (const bf16*)xpad, C, Cout, N2, M, Ho, Wo, Hp, Wp,                            Cin, stride, perm);                                      else                                                                            hipLaunchKernelGGL((gemm_tn_kernel<B1, B2, W, true, false>), grid,                               dim3(256), lds, s, (const bf16*)dy,                                           (const bf16*)xpad, C, Cout, N2, M, Ho, Wo, Hp, Wp,                            Cin, stride, perm);                                    } while (0)
+  if (b1 && b2) TN3_LAUNCH(128, 128, 1);
+  else if (b1) TN3_LAUNCH(128, 64, 2);
+  else if (b2) TN3_LAUNCH(64, 128, 2);
   else
-    hipLaunchKernelGGL((gemm_tn_kernel<64, 64, 4, true>), grid, dim3(256), lds,
-                       s, (const bf16*)dy, (const bf16*)xpad, C, Cout, N2, M,
-                       Ho, Wo, Hp, Wp, Cin, stride, perm);
+    hipLaunchKernelGGL((gemm_tn_kernel<64, 64, 4, true, false>), grid,
+                       dim3(256), lds, s, (const bf16*)dy, (const bf16*)xpad,
+                       C, Cout, N2, M, Ho, Wo, Hp, Wp, Cin, stride, perm);
+#undef TN3_LAUNCH
 }
