@@ -479,8 +479,11 @@ class Conv2dFast(nn.Conv2d):
                 mb = getattr(self.weight, "_edl_bf16", None)
                 if mb is not None:  # engine bucket mirror: free bf16 view
                     w_bf16 = mb.view(self.out_channels, c)
-                    wt_t = self._cached("wt_t", lambda: mb.view(
-                        self.out_channels, c).t().contiguous())
+                    # LDS-tiled transpose kernel instead of torch's
+                    # permute-copy (one per 1x1 layer per step);
+                    # transpose_pad is exact here since Cout % 64 == 0
+                    wt_t = self._cached("wt_t", lambda: ext().transpose_pad(
+                        mb.view(self.out_channels, c)))
                 else:
                     w_bf16 = self._cached("w_bf16", lambda: self.weight.detach()
                                           .view(self.out_channels, c)
