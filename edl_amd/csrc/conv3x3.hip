@@ -65,13 +65,20 @@ extern "C" __global__ void pad_nhwc_kernel(
 // K-range over grid.y and atomically fold fp32 partials into C_part
 // (zeroed by the caller; cast to bf16 afterwards) — measured 56 us/call
 // at 52 blocks before, CU-starved.
+// BN_PART: per-channel sum/sumsq partials of the (bf16-rounded) OUTPUT,
+// one row per m-tile ([tiles_m, 2N]: sums then sumsq) — the following
+// BatchNorm skips its stats kernel entirely (a full activation re-read,
+// measured ~0.53 ms/step). Folded in the epilogue via LDS atomics into
+// the dead staging buffers; disjoint column ranges per block, so the
+// partial rows need no zero-init and no global atomics.
 template <int BM, int BN, int WAVES_M, int WAVES_N, bool GROUPED = false>
 __global__ __launch_bounds__(256, 2) void conv3x3_kernel(
     const bf16* __restrict__ XP, const bf16* __restrict__ B,
     bf16* __restrict__ C_out, const int M, const int N, const int Cin,
     const int HW_out, const int W_out, const int Hp, const int Wp,
     const int stride_hw, const int gw = 0,
-    float* __restrict__ C_part = nullptr) {
+    float* __restrict__ C_part = nullptr,
+    float* __restrict__ bn_part = nullptr) {
   constexpr int BK = 64;
   constexpr int A_BYTES = BM * BK * 2;
   constexpr int B_BYTES = BN * BK * 2;
@@ -192,6 +199,8 @@ __global__ __launch_bounds__(256, 2) void conv3x3_kernel(
 
   const int cn = lane & 15;
   const int r4 = (lane >> 4) * 4;
+  // local per-column stats accumulators (summed into LDS afterwards)
+  float ls[4] = {0, 0, 0, 0}, lq[4] = {0, 0, 0, 0};
 #pragma unroll
   for (int mf = 0; mf < 4; ++mf) {
 #pragma unroll
@@ -206,10 +215,38 @@ __global__ __launch_bounds__(256, 2) void conv3x3_kernel(
         } else {
           bf16* crow = C_out + (long long)m * N + n0 + wn + cn;
 #pragma unroll
-          for (int nf = 0; nf < 4; ++nf)
-            crow[nf * 16] = __float2bfloat16(acc[mf][nf][reg]);
+          for (int nf = 0; nf < 4; ++nf) {
+            const bf16 yb = __float2bfloat16(acc[mf][nf][reg]);
+            crow[nf * 16] = yb;
+            if (bn_part != nullptr) {
+              // stats must see the bf16-ROUNDED value BN will read
+              const float yv = __bfloat162float(yb);
+              ls[nf] += yv;
+              lq[nf] = fmaf(yv, yv, lq[nf]);
+            }
+          }
         }
       }
+    }
+  }
+
+  if (bn_part != nullptr) {
+    // the staging buffers are dead past the last k-loop barrier
+    float* bsum = (float*)smem;
+    __syncthreads();
+    for (int i = threadIdx.x; i < 2 * BN; i += blockDim.x) bsum[i] = 0.0f;
+    __syncthreads();
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+      atomicAdd(&bsum[wn + nf * 16 + cn], ls[nf]);
+      atomicAdd(&bsum[BN + wn + nf * 16 + cn], lq[nf]);
+    }
+    __syncthreads();
+    const int mtile = (bid / (N / BN));
+    float* dst = bn_part + (long long)mtile * 2 * N + n0;
+    for (int i = threadIdx.x; i < BN; i += blockDim.x) {
+      dst[i] = bsum[i];
+      dst[N + i] = bsum[BN + i];
     }
   }
 }
@@ -228,7 +265,7 @@ __global__ __launch_bounds__(256, 2) void conv3x3_small_kernel(
     const bf16* __restrict__ XP, const bf16* __restrict__ B,
     bf16* __restrict__ C_out, const int M, const int Cout_real,
     const int HW_out, const int W_out, const int Hp, const int Wp,
-    const int stride_hw) {
+    const int stride_hw, float* __restrict__ bn_part = nullptr) {
   constexpr int BM = 256, BN = 64;
   constexpr int TPK = 64 / CPT;                    // taps per 64-K step
   constexpr int TAPS_PAD = CPT == 16 ? 12 : (CPT == 32 ? 10 : 9);
@@ -337,6 +374,7 @@ __global__ __launch_bounds__(256, 2) void conv3x3_small_kernel(
 
   const int cn = lane & 15;
   const int r4 = (lane >> 4) * 4;
+  float ls[4] = {0, 0, 0, 0}, lq[4] = {0, 0, 0, 0};
 #pragma unroll
   for (int mf = 0; mf < 4; ++mf) {
 #pragma unroll
@@ -346,9 +384,35 @@ __global__ __launch_bounds__(256, 2) void conv3x3_small_kernel(
         bf16* crow = C_out + (long long)m * Cout_real + cn;
 #pragma unroll
         for (int nf = 0; nf < 4; ++nf)
-          if (nf * 16 + cn < Cout_real)
-            crow[nf * 16] = __float2bfloat16(acc[mf][nf][reg]);
+          if (nf * 16 + cn < Cout_real) {
+            const bf16 yb = __float2bfloat16(acc[mf][nf][reg]);
+            crow[nf * 16] = yb;
+            if (bn_part != nullptr) {
+              const float yv = __bfloat162float(yb);
+              ls[nf] += yv;
+              lq[nf] = fmaf(yv, yv, lq[nf]);
+            }
+          }
       }
+    }
+  }
+
+  if (bn_part != nullptr) {  // see conv3x3_kernel BN_PART comment
+    float* bsum = (float*)smem;
+    __syncthreads();
+    for (int i = threadIdx.x; i < 2 * BN; i += blockDim.x) bsum[i] = 0.0f;
+    __syncthreads();
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf)
+      if (nf * 16 + cn < Cout_real) {
+        atomicAdd(&bsum[nf * 16 + cn], ls[nf]);
+        atomicAdd(&bsum[BN + nf * 16 + cn], lq[nf]);
+      }
+    __syncthreads();
+    float* dst = bn_part + (long long)bid * 2 * Cout_real;
+    for (int i = threadIdx.x; i < Cout_real; i += blockDim.x) {
+      dst[i] = bsum[i];
+      dst[Cout_real + i] = bsum[BN + i];
     }
   }
 }
@@ -401,14 +465,15 @@ extern "C" void launch_pad_nhwc_cpad(const void* x, void* xp, int Nimg, int H,
 extern "C" void launch_conv3x3_small(const void* xp, const void* w3s, void* y,
                                      int M, int Cout_real, int cpt, int HW_out,
                                      int W_out, int Hp, int Wp, int stride,
-                                     hipStream_t s) {
+                                     float* bn_part, hipStream_t s) {
   constexpr int BM = 256;
   const int grid = (M + BM - 1) / BM;
   const int lds_bytes = 2 * (BM * 64 * 2 + 64 * 64 * 2);
 #define SCASE(CPT)                                                          \
   hipLaunchKernelGGL((conv3x3_small_kernel<CPT>), dim3(grid), dim3(256),    \
                      lds_bytes, s, (const bf16*)xp, (const bf16*)w3s,       \
-                     (bf16*)y, M, Cout_real, HW_out, W_out, Hp, Wp, stride)
+                     (bf16*)y, M, Cout_real, HW_out, W_out, Hp, Wp, stride, \
+                     bn_part)
   if (cpt == 16) SCASE(16);
   else if (cpt == 32) SCASE(32);
   else SCASE(64);
@@ -638,12 +703,18 @@ extern "C" int conv3x3_pick_splitk(int M, int Cout, int Cin) {
   return sk < 1 ? 1 : sk;
 }
 
+extern "C" int conv3x3_tiles_m(int M, int Cout) {
+  const int bm = Cout % 128 == 0 ? 128 : 256;
+  return (M + bm - 1) / bm;
+}
+
 extern "C" void launch_conv3x3(const void* xp, const void* w3, void* y, int M,
                                int Cout, int Cin, int HW_out, int W_out, int Hp,
                                int Wp, int stride, float* cpart, int splitk,
-                               hipStream_t s) {
+                               float* bn_part, hipStream_t s) {
   if (splitk < 1) splitk = 1;
   if (cpart == nullptr) splitk = 1;
+  if (splitk > 1) bn_part = nullptr;  // stats need the final values
   if (Cout % 128 == 0) {
     constexpr int BM = 128, BN = 128;
     const int grid = ((M + BM - 1) / BM) * (Cout / BN);
@@ -651,7 +722,8 @@ extern "C" void launch_conv3x3(const void* xp, const void* w3, void* y, int M,
     hipLaunchKernelGGL((conv3x3_kernel<BM, BN, 2, 2>), dim3(grid, splitk),
                        dim3(256), lds_bytes, s, (const bf16*)xp,
                        (const bf16*)w3, (bf16*)y, M, Cout, Cin, HW_out, W_out,
-                       Hp, Wp, stride, 0, splitk > 1 ? cpart : nullptr);
+                       Hp, Wp, stride, 0, splitk > 1 ? cpart : nullptr,
+                       bn_part);
   } else {
     constexpr int BM = 256, BN = 64;
     const int grid = ((M + BM - 1) / BM) * (Cout / BN);
@@ -659,6 +731,7 @@ extern "C" void launch_conv3x3(const void* xp, const void* w3, void* y, int M,
     hipLaunchKernelGGL((conv3x3_kernel<BM, BN, 4, 1>), dim3(grid, splitk),
                        dim3(256), lds_bytes, s, (const bf16*)xp,
                        (const bf16*)w3, (bf16*)y, M, Cout, Cin, HW_out, W_out,
-                       Hp, Wp, stride, 0, splitk > 1 ? cpart : nullptr);
+                       Hp, Wp, stride, 0, splitk > 1 ? cpart : nullptr,
+                       bn_part);
   }
 }

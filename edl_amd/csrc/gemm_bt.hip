@@ -31,7 +31,11 @@ template <int BM, int BN, int WAVES_M, int WAVES_N, bool SPLITK = false,
           bool V2 = false>
 __global__ __launch_bounds__(256, 2) void gemm_bt_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B,
-    void* __restrict__ C_any, const int M, const int N, const int K) {
+    void* __restrict__ C_any, const int M, const int N, const int K,
+    float* __restrict__ bn_part = nullptr) {
+  // bn_part: per-channel sum/sumsq partials of the bf16-rounded output,
+  // [tiles_m, 2N] — lets the following BatchNorm skip its stats kernel
+  // (see conv3x3.hip BN_PART)
   // SPLITK: grid.y k-slices; each block atomically folds its fp32 partial
   // tile into C (fp32, pre-zeroed). Wgrad-shaped GEMMs (tiny [Cout, Cin]
   // output, K = M ~ 1e5) otherwise serialize on 1-2 blocks.
@@ -207,6 +211,7 @@ __global__ __launch_bounds__(256, 2) void gemm_bt_kernel(
   // row=(lane>>4)*4+reg ----
   const int cn = lane & 15;
   const int r4 = (lane >> 4) * 4;
+  float ls[4] = {0, 0, 0, 0}, lq[4] = {0, 0, 0, 0};
 #pragma unroll
   for (int mf = 0; mf < 4; ++mf) {
 #pragma unroll
@@ -223,10 +228,34 @@ __global__ __launch_bounds__(256, 2) void gemm_bt_kernel(
           bf16* crow = C + (long long)m * N + n0 + wn + cn;
 #pragma unroll
           for (int nf = 0; nf < 4; ++nf) {
-            crow[nf * 16] = __float2bfloat16(acc[mf][nf][reg]);
+            const bf16 yb = __float2bfloat16(acc[mf][nf][reg]);
+            crow[nf * 16] = yb;
+            if (bn_part != nullptr) {
+              const float yv = __bfloat162float(yb);
+              ls[nf] += yv;
+              lq[nf] = fmaf(yv, yv, lq[nf]);
+            }
           }
         }
       }
+    }
+  }
+
+  if (!SPLITK && bn_part != nullptr) {
+    float* bsum = (float*)smem;  // staging buffers dead past the loop
+    __syncthreads();
+    for (int i = threadIdx.x; i < 2 * BN; i += blockDim.x) bsum[i] = 0.0f;
+    __syncthreads();
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+      atomicAdd(&bsum[wn + nf * 16 + cn], ls[nf]);
+      atomicAdd(&bsum[BN + wn + nf * 16 + cn], lq[nf]);
+    }
+    __syncthreads();
+    float* dst = bn_part + (long long)tile_m * 2 * N + n0;
+    for (int i = threadIdx.x; i < BN; i += blockDim.x) {
+      dst[i] = bsum[i];
+      dst[N + i] = bsum[BN + i];
     }
   }
 }
@@ -243,8 +272,13 @@ static bool use_v2() {
   return v;
 }
 
+extern "C" int gemm_bt_tiles_m(int M, int N) {
+  const int bm = N % 128 == 0 ? 128 : 256;
+  return (M + bm - 1) / bm;
+}
+
 extern "C" void launch_gemm_bt(const void* A, const void* B, void* C, int M,
-                               int N, int K, hipStream_t s) {
+                               int N, int K, float* bn_part, hipStream_t s) {
   if (N % 128 == 0) {
     constexpr int BM = 128, BN = 128;
     const int grid = ((M + BM - 1) / BM) * (N / BN);
@@ -252,10 +286,11 @@ extern "C" void launch_gemm_bt(const void* A, const void* B, void* C, int M,
     if (use_v2())
       hipLaunchKernelGGL((gemm_bt_kernel<BM, BN, 2, 2, false, true>), dim3(grid),
                          dim3(256), lds_bytes, s, (const bf16*)A, (const bf16*)B,
-                         C, M, N, K);
+                         C, M, N, K, bn_part);
     else
       hipLaunchKernelGGL((gemm_bt_kernel<BM, BN, 2, 2>), dim3(grid), dim3(256),
-                         lds_bytes, s, (const bf16*)A, (const bf16*)B, C, M, N, K);
+                         lds_bytes, s, (const bf16*)A, (const bf16*)B, C, M, N,
+                         K, bn_part);
   } else {  // N % 64 == 0
     constexpr int BM = 256, BN = 64;
     const int grid = ((M + BM - 1) / BM) * (N / BN);
@@ -263,10 +298,11 @@ extern "C" void launch_gemm_bt(const void* A, const void* B, void* C, int M,
     if (use_v2())
       hipLaunchKernelGGL((gemm_bt_kernel<BM, BN, 4, 1, false, true>), dim3(grid),
                          dim3(256), lds_bytes, s, (const bf16*)A, (const bf16*)B,
-                         C, M, N, K);
+                         C, M, N, K, bn_part);
     else
       hipLaunchKernelGGL((gemm_bt_kernel<BM, BN, 4, 1>), dim3(grid), dim3(256),
-                         lds_bytes, s, (const bf16*)A, (const bf16*)B, C, M, N, K);
+                         lds_bytes, s, (const bf16*)A, (const bf16*)B, C, M, N,
+                         K, bn_part);
   }
 }
 

@@ -34,11 +34,13 @@ extern "C" void launch_bn_bwd_reduce(const void*, const unsigned char*,
 extern "C" void launch_bn_bwd_finalize(const float*, int, float*, int, float*,
                                        float*, hipStream_t);
 extern "C" void launch_gemm_bt(const void*, const void*, void*, int, int, int,
-                               hipStream_t);
+                               float*, hipStream_t);
+extern "C" int gemm_bt_tiles_m(int, int);
+extern "C" int conv3x3_tiles_m(int, int);
 extern "C" void launch_pad_nhwc(const void*, void*, int, int, int, int, int, int,
                                 hipStream_t);
 extern "C" void launch_conv3x3(const void*, const void*, void*, int, int, int,
-                               int, int, int, int, int, float*, int,
+                               int, int, int, int, int, float*, int, float*,
                                hipStream_t);
 extern "C" int conv3x3_pick_splitk(int, int, int);
 extern "C" void launch_conv3x3_grouped(const void*, const void*, void*, int, int,
@@ -48,7 +50,7 @@ extern "C" void launch_conv3x3s2_dgrad(const void*, const void*, void*, int,
                                        int, int, int, int, int, int,
                                        hipStream_t);
 extern "C" void launch_conv3x3_small(const void*, const void*, void*, int, int,
-                                     int, int, int, int, int, int,
+                                     int, int, int, int, int, int, float*,
                                      hipStream_t);
 extern "C" void launch_pad_nhwc_cpad(const void*, void*, int, int, int, int,
                                      int, int, int, hipStream_t);
@@ -159,7 +161,8 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor gamma,
                                         torch::Tensor rvar, double momentum,
                                         double eps,
                                         c10::optional<torch::Tensor> res,
-                                        bool relu) {
+                                        bool relu,
+                                        c10::optional<torch::Tensor> pre_part) {
   // x: [M, C] contiguous view of an NHWC tensor (Python side reshapes)
   const int64_t C = gamma.numel();
   TORCH_CHECK(x.dim() == 2 && x.size(1) == C && x.is_contiguous(),
@@ -178,8 +181,22 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor gamma,
   auto msk = relu ? torch::empty({M, C / 8}, x.options().dtype(torch::kUInt8))
                   : torch::empty({0}, x.options().dtype(torch::kUInt8));
   auto s = cur_stream();
-  launch_bn_stats(x.data_ptr(), partial.data_ptr<float>(), grid, M, (int)C, s);
-  launch_bn_finalize(partial.data_ptr<float>(), grid, gamma.data_ptr<float>(),
+  int fin_grid = grid;
+  if (pre_part.has_value()) {
+    // stats already folded into the producing conv's epilogue
+    // ([tiles_m, 2C] fp32, conv3x3.hip BN_PART) — skip the stats kernel
+    // (a full activation re-read)
+    TORCH_CHECK(pre_part->is_contiguous() &&
+                    pre_part->scalar_type() == torch::kFloat32 &&
+                    pre_part->dim() == 2 && pre_part->size(1) == 2 * C,
+                "bn: pre_part [tiles, 2C] fp32");
+    partial = *pre_part;
+    fin_grid = (int)pre_part->size(0);
+  } else {
+    launch_bn_stats(x.data_ptr(), partial.data_ptr<float>(), grid, M, (int)C,
+                    s);
+  }
+  launch_bn_finalize(partial.data_ptr<float>(), fin_grid, gamma.data_ptr<float>(),
                      beta.data_ptr<float>(), mean.data_ptr<float>(),
                      invstd.data_ptr<float>(), scale.data_ptr<float>(),
                      shift.data_ptr<float>(),
@@ -272,9 +289,31 @@ torch::Tensor gemm_bt(torch::Tensor a, torch::Tensor b) {
   const int M = (int)a.size(0), K = (int)a.size(1), N = (int)b.size(0);
   TORCH_CHECK(K % 64 == 0 && N % 64 == 0, "gemm_bt: K,N % 64 == 0");
   auto c = torch::empty({M, N}, a.options());
-  launch_gemm_bt(ac.data_ptr(), bc.data_ptr(), c.data_ptr(), M, N, K,
+  launch_gemm_bt(ac.data_ptr(), bc.data_ptr(), c.data_ptr(), M, N, K, nullptr,
                  cur_stream());
   return c;
+}
+
+std::vector<torch::Tensor> gemm_bt_stats(torch::Tensor a, torch::Tensor b) {
+  // like gemm_bt, additionally returning BN stats partials [tiles_m, 2N]
+  // of the bf16-rounded output (the following BN skips its stats kernel)
+  TORCH_CHECK(a.is_cuda() && b.is_cuda(), "gemm_bt: GPU tensors required");
+  TORCH_CHECK(a.scalar_type() == torch::kBFloat16 &&
+                  b.scalar_type() == torch::kBFloat16,
+              "gemm_bt: bf16 only");
+  TORCH_CHECK(a.dim() == 2 && b.dim() == 2 && a.size(1) == b.size(1),
+              "gemm_bt: [M,K] x [N,K]");
+  auto ac = a.contiguous();
+  auto bc = b.contiguous();
+  const int M = (int)a.size(0), K = (int)a.size(1), N = (int)b.size(0);
+  TORCH_CHECK(K % 64 == 0 && N % 64 == 0, "gemm_bt: K,N % 64 == 0");
+  auto c = torch::empty({M, N}, a.options());
+  const int tiles_m = gemm_bt_tiles_m(M, N);
+  auto part = torch::empty({tiles_m, 2 * N},
+                           a.options().dtype(torch::kFloat32));
+  launch_gemm_bt(ac.data_ptr(), bc.data_ptr(), c.data_ptr(), M, N, K,
+                 part.data_ptr<float>(), cur_stream());
+  return {c, part};
 }
 
 torch::Tensor conv3x3_grouped_fwd(torch::Tensor x, torch::Tensor w3g,
@@ -343,8 +382,50 @@ torch::Tensor conv3x3_small_fwd(torch::Tensor x, torch::Tensor w3s,
   auto y = torch::empty({M, cout_real}, x.options());
   launch_conv3x3_small(xp.data_ptr(), w3s.data_ptr(), y.data_ptr(), (int)M,
                        (int)cout_real, (int)cpt, Hout * Wout, Wout, Hp, Wp,
-                       (int)stride, s);
+                       (int)stride, nullptr, s);
   return y;
+}
+
+std::vector<torch::Tensor> conv3x3_small_fwd_stats(torch::Tensor x,
+                                                   torch::Tensor w3s,
+                                                   int64_t cout_real,
+                                                   int64_t cpt,
+                                                   int64_t stride) {
+  // conv3x3_small_fwd + BN stats partials [tiles_m, 2*cout_real]
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 &&
+                  x.scalar_type() == torch::kBFloat16 &&
+                  x.is_contiguous(torch::MemoryFormat::ChannelsLast),
+              "conv3x3s: 4-D channels_last bf16");
+  TORCH_CHECK(stride == 1 || stride == 2, "conv3x3s: stride 1 or 2");
+  TORCH_CHECK(cpt == 16 || cpt == 32 || cpt == 64, "conv3x3s: cpt 16/32/64");
+  const int Nimg = (int)x.size(0), Cin = (int)x.size(1);
+  const int H = (int)x.size(2), W = (int)x.size(3);
+  TORCH_CHECK(Cin <= cpt, "conv3x3s: Cin <= cpt");
+  const int taps_pad = cpt == 16 ? 12 : (cpt == 32 ? 10 : 9);
+  TORCH_CHECK(w3s.is_contiguous() && w3s.size(0) == 64 &&
+                  w3s.size(1) == taps_pad * cpt,
+              "conv3x3s: w3s [64, taps_pad*cpt]");
+  TORCH_CHECK(cout_real >= 1 && cout_real <= 64, "conv3x3s: cout <= 64");
+  const int Hp = H + 2, Wp = W + 2;
+  const int Hout = (H - 1) / (int)stride + 1;
+  const int Wout = (W - 1) / (int)stride + 1;
+  const long long M = (long long)Nimg * Hout * Wout;
+  auto s = cur_stream();
+  auto xp = torch::empty({(long long)Nimg * Hp * Wp * cpt}, x.options());
+  if (Cin == cpt) {
+    launch_pad_nhwc(x.data_ptr(), xp.data_ptr(), Nimg, H, W, Hp, Wp, Cin, s);
+  } else {
+    launch_pad_nhwc_cpad(x.data_ptr(), xp.data_ptr(), Nimg, H, W, Hp, Wp,
+                         Cin, (int)cpt, s);
+  }
+  auto y = torch::empty({M, cout_real}, x.options());
+  const int tiles_m = (int)((M + 255) / 256);
+  auto bpart = torch::empty({tiles_m, 2 * cout_real},
+                            x.options().dtype(torch::kFloat32));
+  launch_conv3x3_small(xp.data_ptr(), w3s.data_ptr(), y.data_ptr(), (int)M,
+                       (int)cout_real, (int)cpt, Hout * Wout, Wout, Hp, Wp,
+                       (int)stride, bpart.data_ptr<float>(), s);
+  return {y, bpart};
 }
 
 torch::Tensor conv3x3s2_dgrad(torch::Tensor dy, torch::Tensor wcat,
@@ -632,13 +713,55 @@ torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w3, int64_t stride) {
     auto part = torch::zeros({M, Cout}, x.options().dtype(torch::kFloat32));
     launch_conv3x3(xp.data_ptr(), w3.data_ptr(), nullptr, (int)M, Cout, Cin,
                    Hout * Wout, Wout, Hp, Wp, (int)stride,
-                   part.data_ptr<float>(), splitk, s);
+                   part.data_ptr<float>(), splitk, nullptr, s);
     return part.to(torch::kBFloat16);
   }
   auto y = torch::empty({M, Cout}, x.options());
   launch_conv3x3(xp.data_ptr(), w3.data_ptr(), y.data_ptr(), (int)M, Cout, Cin,
-                 Hout * Wout, Wout, Hp, Wp, (int)stride, nullptr, 1, s);
+                 Hout * Wout, Wout, Hp, Wp, (int)stride, nullptr, 1, nullptr,
+                 s);
   return y;
+}
+
+std::vector<torch::Tensor> conv3x3_fwd_stats(torch::Tensor x, torch::Tensor w3,
+                                             int64_t stride) {
+  // conv3x3_fwd + BN stats partials of the output ([tiles_m, 2*Cout]).
+  // Split-K shapes can't fold stats (sumsq is nonlinear over partial
+  // sums) — the second return is then an undefined tensor (Python None).
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 &&
+                  x.scalar_type() == torch::kBFloat16,
+              "conv3x3: 4-D bf16 GPU tensor required");
+  TORCH_CHECK(x.is_contiguous(torch::MemoryFormat::ChannelsLast),
+              "conv3x3: channels_last required");
+  TORCH_CHECK(stride == 1 || stride == 2, "conv3x3: stride 1 or 2");
+  const int Nimg = (int)x.size(0), Cin = (int)x.size(1);
+  const int H = (int)x.size(2), W = (int)x.size(3);
+  const int Cout = (int)w3.size(0);
+  TORCH_CHECK(w3.size(1) == 9 * Cin && w3.is_contiguous(), "conv3x3: w3 shape");
+  TORCH_CHECK(Cin % 64 == 0 && Cout % 64 == 0, "conv3x3: C % 64");
+  const int Hp = H + 2, Wp = W + 2;
+  const int Hout = (H + 2 - 3) / (int)stride + 1;
+  const int Wout = (W + 2 - 3) / (int)stride + 1;
+  const long long M = (long long)Nimg * Hout * Wout;
+  auto s = cur_stream();
+  auto xp = torch::empty({(long long)Nimg * Hp * Wp * Cin}, x.options());
+  launch_pad_nhwc(x.data_ptr(), xp.data_ptr(), Nimg, H, W, Hp, Wp, Cin, s);
+  const int splitk = conv3x3_pick_splitk((int)M, Cout, Cin);
+  if (splitk > 1) {
+    auto part = torch::zeros({M, Cout}, x.options().dtype(torch::kFloat32));
+    launch_conv3x3(xp.data_ptr(), w3.data_ptr(), nullptr, (int)M, Cout, Cin,
+                   Hout * Wout, Wout, Hp, Wp, (int)stride,
+                   part.data_ptr<float>(), splitk, nullptr, s);
+    return {part.to(torch::kBFloat16), torch::Tensor()};
+  }
+  auto y = torch::empty({M, Cout}, x.options());
+  const int tiles_m = conv3x3_tiles_m((int)M, Cout);
+  auto bpart = torch::empty({tiles_m, 2 * Cout},
+                            x.options().dtype(torch::kFloat32));
+  launch_conv3x3(xp.data_ptr(), w3.data_ptr(), y.data_ptr(), (int)M, Cout, Cin,
+                 Hout * Wout, Wout, Hp, Wp, (int)stride, nullptr, 1,
+                 bpart.data_ptr<float>(), s);
+  return {y, bpart};
 }
 
 }  // namespace
@@ -652,7 +775,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("kd_ce_forward", &kd_ce_forward, "KD soft-label CE forward -> per-row loss");
   m.def("kd_ce_backward", &kd_ce_backward, "KD soft-label CE backward -> dlogits");
   m.def("bn_fwd_train", &bn_fwd_train,
-        "fused NHWC bf16 BN(+add)+ReLU train fwd -> (y, mean, invstd, mask)");
+        "fused NHWC bf16 BN(+add)+ReLU train fwd -> (y, mean, invstd, mask)",
+        py::arg("x"), py::arg("gamma"), py::arg("beta"), py::arg("rmean"),
+        py::arg("rvar"), py::arg("momentum"), py::arg("eps"), py::arg("res"),
+        py::arg("relu"), py::arg("pre_part") = py::none());
   m.def("bn_fwd_eval", &bn_fwd_eval, "fused NHWC bf16 BN(+add)+ReLU eval fwd");
   m.def("bn_bwd", &bn_bwd,
         "fused BN(+add)+ReLU bwd -> (dx, dgamma, dbeta, dres?)",
@@ -664,6 +790,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_bt", &gemm_bt, "bf16 MFMA GEMM: C[M,N] = A[M,K] @ B[N,K]^T");
   m.def("conv3x3_fwd", &conv3x3_fwd,
         "implicit-GEMM 3x3 same-pad conv (stride 1/2) -> y2d [M, Cout]");
+  m.def("conv3x3_fwd_stats", &conv3x3_fwd_stats,
+        "conv3x3_fwd + BN stats partials -> (y2d, part-or-None)");
+  m.def("gemm_bt_stats", &gemm_bt_stats,
+        "gemm_bt + BN stats partials -> (C, part)");
   m.def("transpose_pad", &transpose_pad,
         "bf16 [M,C] -> [C, ceil64(M)] transpose with zero pad");
   m.def("gemm_bt_splitk", &gemm_bt_splitk,
@@ -686,6 +816,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "stride-2 3x3 same-pad dgrad (parity implicit GEMM) -> dx2d [N*H*W, Cin]");
   m.def("conv3x3_small_fwd", &conv3x3_small_fwd,
         "small-channel 3x3 conv (deep stem) -> y2d [M, cout_real]");
+  m.def("conv3x3_small_fwd_stats", &conv3x3_small_fwd_stats,
+        "stem conv + BN stats partials -> (y2d, part)");
   m.def("repack_dgrad_w3", &repack_dgrad_w3,
         "dgrad weight repack [Co,9Ci]->[Ci,9Co] (mode 0 rot / 1 s2 order)");
   m.def("maxpool3x3s2_fwd", &maxpool3x3s2_fwd,

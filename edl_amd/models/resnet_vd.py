@@ -26,7 +26,9 @@ class ConvBN(nn.Module):
         self.bn = BNReLU2d(cout, act=act)
 
     def forward(self, x):
-        return self.bn(self.conv(x))
+        # conv folds BN stats partials into its epilogue when training
+        y = self.conv(x, bn_stats=self.bn.training)
+        return self.bn(y, partials=self.conv.pop_bn_part())
 
 
 class VdShortcut(nn.Module):
@@ -67,8 +69,9 @@ class BottleneckVd(nn.Module):
 
     def forward(self, x):
         s = self.shortcut(x)
-        y = self.conv2(self.conv1(self.conv0(x)))
-        return self.bn_add(y, s)
+        y = self.conv2(self.conv1(self.conv0(x)),
+                       bn_stats=self.bn_add.training)
+        return self.bn_add(y, s, partials=self.conv2.pop_bn_part())
 
 
 class BasicBlockVd(nn.Module):
@@ -84,7 +87,8 @@ class BasicBlockVd(nn.Module):
 
     def forward(self, x):
         s = self.shortcut(x)
-        return self.bn_add(self.conv1(self.conv0(x)), s)
+        y = self.conv1(self.conv0(x), bn_stats=self.bn_add.training)
+        return self.bn_add(y, s, partials=self.conv1.pop_bn_part())
 
 
 _DEPTHS = {

@@ -26,9 +26,11 @@ def _direct_grad(p):
 
 class _FusedBN(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x2d, gamma, beta, rmean, rvar, momentum, eps, res2d, relu):
+    def forward(ctx, x2d, gamma, beta, rmean, rvar, momentum, eps, res2d, relu,
+                pre_part=None):
         y, mean, invstd, mask = ext().bn_fwd_train(
-            x2d, gamma, beta, rmean, rvar, momentum, eps, res2d, relu
+            x2d, gamma, beta, rmean, rvar, momentum, eps, res2d, relu,
+            pre_part
         )
         # backward reads the 1-bit ReLU mask instead of y (16x fewer bytes)
         ctx.save_for_backward(x2d, mask, mean, invstd, gamma)
@@ -49,7 +51,7 @@ class _FusedBN(torch.autograd.Function):
                 None if ctx.dg_t is not None else dgamma,
                 None if ctx.db_t is not None else dbeta,
                 None, None, None, None,
-                dres if ctx.has_res else None, None)
+                dres if ctx.has_res else None, None, None)
 
 
 def _to_2d(t):
@@ -121,13 +123,13 @@ class BNReLU2d(nn.Module):
             y = F.relu(y)
         return y.to(x.dtype)
 
-    def _fused(self, x, res=None):
+    def _fused(self, x, res=None, partials=None):
         x2d = _to_2d(x)
         res2d = _to_2d(res) if res is not None else None
         if self.training:
             y2d = _FusedBN.apply(
                 x2d, self.weight, self.bias, self.running_mean, self.running_var,
-                self.momentum, self.eps, res2d, self.act,
+                self.momentum, self.eps, res2d, self.act, partials,
             )
         else:
             invstd = torch.rsqrt(self.running_var + self.eps)
@@ -136,12 +138,15 @@ class BNReLU2d(nn.Module):
             y2d = ext().bn_fwd_eval(x2d, scale, shift, res2d, self.act)
         return _from_2d(y2d, x.shape)
 
-    def forward(self, x, res=None):
+    def forward(self, x, res=None, partials=None):
+        # partials: BN stats pre-folded into the producing conv's
+        # epilogue ([tiles, 2C] fp32) — skips the stats kernel
         if self.training:
             self._batches_seen += 1
         if self._use_fused(x, res):
             if self.training or not torch.is_grad_enabled():
-                return self._fused(x, res)
+                return self._fused(x, res,
+                                   partials if self.training else None)
         return self._fallback(x, res)
 
     def extra_repr(self):
@@ -154,5 +159,5 @@ class BNAddReLU2d(BNReLU2d):
     def __init__(self, num_features, eps=1e-5, momentum=0.1):
         super().__init__(num_features, eps=eps, momentum=momentum, act=True)
 
-    def forward(self, x, res):  # res is mandatory here
-        return super().forward(x, res)
+    def forward(self, x, res, partials=None):  # res is mandatory here
+        return super().forward(x, res, partials)

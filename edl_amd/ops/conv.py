@@ -41,7 +41,8 @@ class _Conv1x1Hip(torch.autograd.Function):
     optionally accumulating straight into the bucket-view grad."""
 
     @staticmethod
-    def forward(ctx, x2d, w_param, w_bf16, wt_cached, grad_tgt):
+    def forward(ctx, x2d, w_param, w_bf16, wt_cached, grad_tgt,
+                part_out=None):
         # w_param: the fp32 (or bf16) parameter view — the DIFFERENTIABLE
         # input; w_bf16/wt_cached: per-step cached compute copies. Grads
         # come back in fp32 straight from the split-K kernel: no cast
@@ -52,6 +53,10 @@ class _Conv1x1Hip(torch.autograd.Function):
         ctx.wt = wt_cached  # [Cin, Cout] bf16, derived per weight epoch
         ctx.w_dtype = w_param.dtype
         ctx.grad_tgt = grad_tgt
+        if part_out is not None:
+            y2d, part = ext().gemm_bt_stats(x2d, w_bf16)
+            part_out.append(part)
+            return y2d
         return ext().gemm_bt(x2d, w_bf16)
 
     @staticmethod
@@ -73,13 +78,13 @@ class _Conv1x1Hip(torch.autograd.Function):
                 e.gemm_tn_splitk(
                     dy2d, x2d, 0,
                     ctx.grad_tgt.view(ctx.grad_tgt.shape[0], -1))
-                return dx, None, None, None, None
+                return dx, None, None, None, None, None
             dw = e.gemm_tn_splitk(dy2d, x2d, 0)
         else:
             dw = e.gemm_bt_splitk(e.transpose_pad(dy2d), e.transpose_pad(x2d), 0)
         if dw.dtype != ctx.w_dtype:
             dw = dw.to(ctx.w_dtype)
-        return dx, dw, None, None, None
+        return dx, dw, None, None, None, None
 
 
 _CONV3X3 = os.environ.get("EDL_CONV3X3", "hip")
@@ -91,6 +96,10 @@ _CONV3X3 = os.environ.get("EDL_CONV3X3", "hip")
 # even the 4x-zero cpg=16 block-diagonal beats MIOpen once the rest of
 # the net runs in-repo kernels.
 _GROUPED_MINC = int(os.environ.get("EDL_CONV3X3_GROUPED_MINC", "16"))
+
+# Fold BN stats partials into the conv fwd epilogues (the following BN
+# skips its stats kernel — a full activation re-read, ~0.53 ms/step).
+_BN_STATS_FUSED = os.environ.get("EDL_BN_STATS_FUSED", "1") == "1"
 
 # Weight-derived tensors (bf16 casts, transposed/repacked layouts) are
 # immutable within one optimizer step; FusedSGD.step() bumps this epoch and
@@ -121,13 +130,18 @@ class _Conv3x3Hip(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, w_param, w_bf16, w3_cached, w3rot_cached, stride,
-                grad_tgt):
+                grad_tgt, part_out=None):
         ctx.save_for_backward(x, w_bf16)
         ctx.stride = stride
         ctx.w3rot = w3rot_cached  # s1: rotated fwd repack; s2: s2dgrad wcat
         ctx.w_dtype = w_param.dtype
         ctx.grad_tgt = grad_tgt
-        y2d = ext().conv3x3_fwd(x, w3_cached, stride)
+        if part_out is not None:
+            y2d, part = ext().conv3x3_fwd_stats(x, w3_cached, stride)
+            if part is not None:  # split-K shapes return None
+                part_out.append(part)
+        else:
+            y2d = ext().conv3x3_fwd(x, w3_cached, stride)
         n, _, h, w = x.shape
         ho = (h - 1) // stride + 1
         wo = (w - 1) // stride + 1
@@ -168,7 +182,7 @@ class _Conv3x3Hip(torch.autograd.Function):
                 # [Cout, Cin, 3, 3] (kernel epilogue remaps)
                 e.gemm_tn3x3_splitk(dy2d.to(torch.bfloat16), x, stride, 0,
                                     ctx.grad_tgt)
-                return dx, None, None, None, None, None, None
+                return dx, None, None, None, None, None, None, None
             dw3 = e.gemm_tn3x3_splitk(dy2d.to(torch.bfloat16), x, stride, 0)
         else:
             dw3 = e.gemm_bt_splitk(
@@ -177,7 +191,7 @@ class _Conv3x3Hip(torch.autograd.Function):
         dw = dw3.view(co, 3, 3, ci).permute(0, 3, 1, 2)  # fp32
         if dw.dtype != ctx.w_dtype:
             dw = dw.to(ctx.w_dtype)
-        return dx, dw, None, None, None, None, None
+        return dx, dw, None, None, None, None, None, None
 
 
 def _small_cpt(cin):
@@ -211,14 +225,19 @@ class _Conv3x3SmallHip(torch.autograd.Function):
     GEMM per the north star's library-GEMM allowance)."""
 
     @staticmethod
-    def forward(ctx, x, w_param, w3s, w3srot, stride, cout):
+    def forward(ctx, x, w_param, w3s, w3srot, stride, cout, part_out=None):
         ctx.save_for_backward(x)
         ctx.stride = stride
         ctx.w3srot = w3srot  # None when dx is not needed (stem conv0)
         ctx.w_dtype = w_param.dtype
         ctx.w_shape = tuple(w_param.shape)
         cpt = _small_cpt(x.shape[1])
-        y2d = ext().conv3x3_small_fwd(x, w3s, cout, cpt, stride)
+        if part_out is not None:
+            y2d, part = ext().conv3x3_small_fwd_stats(x, w3s, cout, cpt,
+                                                      stride)
+            part_out.append(part)
+        else:
+            y2d = ext().conv3x3_small_fwd(x, w3s, cout, cpt, stride)
         n, _, h, w = x.shape
         ho = (h - 1) // stride + 1
         wo = (w - 1) // stride + 1
@@ -254,7 +273,7 @@ class _Conv3x3SmallHip(torch.autograd.Function):
                 padding=(1, 1))
         if dw.dtype != ctx.w_dtype:
             dw = dw.to(ctx.w_dtype)
-        return dx, dw, None, None, None, None
+        return dx, dw, None, None, None, None, None
 
 
 _S2D_TAP_ORDER = [(1, 1), (1, 0), (1, 2), (0, 1), (2, 1),
@@ -328,7 +347,16 @@ class Conv2dFast(nn.Conv2d):
             d[key] = fn()
         return d[key]
 
-    def forward(self, x):
+    def pop_bn_part(self):
+        """BN stats partials of the LAST forward (or None) — consumed by
+        the following BatchNorm to skip its stats kernel."""
+        p = getattr(self, "_bn_part", None)
+        self._bn_part = None
+        return p
+
+    def forward(self, x, bn_stats=False):
+        self._bn_part = None
+        bn_stats = bn_stats and _BN_STATS_FUSED
         if (
             _CONV3X3 == "hip"
             and x.is_cuda
@@ -384,8 +412,12 @@ class Conv2dFast(nn.Conv2d):
                     and ci <= _WGRAD3_MAXC
                     and torch.is_grad_enabled() and self.training):
                 grad_tgt = w.grad.permute(0, 2, 3, 1).reshape(co, 9 * ci)
-            return _Conv3x3Hip.apply(x, self.weight, w_bf16, w3, w3rot,
-                                     self.stride[0], grad_tgt)
+            holder = [] if bn_stats else None
+            y = _Conv3x3Hip.apply(x, self.weight, w_bf16, w3, w3rot,
+                                  self.stride[0], grad_tgt, holder)
+            if holder:
+                self._bn_part = holder[0]
+            return y
         if (
             # deep-stem 3x3s (3->32->32->64): small-channel kernel
             _CONV3X3 == "hip"
@@ -417,8 +449,13 @@ class Conv2dFast(nn.Conv2d):
                 w3srot = self._cached("w3srot", lambda: _repack_w3_small(
                     wsrc.permute(1, 0, 2, 3).flip(2, 3),
                     _small_cpt(self.out_channels)))
-            return _Conv3x3SmallHip.apply(x, self.weight, w3s, w3srot,
-                                          self.stride[0], self.out_channels)
+            holder = [] if bn_stats else None
+            y = _Conv3x3SmallHip.apply(x, self.weight, w3s, w3srot,
+                                       self.stride[0], self.out_channels,
+                                       holder)
+            if holder:
+                self._bn_part = holder[0]
+            return y
         if (
             # Grouped teacher 3x3s. cpg >= _GROUPED_MINC runs the in-repo
             # grouped implicit-GEMM kernel (cpg >= 64: exact, zero wasted
@@ -490,8 +527,11 @@ class Conv2dFast(nn.Conv2d):
                                           .to(torch.bfloat16).contiguous())
                     wt_t = self._cached("wt_t", lambda: self.weight.detach().view(
                         self.out_channels, c).to(torch.bfloat16).t().contiguous())
+                holder = [] if bn_stats else None
                 y2d = _Conv1x1Hip.apply(x2d, wt, w_bf16, wt_t,
-                                        self._grad_tgt())
+                                        self._grad_tgt(), holder)
+                if holder:
+                    self._bn_part = holder[0]
             else:
                 y2d = x2d @ wt.t()
             return (

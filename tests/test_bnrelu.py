@@ -136,3 +136,74 @@ class TestFusedGPU:
             y_ref = model(x)  # fp32, fallback path inside BNReLU2d
         assert torch.allclose(y_fused.float(), y_ref, atol=0.5, rtol=0.1), \
             (y_fused.float() - y_ref).abs().max().item()
+
+
+@pytest.mark.gpu
+class TestConvFusedStats:
+    def _gpu(self):
+        if not torch.cuda.is_available():
+            pytest.skip("no GPU")
+
+    @pytest.mark.parametrize("shape", [
+        (2, 64, 64, 14, 1),    # 1x1 (gemm_bt epilogue)
+        (2, 64, 128, 14, 3),   # 3x3 (conv3x3 epilogue)
+        (2, 3, 32, 16, 3),     # stem (small kernel epilogue)
+    ])
+    def test_conv_epilogue_stats_match_bn_stats(self, shape, monkeypatch):
+        """BN fed conv-folded stats partials must match BN computing its
+        own stats on the same bf16 output (train mode, running stats and
+        y compared)."""
+        self._gpu()
+        import copy
+
+        import edl_amd.ops.conv as conv_mod
+        from edl_amd.models.resnet_vd import ConvBN
+
+        n, ci, co, hw, k = shape
+        torch.manual_seed(31)
+        m1 = ConvBN(ci, co, k).cuda().train()
+        m1.conv.to(torch.bfloat16)
+        m2 = copy.deepcopy(m1)
+        x = torch.randn(n, ci, hw, hw, device="cuda").to(torch.bfloat16)
+        x = x.contiguous(memory_format=torch.channels_last)
+
+        monkeypatch.setattr(conv_mod, "_BN_STATS_FUSED", True)
+        y1 = m1(x)
+        monkeypatch.setattr(conv_mod, "_BN_STATS_FUSED", False)
+        y2 = m2(x)
+        assert torch.allclose(y1.float(), y2.float(), atol=1e-2, rtol=1e-2), \
+            (y1.float() - y2.float()).abs().max().item()
+        assert torch.allclose(m1.bn.running_mean, m2.bn.running_mean,
+                              atol=1e-4, rtol=1e-4)
+        assert torch.allclose(m1.bn.running_var, m2.bn.running_var,
+                              atol=1e-4, rtol=1e-4)
+
+    def test_bottleneck_trains_with_fused_stats(self, monkeypatch):
+        """Full bottleneck fwd+bwd with conv-folded stats: finite grads,
+        same trajectory as the unfused-stats path."""
+        self._gpu()
+        import copy
+
+        import edl_amd.ops.conv as conv_mod
+        from edl_amd.models.resnet_vd import BottleneckVd
+
+        torch.manual_seed(32)
+        b1 = BottleneckVd(64, 64, stride=1, if_first=True).cuda().train()
+        for mod in b1.modules():
+            if isinstance(mod, conv_mod.Conv2dFast):
+                mod.to(torch.bfloat16)
+        b2 = copy.deepcopy(b1)
+        x = torch.randn(2, 64, 14, 14, device="cuda").to(torch.bfloat16)
+        x = x.contiguous(memory_format=torch.channels_last)
+
+        monkeypatch.setattr(conv_mod, "_BN_STATS_FUSED", True)
+        y1 = b1(x)
+        y1.float().square().mean().backward()
+        monkeypatch.setattr(conv_mod, "_BN_STATS_FUSED", False)
+        y2 = b2(x)
+        y2.float().square().mean().backward()
+        assert torch.allclose(y1.float(), y2.float(), atol=1e-2, rtol=1e-2)
+        for p1, p2 in zip(b1.parameters(), b2.parameters()):
+            if p1.grad is not None:
+                assert torch.allclose(p1.grad.float(), p2.grad.float(),
+                                      atol=5e-2, rtol=5e-2), p1.shape
