@@ -242,11 +242,23 @@ __global__ __launch_bounds__(256, 2) void conv3x3_kernel(
       atomicAdd(&bsum[BN + wn + nf * 16 + cn], lq[nf]);
     }
     __syncthreads();
-    const int mtile = (bid / (N / BN));
+    // fold into <=192 partial rows (finalize latency/traffic cap): rows
+    // beyond the cap atomically add into row mtile %% cap (buffer
+    // pre-zeroed by the caller in that case)
+    const int tiles_m = gridDim.x / (N / BN);
+    const int cap = tiles_m < 192 ? tiles_m : 192;
+    const int mtile = (bid / (N / BN)) % cap;
     float* dst = bn_part + (long long)mtile * 2 * N + n0;
-    for (int i = threadIdx.x; i < BN; i += blockDim.x) {
-      dst[i] = bsum[i];
-      dst[N + i] = bsum[BN + i];
+    if (tiles_m > cap) {
+      for (int i = threadIdx.x; i < BN; i += blockDim.x) {
+        atomicAdd(&dst[i], bsum[i]);
+        atomicAdd(&dst[N + i], bsum[BN + i]);
+      }
+    } else {
+      for (int i = threadIdx.x; i < BN; i += blockDim.x) {
+        dst[i] = bsum[i];
+        dst[N + i] = bsum[BN + i];
+      }
     }
   }
 }
@@ -409,10 +421,18 @@ __global__ __launch_bounds__(256, 2) void conv3x3_small_kernel(
         atomicAdd(&bsum[BN + nf * 16 + cn], lq[nf]);
       }
     __syncthreads();
-    float* dst = bn_part + (long long)bid * 2 * Cout_real;
-    for (int i = threadIdx.x; i < Cout_real; i += blockDim.x) {
-      dst[i] = bsum[i];
-      dst[Cout_real + i] = bsum[BN + i];
+    const int cap = (int)gridDim.x < 192 ? (int)gridDim.x : 192;
+    float* dst = bn_part + (long long)(bid % cap) * 2 * Cout_real;
+    if ((int)gridDim.x > cap) {
+      for (int i = threadIdx.x; i < Cout_real; i += blockDim.x) {
+        atomicAdd(&dst[i], bsum[i]);
+        atomicAdd(&dst[Cout_real + i], bsum[BN + i]);
+      }
+    } else {
+      for (int i = threadIdx.x; i < Cout_real; i += blockDim.x) {
+        dst[i] = bsum[i];
+        dst[Cout_real + i] = bsum[BN + i];
+      }
     }
   }
 }

@@ -252,10 +252,19 @@ __global__ __launch_bounds__(256, 2) void gemm_bt_kernel(
       atomicAdd(&bsum[BN + wn + nf * 16 + cn], lq[nf]);
     }
     __syncthreads();
-    float* dst = bn_part + (long long)tile_m * 2 * N + n0;
-    for (int i = threadIdx.x; i < BN; i += blockDim.x) {
-      dst[i] = bsum[i];
-      dst[N + i] = bsum[BN + i];
+    const int tiles_m_g = gridDim.x / tiles_n;
+    const int cap = tiles_m_g < 192 ? tiles_m_g : 192;
+    float* dst = bn_part + (long long)(tile_m % cap) * 2 * N + n0;
+    if (tiles_m_g > cap) {  // pre-zeroed by the caller
+      for (int i = threadIdx.x; i < BN; i += blockDim.x) {
+        atomicAdd(&dst[i], bsum[i]);
+        atomicAdd(&dst[N + i], bsum[BN + i]);
+      }
+    } else {
+      for (int i = threadIdx.x; i < BN; i += blockDim.x) {
+        dst[i] = bsum[i];
+        dst[N + i] = bsum[BN + i];
+      }
     }
   }
 }

@@ -309,8 +309,12 @@ std::vector<torch::Tensor> gemm_bt_stats(torch::Tensor a, torch::Tensor b) {
   TORCH_CHECK(K % 64 == 0 && N % 64 == 0, "gemm_bt: K,N % 64 == 0");
   auto c = torch::empty({M, N}, a.options());
   const int tiles_m = gemm_bt_tiles_m(M, N);
-  auto part = torch::empty({tiles_m, 2 * N},
-                           a.options().dtype(torch::kFloat32));
+  const int rows = std::min(tiles_m, 192);
+  auto part = tiles_m > rows
+                  ? torch::zeros({rows, 2 * N},
+                                 a.options().dtype(torch::kFloat32))
+                  : torch::empty({rows, 2 * N},
+                                 a.options().dtype(torch::kFloat32));
   launch_gemm_bt(ac.data_ptr(), bc.data_ptr(), c.data_ptr(), M, N, K,
                  part.data_ptr<float>(), cur_stream());
   return {c, part};
@@ -420,8 +424,12 @@ std::vector<torch::Tensor> conv3x3_small_fwd_stats(torch::Tensor x,
   }
   auto y = torch::empty({M, cout_real}, x.options());
   const int tiles_m = (int)((M + 255) / 256);
-  auto bpart = torch::empty({tiles_m, 2 * cout_real},
-                            x.options().dtype(torch::kFloat32));
+  const int rows = std::min(tiles_m, 192);
+  auto bpart = tiles_m > rows
+                   ? torch::zeros({rows, 2 * cout_real},
+                                  x.options().dtype(torch::kFloat32))
+                   : torch::empty({rows, 2 * cout_real},
+                                  x.options().dtype(torch::kFloat32));
   launch_conv3x3_small(xp.data_ptr(), w3s.data_ptr(), y.data_ptr(), (int)M,
                        (int)cout_real, (int)cpt, Hout * Wout, Wout, Hp, Wp,
                        (int)stride, bpart.data_ptr<float>(), s);
@@ -756,8 +764,12 @@ std::vector<torch::Tensor> conv3x3_fwd_stats(torch::Tensor x, torch::Tensor w3,
   }
   auto y = torch::empty({M, Cout}, x.options());
   const int tiles_m = conv3x3_tiles_m((int)M, Cout);
-  auto bpart = torch::empty({tiles_m, 2 * Cout},
-                            x.options().dtype(torch::kFloat32));
+  const int rows = std::min(tiles_m, 192);
+  auto bpart = tiles_m > rows
+                   ? torch::zeros({rows, 2 * Cout},
+                                  x.options().dtype(torch::kFloat32))
+                   : torch::empty({rows, 2 * Cout},
+                                  x.options().dtype(torch::kFloat32));
   launch_conv3x3(xp.data_ptr(), w3.data_ptr(), y.data_ptr(), (int)M, Cout, Cin,
                  Hout * Wout, Wout, Hp, Wp, (int)stride, nullptr, 1,
                  bpart.data_ptr<float>(), s);
