@@ -73,7 +73,12 @@ __device__ __forceinline__ int tn_swz(int r) {
 // zero packing VALU. Staging stays glds lane-linear: chunk ch = region
 // (nb = ch>>1, r = ch&1), lane l holds k = (l>>3)*8 + 4r + ((l>>1)&3),
 // n = nb*16 + (l&1)*8.
-template <int BN1, int BN2, int KW, bool G3 = false, bool TR = false>
+// DEEP (KW==1 only): 3 LDS buffers, raw s_barrier + COUNTED vmcnt so two
+// staged k-tiles stay in flight across barriers (the __syncthreads form
+// drains every glds per iteration — guide 'pipelining across barriers';
+// costs 96 KB LDS -> 1 block/CU).
+template <int BN1, int BN2, int KW, bool G3 = false, bool TR = false,
+          bool DEEP = false>
 __global__ __launch_bounds__(256, 2) void gemm_tn_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B,
     float* __restrict__ C, const int N1, const int N2, const int K,
@@ -293,12 +298,34 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_kernel(
     return d;
   };
 
-  do_stage(0, 0);
-  __syncthreads();  // implicit vmcnt(0)+lgkmcnt(0) drains glds/ds stores
+  static_assert(!DEEP || KW == 1, "DEEP needs KW == 1");
+  static_assert(!DEEP || (A_BYTES + B_BYTES) / 1024 / GW == 8,
+                "DEEP wait literals assume 8 glds per wave per stage");
+
+  if constexpr (DEEP) {
+    do_stage(0, 0);
+    if (iters > 1) do_stage(1, 1);
+  } else {
+    do_stage(0, 0);
+    __syncthreads();  // implicit vmcnt(0)+lgkmcnt(0) drains glds/ds stores
+  }
 
   for (int i = 0; i < iters; ++i) {
-    const int cur = i & 1;
-    if (i + 1 < iters) do_stage(cur ^ 1, i + 1);
+    int cur;
+    if constexpr (DEEP) {
+      // own stage(i) must have landed; stage(i+1) may stay in flight
+      if (i + 1 < iters)
+        asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+      else
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");  // tail ds_writes
+      __builtin_amdgcn_s_barrier();
+      if (i + 2 < iters) do_stage((i + 2) % 3, i + 2);
+      cur = i % 3;
+    } else {
+      cur = i & 1;
+      if (i + 1 < iters) do_stage(cur ^ 1, i + 1);
+    }
     if (c0 + grp + i * KW < c1) {
 #pragma unroll
       for (int kk = 0; kk < 2; ++kk) {
@@ -343,8 +370,9 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_kernel(
                 a[mf], b[nf], acc[mf][nf], 0, 0, 0);
       }
     }
-    __syncthreads();
+    if constexpr (!DEEP) __syncthreads();
   }
+  if constexpr (DEEP) __syncthreads();  // drain before epilogue reuse
 
   // ---- KW > 1: fold the k-group partials in LDS first — without this
   // every group wave atomicAdds the SAME 64x64 tile (KW x the atomic
@@ -417,6 +445,14 @@ static bool tn_use_tr() {
   return v;
 }
 
+static bool tn_use_deep() {
+  static const bool v = []() {
+    const char* e = getenv("EDL_TN_DEEP");
+    return e && atoi(e) != 0;  // opt-in experiment (3-buffer pipeline)
+  }();
+  return v;
+}
+
 extern "C" void launch_gemm_tn_splitk(const void* A, const void* B, float* C,
                                       int N1, int N2, int K, int splitk,
                                       hipStream_t s) {
@@ -426,7 +462,21 @@ extern "C" void launch_gemm_tn_splitk(const void* A, const void* B, float* C,
   const int KW = 4 / ((BN1 / 64) * (BN2 / 64));
   const int lds = KW * 2 * (64 * BN1 * 2 + 64 * BN2 * 2);
 #define TN_LAUNCH(B1, B2, W)                                                  do {                                                                          if (tn_use_tr())                                                              hipLaunchKernelGGL((gemm_tn_kernel<B1, B2, W, false, true>), grid,                             dim3(256), lds, s, (const bf16*)A, (const bf16*)B,                          C, N1, N2, K);                                         else                                                                          hipLaunchKernelGGL((gemm_tn_kernel<B1, B2, W, false, false>), grid,                            dim3(256), lds, s, (const bf16*)A, (const bf16*)B,                          C, N1, N2, K);                                       } while (0)
-  if (b1 && b2) TN_LAUNCH(128, 128, 1);
+  if (b1 && b2) {
+    if (tn_use_deep()) {
+      const int lds3 = 3 * (64 * 128 * 2 + 64 * 128 * 2);
+      if (tn_use_tr())
+        hipLaunchKernelGGL((gemm_tn_kernel<128, 128, 1, false, true, true>),
+                           grid, dim3(256), lds3, s, (const bf16*)A,
+                           (const bf16*)B, C, N1, N2, K);
+      else
+        hipLaunchKernelGGL((gemm_tn_kernel<128, 128, 1, false, false, true>),
+                           grid, dim3(256), lds3, s, (const bf16*)A,
+                           (const bf16*)B, C, N1, N2, K);
+    } else {
+      TN_LAUNCH(128, 128, 1);
+    }
+  }
   else if (b1) TN_LAUNCH(128, 64, 2);
   else if (b2) TN_LAUNCH(64, 128, 2);
   else  // KW=4 spills 28 B scratch under TR (16 live u64 reads + acc)
@@ -450,7 +500,23 @@ extern "C" void launch_gemm_tn3x3_splitk(const void* dy, const void* xpad,
   const int lds = (4 / ((BN1 / 64) * (BN2 / 64))) * 2 *
                   (64 * BN1 * 2 + 64 * BN2 * 2);
 #define TN3_LAUNCH(B1, B2, W)                                                   do {                                                                            if (tn_use_tr())                                                                hipLaunchKernelGGL((gemm_tn_kernel<B1, B2, W, true, true>), grid,                                dim3(256), lds, s, (const bf16*)dy,                                           (const bf16*)xpad, C, Cout, N2, M, Ho, Wo, Hp, Wp,                            Cin, stride, perm);                                      else                                                                            hipLaunchKernelGGL((gemm_tn_kernel<B1, B2, W, true, false>), grid,                               dim3(256), lds, s, (const bf16*)dy,                                           (const bf16*)xpad, C, Cout, N2, M, Ho, Wo, Hp, Wp,                            Cin, stride, perm);                                    } while (0)
-  if (b1 && b2) TN3_LAUNCH(128, 128, 1);
+  if (b1 && b2) {
+    if (tn_use_deep()) {
+      const int lds3 = 3 * (64 * 128 * 2 + 64 * 128 * 2);
+      if (tn_use_tr())
+        hipLaunchKernelGGL((gemm_tn_kernel<128, 128, 1, true, true, true>),
+                           grid, dim3(256), lds3, s, (const bf16*)dy,
+                           (const bf16*)xpad, C, Cout, N2, M, Ho, Wo, Hp, Wp,
+                           Cin, stride, perm);
+      else
+        hipLaunchKernelGGL((gemm_tn_kernel<128, 128, 1, true, false, true>),
+                           grid, dim3(256), lds3, s, (const bf16*)dy,
+                           (const bf16*)xpad, C, Cout, N2, M, Ho, Wo, Hp, Wp,
+                           Cin, stride, perm);
+    } else {
+      TN3_LAUNCH(128, 128, 1);
+    }
+  }
   else if (b1) TN3_LAUNCH(128, 64, 2);
   else if (b2) TN3_LAUNCH(64, 128, 2);
   else
