@@ -31,8 +31,11 @@ def _run_probe(nproc, port, extra_env=None):
         env=env, cwd=REPO, capture_output=True, text=True, timeout=600,
     )
     sys.stderr.write(r.stdout[-3000:] + r.stderr[-2000:])
-    verdicts = [json.loads(l) for l in r.stdout.splitlines()
-                if l.startswith('{"probe"')]
+    import re
+
+    # torchrun can interleave both ranks' lines without a newline
+    verdicts = [json.loads(m) for m in
+                re.findall(r'\{"probe".*?"ok": (?:true|false)\}', r.stdout)]
     return r.returncode, verdicts
 
 
