@@ -238,8 +238,20 @@ __global__ __launch_bounds__(256, 2) void conv3x3_kernel(
     __syncthreads();
 #pragma unroll
     for (int nf = 0; nf < 4; ++nf) {
-      atomicAdd(&bsum[wn + nf * 16 + cn], ls[nf]);
-      atomicAdd(&bsum[BN + wn + nf * 16 + cn], lq[nf]);
+      // fold the 4 lanes sharing column cn (lane, +16, +32, +48): ONE
+      // LDS atomic per address per wave instead of a 4-way serialized
+      // same-address conflict
+      ls[nf] += __shfl_down(ls[nf], 32);
+      ls[nf] += __shfl_down(ls[nf], 16);
+      lq[nf] += __shfl_down(lq[nf], 32);
+      lq[nf] += __shfl_down(lq[nf], 16);
+    }
+    if (lane < 16) {
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+        atomicAdd(&bsum[wn + nf * 16 + cn], ls[nf]);
+        atomicAdd(&bsum[BN + wn + nf * 16 + cn], lq[nf]);
+      }
     }
     __syncthreads();
     // fold into <=192 partial rows (finalize latency/traffic cap): rows
@@ -415,11 +427,23 @@ __global__ __launch_bounds__(256, 2) void conv3x3_small_kernel(
     for (int i = threadIdx.x; i < 2 * BN; i += blockDim.x) bsum[i] = 0.0f;
     __syncthreads();
 #pragma unroll
-    for (int nf = 0; nf < 4; ++nf)
-      if (nf * 16 + cn < Cout_real) {
-        atomicAdd(&bsum[nf * 16 + cn], ls[nf]);
-        atomicAdd(&bsum[BN + nf * 16 + cn], lq[nf]);
-      }
+    for (int nf = 0; nf < 4; ++nf) {
+      // fold the 4 lanes sharing column cn (lane, +16, +32, +48): ONE
+      // LDS atomic per address per wave instead of a 4-way serialized
+      // same-address conflict
+      ls[nf] += __shfl_down(ls[nf], 32);
+      ls[nf] += __shfl_down(ls[nf], 16);
+      lq[nf] += __shfl_down(lq[nf], 32);
+      lq[nf] += __shfl_down(lq[nf], 16);
+    }
+    if (lane < 16) {
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf)
+        if (nf * 16 + cn < Cout_real) {
+          atomicAdd(&bsum[nf * 16 + cn], ls[nf]);
+          atomicAdd(&bsum[BN + nf * 16 + cn], lq[nf]);
+        }
+    }
     __syncthreads();
     const int cap = (int)gridDim.x < 192 ? (int)gridDim.x : 192;
     float* dst = bn_part + (long long)(bid % cap) * 2 * Cout_real;

@@ -248,8 +248,20 @@ __global__ __launch_bounds__(256, 2) void gemm_bt_kernel(
     __syncthreads();
 #pragma unroll
     for (int nf = 0; nf < 4; ++nf) {
-      atomicAdd(&bsum[wn + nf * 16 + cn], ls[nf]);
-      atomicAdd(&bsum[BN + wn + nf * 16 + cn], lq[nf]);
+      // fold the 4 lanes sharing column cn (lane, +16, +32, +48): ONE
+      // LDS atomic per address per wave instead of a 4-way serialized
+      // same-address conflict
+      ls[nf] += __shfl_down(ls[nf], 32);
+      ls[nf] += __shfl_down(ls[nf], 16);
+      lq[nf] += __shfl_down(lq[nf], 32);
+      lq[nf] += __shfl_down(lq[nf], 16);
+    }
+    if (lane < 16) {
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+        atomicAdd(&bsum[wn + nf * 16 + cn], ls[nf]);
+        atomicAdd(&bsum[BN + wn + nf * 16 + cn], lq[nf]);
+      }
     }
     __syncthreads();
     const int tiles_m_g = gridDim.x / tiles_n;
