@@ -123,7 +123,7 @@ __global__ void bn_stats_kernel(
 // strips over the block range (v1 ran C threads total: at C=64 a single
 // wave serially read every partial, latency-bound).
 extern "C" __global__ void bn_finalize_v2_kernel(
-    const float* __restrict__ partial, const int nblocks,
+    float* __restrict__ partial, const int nblocks, const int zero_src,
     const float* __restrict__ gamma,
     const float* __restrict__ beta, float* __restrict__ mean_out,
     float* __restrict__ invstd_out, float* __restrict__ scale_out,
@@ -142,13 +142,17 @@ extern "C" __global__ void bn_finalize_v2_kernel(
     for (; b + 12 < nblocks; b += 16) {
 #pragma unroll
       for (int u = 0; u < 4; ++u) {
-        sa[u] += partial[(long long)(b + 4 * u) * st + c];
-        qa[u] += partial[(long long)(b + 4 * u) * st + C + c];
+        const long long i0 = (long long)(b + 4 * u) * st + c;
+        sa[u] += partial[i0];
+        qa[u] += partial[i0 + C];
+        if (zero_src) { partial[i0] = 0.0f; partial[i0 + C] = 0.0f; }
       }
     }
     for (; b < nblocks; b += 4) {
-      sa[0] += partial[(long long)b * st + c];
-      qa[0] += partial[(long long)b * st + C + c];
+      const long long i0 = (long long)b * st + c;
+      sa[0] += partial[i0];
+      qa[0] += partial[i0 + C];
+      if (zero_src) { partial[i0] = 0.0f; partial[i0 + C] = 0.0f; }
     }
   }
   accs[strip][lane] = (sa[0] + sa[1]) + (sa[2] + sa[3]);
@@ -179,7 +183,7 @@ extern "C" __global__ void bn_finalize_v2_kernel(
 // stats. One thread per channel; thread c's reads of partial[b][c] are
 // coalesced across the warp for each fixed b.
 extern "C" __global__ void bn_finalize_kernel(
-    const float* __restrict__ partial, const int nblocks,
+    float* __restrict__ partial, const int nblocks, const int zero_src,
     const float* __restrict__ gamma,
     const float* __restrict__ beta, float* __restrict__ mean_out,
     float* __restrict__ invstd_out, float* __restrict__ scale_out,
@@ -196,13 +200,17 @@ extern "C" __global__ void bn_finalize_kernel(
   for (; b + 8 <= nblocks; b += 8) {
 #pragma unroll
     for (int u = 0; u < 8; ++u) {
-      sa[u] += partial[(long long)(b + u) * st + c];
-      qa[u] += partial[(long long)(b + u) * st + C + c];
+      const long long i0 = (long long)(b + u) * st + c;
+      sa[u] += partial[i0];
+      qa[u] += partial[i0 + C];
+      if (zero_src) { partial[i0] = 0.0f; partial[i0 + C] = 0.0f; }
     }
   }
   for (; b < nblocks; ++b) {
-    sa[0] += partial[(long long)b * st + c];
-    qa[0] += partial[(long long)b * st + C + c];
+    const long long i0 = (long long)b * st + c;
+    sa[0] += partial[i0];
+    qa[0] += partial[i0 + C];
+    if (zero_src) { partial[i0] = 0.0f; partial[i0 + C] = 0.0f; }
   }
   float s = 0.0f, q = 0.0f;
 #pragma unroll
@@ -528,21 +536,23 @@ extern "C" void launch_bn_stats(const void* x, float* partial, int grid,
   }
 }
 
-extern "C" void launch_bn_finalize(const float* partial, int nblocks,
+extern "C" void launch_bn_finalize(float* partial, int nblocks, int zero_src,
                                    const float* gamma,
                                    const float* beta, float* mean, float* invstd,
                                    float* scale, float* shift, float* rmean,
                                    float* rvar, float momentum, float eps,
                                    long long M, int C, hipStream_t s) {
+  // zero_src: store 0 back after each read — returns a pooled pre_part
+  // buffer to the pool clean, killing the per-step FillFunctor launches
   if (env_ll("EDL_BN_FIN_V2", 1)) {
     hipLaunchKernelGGL(bn_finalize_v2_kernel, dim3((C + 63) / 64), dim3(256),
-                       0, s, partial, nblocks, gamma, beta, mean, invstd,
-                       scale, shift, rmean, rvar, momentum, eps, M, C);
+                       0, s, partial, nblocks, zero_src, gamma, beta, mean,
+                       invstd, scale, shift, rmean, rvar, momentum, eps, M, C);
     return;
   }
   hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 255) / 256), dim3(256), 0, s,
-                     partial, nblocks, gamma, beta, mean, invstd, scale, shift,
-                     rmean, rvar, momentum, eps, M, C);
+                     partial, nblocks, zero_src, gamma, beta, mean, invstd,
+                     scale, shift, rmean, rvar, momentum, eps, M, C);
 }
 
 extern "C" void launch_bn_bwd_finalize(const float* partial, int nblocks,

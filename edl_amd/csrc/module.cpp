@@ -1,5 +1,6 @@
 // Python bindings for the edl_amd CDNA4 kernel layer (edl_amd._C).
 // Compiled by hipcc directly (no hipify, no CUDA path) — see build_hip.py.
+#include <unordered_map>
 #include <torch/extension.h>
 
 #include <c10/hip/HIPStream.h>
@@ -21,7 +22,7 @@ extern "C" int bn_stats_grid(long long, int);
 extern "C" int bn_bwd_grid(long long, int);
 extern "C" void launch_bn_stats(const void*, float*, int, long long, int,
                                 hipStream_t);
-extern "C" void launch_bn_finalize(const float*, int, const float*, const float*,
+extern "C" void launch_bn_finalize(float*, int, int, const float*, const float*,
                                    float*, float*, float*, float*, float*, float*,
                                    float, float, long long, int, hipStream_t);
 extern "C" void launch_bn_apply(const void*, const void*, void*, unsigned char*,
@@ -196,7 +197,8 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor gamma,
     launch_bn_stats(x.data_ptr(), partial.data_ptr<float>(), grid, M, (int)C,
                     s);
   }
-  launch_bn_finalize(partial.data_ptr<float>(), fin_grid, gamma.data_ptr<float>(),
+  launch_bn_finalize(partial.data_ptr<float>(), fin_grid,
+                     pre_part.has_value() ? 1 : 0, gamma.data_ptr<float>(),
                      beta.data_ptr<float>(), mean.data_ptr<float>(),
                      invstd.data_ptr<float>(), scale.data_ptr<float>(),
                      shift.data_ptr<float>(),
@@ -294,6 +296,24 @@ torch::Tensor gemm_bt(torch::Tensor a, torch::Tensor b) {
   return c;
 }
 
+
+static torch::Tensor part_pool_get(int rows, int64_t cols,
+                                   torch::TensorOptions opts) {
+  // Capped BN-stats partial buffers are accumulated with atomics, so they
+  // must start zeroed. Pool them per (device, shape): zeroed ONCE here,
+  // and bn_finalize(zero_src=1) stores zeros back after reading, so reuse
+  // needs no per-step fill launch. If a producer's partials are ever
+  // dropped unconsumed, the Python side zeroes them (bnrelu.forward).
+  static std::unordered_map<int64_t, torch::Tensor> pool;
+  const int64_t key = ((int64_t)opts.device().index() << 48) |
+                      ((int64_t)rows << 28) | cols;
+  auto it = pool.find(key);
+  if (it != pool.end()) return it->second;
+  auto t = torch::zeros({rows, cols}, opts);
+  pool.emplace(key, t);
+  return t;
+}
+
 std::vector<torch::Tensor> gemm_bt_stats(torch::Tensor a, torch::Tensor b) {
   // like gemm_bt, additionally returning BN stats partials [tiles_m, 2N]
   // of the bf16-rounded output (the following BN skips its stats kernel)
@@ -311,8 +331,8 @@ std::vector<torch::Tensor> gemm_bt_stats(torch::Tensor a, torch::Tensor b) {
   const int tiles_m = gemm_bt_tiles_m(M, N);
   const int rows = std::min(tiles_m, 192);
   auto part = tiles_m > rows
-                  ? torch::zeros({rows, 2 * N},
-                                 a.options().dtype(torch::kFloat32))
+                  ? part_pool_get(rows, 2 * N,
+                                  a.options().dtype(torch::kFloat32))
                   : torch::empty({rows, 2 * N},
                                  a.options().dtype(torch::kFloat32));
   launch_gemm_bt(ac.data_ptr(), bc.data_ptr(), c.data_ptr(), M, N, K,
@@ -426,8 +446,8 @@ std::vector<torch::Tensor> conv3x3_small_fwd_stats(torch::Tensor x,
   const int tiles_m = (int)((M + 255) / 256);
   const int rows = std::min(tiles_m, 192);
   auto bpart = tiles_m > rows
-                   ? torch::zeros({rows, 2 * cout_real},
-                                  x.options().dtype(torch::kFloat32))
+                   ? part_pool_get(rows, 2 * cout_real,
+                                   x.options().dtype(torch::kFloat32))
                    : torch::empty({rows, 2 * cout_real},
                                   x.options().dtype(torch::kFloat32));
   launch_conv3x3_small(xp.data_ptr(), w3s.data_ptr(), y.data_ptr(), (int)M,
@@ -766,8 +786,8 @@ std::vector<torch::Tensor> conv3x3_fwd_stats(torch::Tensor x, torch::Tensor w3,
   const int tiles_m = conv3x3_tiles_m((int)M, Cout);
   const int rows = std::min(tiles_m, 192);
   auto bpart = tiles_m > rows
-                   ? torch::zeros({rows, 2 * Cout},
-                                  x.options().dtype(torch::kFloat32))
+                   ? part_pool_get(rows, 2 * Cout,
+                                   x.options().dtype(torch::kFloat32))
                    : torch::empty({rows, 2 * Cout},
                                   x.options().dtype(torch::kFloat32));
   launch_conv3x3(xp.data_ptr(), w3.data_ptr(), y.data_ptr(), (int)M, Cout, Cin,
