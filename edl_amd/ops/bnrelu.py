@@ -147,6 +147,11 @@ class BNReLU2d(nn.Module):
             if self.training or not torch.is_grad_enabled():
                 return self._fused(x, res,
                                    partials if self.training else None)
+        if partials is not None:
+            # capped partials come from a pooled buffer that the finalize
+            # kernel normally returns clean; if we're not consuming them,
+            # scrub before dropping so the pool stays zeroed
+            partials.zero_()
         return self._fallback(x, res)
 
     def extra_repr(self):

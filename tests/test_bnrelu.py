@@ -178,6 +178,41 @@ class TestConvFusedStats:
         assert torch.allclose(m1.bn.running_var, m2.bn.running_var,
                               atol=1e-4, rtol=1e-4)
 
+    @pytest.mark.parametrize("shape", [
+        (4, 64, 64, 112, 1),   # gemm_bt epilogue, tiles_m=196 > 192 cap
+        (4, 64, 64, 112, 3),   # conv3x3 epilogue, capped
+        (2, 3, 32, 160, 3),    # stem small-kernel epilogue, capped
+    ])
+    def test_capped_pooled_partials_stay_clean(self, shape, monkeypatch):
+        """Shapes with tiles_m > 192 accumulate stats into a POOLED zeroed
+        buffer with atomics; bn_finalize(zero_src) must return it to the
+        pool clean. Two successive steps must both match the unfused-stats
+        reference — step 2 is wrong if the pool came back dirty."""
+        self._gpu()
+        import copy
+
+        import edl_amd.ops.conv as conv_mod
+        from edl_amd.models.resnet_vd import ConvBN
+
+        n, ci, co, hw, k = shape
+        torch.manual_seed(33)
+        m1 = ConvBN(ci, co, k).cuda().train()
+        m1.conv.to(torch.bfloat16)
+        m2 = copy.deepcopy(m1)
+        for it in range(2):
+            x = torch.randn(n, ci, hw, hw, device="cuda").to(torch.bfloat16)
+            x = x.contiguous(memory_format=torch.channels_last)
+            monkeypatch.setattr(conv_mod, "_BN_STATS_FUSED", True)
+            y1 = m1(x)
+            monkeypatch.setattr(conv_mod, "_BN_STATS_FUSED", False)
+            y2 = m2(x)
+            assert torch.allclose(y1.float(), y2.float(), atol=1e-2,
+                                  rtol=1e-2), (it, (y1.float() - y2.float()).abs().max().item())
+            assert torch.allclose(m1.bn.running_mean, m2.bn.running_mean,
+                                  atol=1e-4, rtol=1e-4), it
+            assert torch.allclose(m1.bn.running_var, m2.bn.running_var,
+                                  atol=1e-4, rtol=1e-4), it
+
     def test_bottleneck_trains_with_fused_stats(self, monkeypatch):
         """Full bottleneck fwd+bwd with conv-folded stats: finite grads,
         same trajectory as the unfused-stats path."""
