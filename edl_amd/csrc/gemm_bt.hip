@@ -32,7 +32,7 @@ template <int BM, int BN, int WAVES_M, int WAVES_N, bool SPLITK = false,
 __global__ __launch_bounds__(256, 2) void gemm_bt_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B,
     void* __restrict__ C_any, const int M, const int N, const int K,
-    float* __restrict__ bn_part = nullptr) {
+    float* __restrict__ bn_part = nullptr, const int ldsb = 0) {
   // bn_part: per-channel sum/sumsq partials of the bf16-rounded output,
   // [tiles_m, 2N] — lets the following BatchNorm skip its stats kernel
   // (see conv3x3.hip BN_PART)
@@ -224,7 +224,7 @@ __global__ __launch_bounds__(256, 2) void gemm_bt_kernel(
           for (int nf = 0; nf < 4; ++nf) {
             atomicAdd(&crow[nf * 16], acc[mf][nf][reg]);
           }
-        } else {
+        } else if (!ldsb) {
           bf16* crow = C + (long long)m * N + n0 + wn + cn;
 #pragma unroll
           for (int nf = 0; nf < 4; ++nf) {
@@ -238,6 +238,43 @@ __global__ __launch_bounds__(256, 2) void gemm_bt_kernel(
           }
         }
       }
+      if (SPLITK || !ldsb) continue;
+      // EDL_BT_STORE_LDS: stage the wave's 64x64 bf16 tile in LDS
+      // ([64][72] row-major, 16-B pad breaks the b128 bank pattern), then
+      // store full 128-B lines. Replaces 64 2-B MUBUF stores/thread with
+      // 8 dwordx4. Intra-wave only: no __syncthreads needed, just a
+      // lgkmcnt wait before the cross-lane reads. Staging buffers are
+      // dead past the K loop; bsum reuse below comes after a barrier.
+      bf16* wtile = (bf16*)smem + wave * (64 * 72);
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int m = m0 + wm + mf * 16 + r4 + reg;
+#pragma unroll
+        for (int nf = 0; nf < 4; ++nf) {
+          const bf16 yb = __float2bfloat16(acc[mf][nf][reg]);
+          wtile[(mf * 16 + r4 + reg) * 72 + nf * 16 + cn] = yb;
+          if (bn_part != nullptr && m < M) {
+            const float yv = __bfloat162float(yb);
+            ls[nf] += yv;
+            lq[nf] = fmaf(yv, yv, lq[nf]);
+          }
+        }
+      }
+    }
+  }
+  if (!SPLITK && ldsb) {
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    bf16* wtile = (bf16*)smem + wave * (64 * 72);
+    const int rrow = lane >> 3, rcol = (lane & 7) * 8;
+#pragma unroll
+    for (int it = 0; it < 8; ++it) {
+      const int r = it * 8 + rrow;
+      const bf16x8 v = *(const bf16x8*)__builtin_assume_aligned(
+          wtile + r * 72 + rcol, 16);
+      const int m = m0 + wm + r;
+      if (m < M)
+        *(bf16x8*)__builtin_assume_aligned(
+            C + (long long)m * N + n0 + wn + rcol, 16) = v;
     }
   }
 
@@ -298,6 +335,17 @@ extern "C" int gemm_bt_tiles_m(int M, int N) {
   return (M + bm - 1) / bm;
 }
 
+static int use_store_lds() {
+  // EDL_BT_STORE_LDS=1: bounce epilogue stores through LDS (dwordx4
+  // full-line stores instead of 64 2-B stores per thread). Opt-in until
+  // measured on the flagship step.
+  static const int v = []() {
+    const char* e = getenv("EDL_BT_STORE_LDS");
+    return e ? atoi(e) : 0;
+  }();
+  return v;
+}
+
 extern "C" void launch_gemm_bt(const void* A, const void* B, void* C, int M,
                                int N, int K, float* bn_part, hipStream_t s) {
   if (N % 128 == 0) {
@@ -307,11 +355,11 @@ extern "C" void launch_gemm_bt(const void* A, const void* B, void* C, int M,
     if (use_v2())
       hipLaunchKernelGGL((gemm_bt_kernel<BM, BN, 2, 2, false, true>), dim3(grid),
                          dim3(256), lds_bytes, s, (const bf16*)A, (const bf16*)B,
-                         C, M, N, K, bn_part);
+                         C, M, N, K, bn_part, use_store_lds());
     else
       hipLaunchKernelGGL((gemm_bt_kernel<BM, BN, 2, 2>), dim3(grid), dim3(256),
                          lds_bytes, s, (const bf16*)A, (const bf16*)B, C, M, N,
-                         K, bn_part);
+                         K, bn_part, use_store_lds());
   } else {  // N % 64 == 0
     constexpr int BM = 256, BN = 64;
     const int grid = ((M + BM - 1) / BM) * (N / BN);
@@ -319,11 +367,11 @@ extern "C" void launch_gemm_bt(const void* A, const void* B, void* C, int M,
     if (use_v2())
       hipLaunchKernelGGL((gemm_bt_kernel<BM, BN, 4, 1, false, true>), dim3(grid),
                          dim3(256), lds_bytes, s, (const bf16*)A, (const bf16*)B,
-                         C, M, N, K, bn_part);
+                         C, M, N, K, bn_part, use_store_lds());
     else
       hipLaunchKernelGGL((gemm_bt_kernel<BM, BN, 4, 1>), dim3(grid), dim3(256),
                          lds_bytes, s, (const bf16*)A, (const bf16*)B, C, M, N,
-                         K, bn_part);
+                         K, bn_part, use_store_lds());
   }
 }
 
