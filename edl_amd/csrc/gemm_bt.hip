@@ -214,37 +214,11 @@ __global__ __launch_bounds__(256, 2) void gemm_bt_kernel(
   float ls[4] = {0, 0, 0, 0}, lq[4] = {0, 0, 0, 0};
 #pragma unroll
   for (int mf = 0; mf < 4; ++mf) {
-#pragma unroll
-    for (int reg = 0; reg < 4; ++reg) {
-      const int m = m0 + wm + mf * 16 + r4 + reg;
-      if (m < M) {
-        if (SPLITK) {
-          float* crow = Cf + (long long)m * N + n0 + wn + cn;
-#pragma unroll
-          for (int nf = 0; nf < 4; ++nf) {
-            atomicAdd(&crow[nf * 16], acc[mf][nf][reg]);
-          }
-        } else if (!ldsb) {
-          bf16* crow = C + (long long)m * N + n0 + wn + cn;
-#pragma unroll
-          for (int nf = 0; nf < 4; ++nf) {
-            const bf16 yb = __float2bfloat16(acc[mf][nf][reg]);
-            crow[nf * 16] = yb;
-            if (bn_part != nullptr) {
-              const float yv = __bfloat162float(yb);
-              ls[nf] += yv;
-              lq[nf] = fmaf(yv, yv, lq[nf]);
-            }
-          }
-        }
-      }
-      if (SPLITK || !ldsb) continue;
+    if (!SPLITK && ldsb) {
       // EDL_BT_STORE_LDS: stage the wave's 64x64 bf16 tile in LDS
       // ([64][72] row-major, 16-B pad breaks the b128 bank pattern), then
-      // store full 128-B lines. Replaces 64 2-B MUBUF stores/thread with
-      // 8 dwordx4. Intra-wave only: no __syncthreads needed, just a
-      // lgkmcnt wait before the cross-lane reads. Staging buffers are
-      // dead past the K loop; bsum reuse below comes after a barrier.
+      // store full 128-B lines below. Intra-wave only: no __syncthreads
+      // needed, just a lgkmcnt wait before the cross-lane reads.
       bf16* wtile = (bf16*)smem + wave * (64 * 72);
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg) {
@@ -257,6 +231,32 @@ __global__ __launch_bounds__(256, 2) void gemm_bt_kernel(
             const float yv = __bfloat162float(yb);
             ls[nf] += yv;
             lq[nf] = fmaf(yv, yv, lq[nf]);
+          }
+        }
+      }
+      continue;
+    }
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int m = m0 + wm + mf * 16 + r4 + reg;
+      if (m < M) {
+        if (SPLITK) {
+          float* crow = Cf + (long long)m * N + n0 + wn + cn;
+#pragma unroll
+          for (int nf = 0; nf < 4; ++nf) {
+            atomicAdd(&crow[nf * 16], acc[mf][nf][reg]);
+          }
+        } else {
+          bf16* crow = C + (long long)m * N + n0 + wn + cn;
+#pragma unroll
+          for (int nf = 0; nf < 4; ++nf) {
+            const bf16 yb = __float2bfloat16(acc[mf][nf][reg]);
+            crow[nf * 16] = yb;
+            if (bn_part != nullptr) {
+              const float yv = __bfloat162float(yb);
+              ls[nf] += yv;
+              lq[nf] = fmaf(yv, yv, lq[nf]);
+            }
           }
         }
       }

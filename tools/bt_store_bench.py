@@ -55,9 +55,11 @@ if __name__ == "__main__":
         worker()
         sys.exit(0)
     for v in ("0", "1"):
-        env = dict(os.environ, EDL_BT_STORE_LDS=v)
+        repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+        env = dict(os.environ, EDL_BT_STORE_LDS=v,
+                   PYTHONPATH=repo + os.pathsep + os.environ.get("PYTHONPATH", ""))
         r = subprocess.run([sys.executable, __file__, "--worker"], env=env,
-                           capture_output=True, text=True, timeout=300)
+                           cwd=repo, capture_output=True, text=True, timeout=300)
         sys.stdout.write(r.stdout)
         if r.returncode != 0:
             sys.stderr.write(r.stderr[-2000:])
