@@ -480,6 +480,179 @@ __global__ void bn_bwd_dx_kernel(
   }
 }
 
+
+// B-fused: reduce + finalize + dx in ONE launch (EDL_BN_BWD_FUSED).
+// The three-kernel backward pays two launch/drain boundaries per BN layer
+// (53 layers in resnet50_vd). Here the grid rendezvouses in device memory
+// instead: every block writes its partial row, the LAST-arriving block
+// reduces partials -> sums (+ direct-grad accumulation), release-stores a
+// flag, and the rest spin on it (bounded; agent-scope atomics because L2
+// is per-XCD). grid <= 192 (bn_bwd_grid) guarantees co-residency on 256
+// CUs, so the spin cannot deadlock. ws = persistent int[4] workspace
+// {arrive, flag, depart, pad}; the last block OUT resets it, so replays
+// (hipGraph) and back-to-back layers reuse it with no host zeroing.
+template <bool RELU, bool ADD>
+__global__ __launch_bounds__(256, 2) void bn_bwd_fused_kernel(
+    const bf16* __restrict__ dy, const unsigned char* __restrict__ mask,
+    const bf16* __restrict__ x, const float* __restrict__ mean,
+    const float* __restrict__ invstd, const float* __restrict__ gamma,
+    float* __restrict__ partial, float* __restrict__ sums,
+    float* __restrict__ db_acc, float* __restrict__ dg_acc,
+    bf16* __restrict__ dx, bf16* __restrict__ dres, int* __restrict__ ws,
+    const long long M, const int C) {
+  __shared__ float lsum[2 * 2048];
+  __shared__ int role;
+  const int c8 = C >> 3;
+  const int tpr = c8;
+  const int tid = blockIdx.x * blockDim.x + threadIdx.x;
+  const long long rstride = ((long long)gridDim.x * blockDim.x) / tpr;
+  const int lane_c = (int)(threadIdx.x % tpr);
+  const int c0 = lane_c * 8;
+
+  for (int i = threadIdx.x; i < 2 * C; i += blockDim.x) lsum[i] = 0.0f;
+  __syncthreads();
+
+  // ---- phase 1: per-block channel reduction (bn_bwd_reduce, ST=4) ----
+  float mu[8], is[8];
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    mu[i] = mean[c0 + i];
+    is[i] = invstd[c0 + i];
+  }
+  float s1[8] = {0}, s2[8] = {0};
+  long long r = tid / tpr;
+  for (; r + 3 * rstride < M; r += 4 * rstride) {
+#pragma unroll
+    for (int u = 0; u < 2; ++u) {
+      const long long r0 = r + 2 * u * rstride, r1 = r + (2 * u + 1) * rstride;
+      const long long e0 = r0 * C + c0, e1 = r1 * C + c0;
+      F8 g0 = load8(dy + e0), g1 = load8(dy + e1);
+      F8 x0 = load8(x + e0), x1 = load8(x + e1);
+      if (RELU) {
+        const unsigned m0 = mask[r0 * c8 + lane_c];
+        const unsigned m1 = mask[r1 * c8 + lane_c];
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+          g0.v[i] = (m0 >> i) & 1 ? g0.v[i] : 0.0f;
+          g1.v[i] = (m1 >> i) & 1 ? g1.v[i] : 0.0f;
+        }
+      }
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        s1[i] += g0.v[i] + g1.v[i];
+        s2[i] = fmaf(g0.v[i], (x0.v[i] - mu[i]) * is[i],
+                     fmaf(g1.v[i], (x1.v[i] - mu[i]) * is[i], s2[i]));
+      }
+    }
+  }
+  for (; r < M; r += rstride) {
+    const long long eoff = r * C + c0;
+    F8 g = load8(dy + eoff);
+    F8 xv = load8(x + eoff);
+    if (RELU) {
+      const unsigned mb = mask[r * c8 + lane_c];
+#pragma unroll
+      for (int i = 0; i < 8; ++i) g.v[i] = (mb >> i) & 1 ? g.v[i] : 0.0f;
+    }
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      s1[i] += g.v[i];
+      s2[i] = fmaf(g.v[i], (xv.v[i] - mu[i]) * is[i], s2[i]);
+    }
+  }
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    atomicAdd(&lsum[c0 + i], s1[i]);
+    atomicAdd(&lsum[C + c0 + i], s2[i]);
+  }
+  __syncthreads();
+  {
+    float* out = partial + (long long)blockIdx.x * 2 * C;
+    for (int i = threadIdx.x; i < 2 * C; i += blockDim.x) out[i] = lsum[i];
+  }
+
+  // ---- rendezvous: last-arriving block finalizes ----
+  __threadfence();
+  if (threadIdx.x == 0) {
+    const int arrive = __hip_atomic_fetch_add(&ws[0], 1, __ATOMIC_ACQ_REL,
+                                              __HIP_MEMORY_SCOPE_AGENT);
+    role = (arrive == (int)gridDim.x - 1);
+  }
+  __syncthreads();
+  if (role) {
+    const int g = gridDim.x;
+    const long long st = 2 * C;
+    for (int i = threadIdx.x; i < 2 * C; i += blockDim.x) {
+      float sa[4] = {0, 0, 0, 0};
+      int b = 0;
+      for (; b + 4 <= g; b += 4)
+#pragma unroll
+        for (int u = 0; u < 4; ++u) sa[u] += partial[(long long)(b + u) * st + i];
+      for (; b < g; ++b) sa[0] += partial[(long long)b * st + i];
+      const float v = (sa[0] + sa[1]) + (sa[2] + sa[3]);
+      sums[i] = v;
+      if (i < C) {
+        if (db_acc) db_acc[i] += v;
+      } else if (dg_acc) {
+        dg_acc[i - C] += v;
+      }
+    }
+    __syncthreads();
+    __threadfence();
+    if (threadIdx.x == 0)
+      __hip_atomic_store(&ws[1], 1, __ATOMIC_RELEASE,
+                         __HIP_MEMORY_SCOPE_AGENT);
+  } else if (threadIdx.x == 0) {
+    // bounded spin: ~50M sleeps is minutes — unreachable unless the
+    // finalizer died; never hangs the box
+    for (long long it = 0; it < 50000000LL; ++it) {
+      if (__hip_atomic_load(&ws[1], __ATOMIC_ACQUIRE,
+                            __HIP_MEMORY_SCOPE_AGENT))
+        break;
+      __builtin_amdgcn_s_sleep(32);
+    }
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    const int depart = __hip_atomic_fetch_add(&ws[2], 1, __ATOMIC_ACQ_REL,
+                                              __HIP_MEMORY_SCOPE_AGENT);
+    if (depart == (int)gridDim.x - 1) {  // everyone passed the spin: reset
+      __hip_atomic_store(&ws[0], 0, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+      __hip_atomic_store(&ws[1], 0, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+      __hip_atomic_store(&ws[2], 0, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    }
+  }
+
+  // ---- phase 2: dx (+dres), training form (bn_bwd_dx) ----
+  const float inv_m = 1.0f / (float)M;
+  const long long total = M * c8;
+  long long i2 = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long stride2 = (long long)gridDim.x * blockDim.x;
+  for (; i2 < total; i2 += stride2) {
+    const int oct = (int)(i2 % c8);
+    const int cc0 = oct * 8;
+    const long long eoff = (i2 / c8) * C + cc0;
+    F8 g = load8(dy + eoff);
+    if (RELU) {
+      const unsigned mb = mask[i2];
+#pragma unroll
+      for (int k = 0; k < 8; ++k) g.v[k] = (mb >> k) & 1 ? g.v[k] : 0.0f;
+    }
+    if (ADD) store8(dres + eoff, g);
+    F8 xv = load8(x + eoff);
+    F8 o;
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      const int c = cc0 + k;
+      const float isv = invstd[c];
+      const float xhat = (xv.v[k] - mean[c]) * isv;
+      o.v[k] = gamma[c] * isv *
+               (g.v[k] - sums[c] * inv_m - xhat * sums[C + c] * inv_m);
+    }
+    store8(dx + eoff, o);
+  }
+}
+
 // ---------------- launchers ----------------
 
 static long long env_ll(const char* name, long long dflt) {
@@ -648,4 +821,23 @@ extern "C" void launch_bn_bwd_dx(const void* dy, const unsigned char* mask,
     else CASE(false, false, false);
   }
 #undef CASE
+}
+
+extern "C" void launch_bn_bwd_fused(const void* dy, const unsigned char* mask,
+                                    const void* x, const float* mean,
+                                    const float* invstd, const float* gamma,
+                                    float* partial, float* sums, float* db_acc,
+                                    float* dg_acc, void* dx, void* dres,
+                                    int* ws, int grid, long long M, int C,
+                                    bool relu, bool add, hipStream_t s) {
+#define FCASE(R, A)                                                            \
+  hipLaunchKernelGGL((bn_bwd_fused_kernel<R, A>), dim3(grid), dim3(256), 0, s, \
+                     (const bf16*)dy, (R) ? mask : nullptr, (const bf16*)x,    \
+                     mean, invstd, gamma, partial, sums, db_acc, dg_acc,       \
+                     (bf16*)dx, (bf16*)dres, ws, M, C)
+  if (relu && add) FCASE(true, true);
+  else if (relu) FCASE(true, false);
+  else if (add) FCASE(false, true);
+  else FCASE(false, false);
+#undef FCASE
 }

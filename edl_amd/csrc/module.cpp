@@ -28,6 +28,11 @@ extern "C" void launch_bn_finalize(float*, int, int, const float*, const float*,
 extern "C" void launch_bn_apply(const void*, const void*, void*, unsigned char*,
                                 const float*, const float*, long long, int,
                                 bool, bool, hipStream_t);
+extern "C" void launch_bn_bwd_fused(const void*, const unsigned char*,
+                                    const void*, const float*, const float*,
+                                    const float*, float*, float*, float*,
+                                    float*, void*, void*, int*, int, long long,
+                                    int, bool, bool, hipStream_t);
 extern "C" void launch_bn_bwd_reduce(const void*, const unsigned char*,
                                      const void*, const float*, const float*,
                                      float*, int, long long, int, bool,
@@ -248,9 +253,6 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, c10::optional<torch::Tensor>
               "bn_bwd: relu path needs the fwd mask");
   const unsigned char* mp =
       relu ? (const unsigned char*)msk->data_ptr() : nullptr;
-  launch_bn_bwd_reduce(dyc.data_ptr(), mp, x.data_ptr(),
-                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                       partial.data_ptr<float>(), grid, M, (int)C, relu, s);
   // direct-grad mode: accumulate dgamma/dbeta straight into the params'
   // bucket-view gradients (skips the per-param AccumulateGrad kernels)
   float* dg_acc = nullptr;
@@ -267,6 +269,30 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, c10::optional<torch::Tensor>
                 "bn_bwd: dbeta_acc fp32 [C]");
     db_acc = dbeta_acc->data_ptr<float>();
   }
+  // read per call (not static) so tests can A/B via os.environ
+  const char* fenv = getenv("EDL_BN_BWD_FUSED");
+  const bool fused_bwd = fenv ? atoi(fenv) != 0 : false;
+  if (fused_bwd && training) {
+    // one launch: reduce + last-block finalize + dx (grid-wide flag sync);
+    // ws = persistent per-device {arrive, flag, depart} ints, self-resetting
+    static std::unordered_map<int, torch::Tensor> ws_pool;
+    const int dev = (int)x.device().index();
+    auto wit = ws_pool.find(dev);
+    if (wit == ws_pool.end())
+      wit = ws_pool.emplace(dev, torch::zeros({4},
+          x.options().dtype(torch::kInt32))).first;
+    launch_bn_bwd_fused(dyc.data_ptr(), mp, x.data_ptr(),
+                        mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                        gamma.data_ptr<float>(), partial.data_ptr<float>(),
+                        sums.data_ptr<float>(), db_acc, dg_acc, dx.data_ptr(),
+                        add ? dres.data_ptr() : nullptr,
+                        wit->second.data_ptr<int>(), grid, M, (int)C, relu,
+                        add, s);
+    return {dx, sums[1], sums[0], dres};
+  }
+  launch_bn_bwd_reduce(dyc.data_ptr(), mp, x.data_ptr(),
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       partial.data_ptr<float>(), grid, M, (int)C, relu, s);
   launch_bn_bwd_finalize(partial.data_ptr<float>(), grid, sums.data_ptr<float>(),
                          (int)C, db_acc, dg_acc, s);
   launch_bn_bwd_dx(dyc.data_ptr(), mp, x.data_ptr(),

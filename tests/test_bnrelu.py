@@ -242,3 +242,47 @@ class TestConvFusedStats:
             if p1.grad is not None:
                 assert torch.allclose(p1.grad.float(), p2.grad.float(),
                                       atol=5e-2, rtol=5e-2), p1.shape
+
+
+@pytest.mark.gpu
+class TestFusedBwdSingleLaunch:
+    """EDL_BN_BWD_FUSED: reduce+finalize+dx as ONE kernel with a grid-wide
+    flag rendezvous (bnrelu.hip bn_bwd_fused_kernel). Two back-to-back
+    calls verify the self-resetting workspace."""
+
+    @pytest.mark.parametrize("relu,add", [(True, False), (True, True),
+                                          (False, False)])
+    @pytest.mark.parametrize("mc", [(4 * 56 * 56, 64), (4 * 7 * 7, 2048)])
+    def test_matches_unfused(self, relu, add, mc, monkeypatch):
+        if not torch.cuda.is_available():
+            pytest.skip("no GPU")
+        import os
+
+        from edl_amd.ops import ext
+
+        M, C = mc
+        torch.manual_seed(40)
+        for it in range(2):  # second call checks the ws reset
+            x2d = torch.randn(M, C, device="cuda").to(torch.bfloat16)
+            gamma = torch.rand(C, device="cuda") + 0.5
+            beta = torch.randn(C, device="cuda")
+            res = (torch.randn(M, C, device="cuda").to(torch.bfloat16)
+                   if add else None)
+            y, mean, invstd, mask = ext().bn_fwd_train(
+                x2d, gamma, beta, torch.zeros(C, device="cuda"),
+                torch.ones(C, device="cuda"), 0.1, 1e-5, res, relu, None)
+            dy = torch.randn(M, C, device="cuda").to(torch.bfloat16)
+
+            monkeypatch.setenv("EDL_BN_BWD_FUSED", "0")
+            r0 = ext().bn_bwd(dy, mask, x2d, mean, invstd, gamma, relu, add,
+                              True, None, None)
+            monkeypatch.setenv("EDL_BN_BWD_FUSED", "1")
+            r1 = ext().bn_bwd(dy, mask, x2d, mean, invstd, gamma, relu, add,
+                              True, None, None)
+            monkeypatch.setenv("EDL_BN_BWD_FUSED", "0")
+            names = ["dx", "dgamma", "dbeta", "dres"]
+            for n, a, b in zip(names, r0, r1):
+                if a is None or (hasattr(a, "numel") and a.numel() == 0):
+                    continue
+                assert torch.allclose(a.float(), b.float(), atol=2e-2,
+                                      rtol=2e-2), (it, n)
