@@ -281,12 +281,16 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, c10::optional<torch::Tensor>
     if (wit == ws_pool.end())
       wit = ws_pool.emplace(dev, torch::zeros({4},
           x.options().dtype(torch::kInt32))).first;
+    // co-residency bound: every block must be resident for the spin to
+    // resolve (256 CUs, >=1 block/CU) — clamp whatever the grid-cap env
+    // says; partial has `grid` rows, the first fgrid are used
+    const int fgrid = grid > 256 ? 256 : grid;
     launch_bn_bwd_fused(dyc.data_ptr(), mp, x.data_ptr(),
                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
                         gamma.data_ptr<float>(), partial.data_ptr<float>(),
                         sums.data_ptr<float>(), db_acc, dg_acc, dx.data_ptr(),
                         add ? dres.data_ptr() : nullptr,
-                        wit->second.data_ptr<int>(), grid, M, (int)C, relu,
+                        wit->second.data_ptr<int>(), fgrid, M, (int)C, relu,
                         add, s);
     return {dx, sums[1], sums[0], dres};
   }
