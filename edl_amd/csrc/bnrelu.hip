@@ -483,14 +483,18 @@ __global__ void bn_bwd_dx_kernel(
 
 // B-fused: reduce + finalize + dx in ONE launch (EDL_BN_BWD_FUSED).
 // The three-kernel backward pays two launch/drain boundaries per BN layer
-// (53 layers in resnet50_vd). Here the grid rendezvouses in device memory
-// instead: every block writes its partial row, the LAST-arriving block
-// reduces partials -> sums (+ direct-grad accumulation), release-stores a
-// flag, and the rest spin on it (bounded; agent-scope atomics because L2
-// is per-XCD). grid <= 192 (bn_bwd_grid) guarantees co-residency on 256
-// CUs, so the spin cannot deadlock. ws = persistent int[4] workspace
-// {arrive, flag, depart, pad}; the last block OUT resets it, so replays
-// (hipGraph) and back-to-back layers reuse it with no host zeroing.
+// (53 layers in resnet50_vd). Here the grid rendezvouses in device memory:
+// every block writes its partial row and bumps an arrival counter; the
+// first F blocks wait for full arrival, then finalize a channel SLICE each
+// (parallel finalize — a single finalizer block measured 45+ us on 512
+// partial rows) and bump a done counter; everyone spins RELAXED (an
+// acquire per poll invalidates L2 every iteration — measured as an ~85 us
+// rendezvous storm) with ONE acquire fence on exit. grid <= 512 with
+// __launch_bounds__(256, 2) guarantees co-residency on 256 CUs (the spin
+// cannot deadlock); all spins are iteration-bounded regardless. ws = a
+// persistent int[4] {arrive, done, depart, pad}: the last block OUT
+// resets it, so hipGraph replays and back-to-back layers need no host
+// zeroing.
 template <bool RELU, bool ADD>
 __global__ __launch_bounds__(256, 2) void bn_bwd_fused_kernel(
     const bf16* __restrict__ dy, const unsigned char* __restrict__ mask,
@@ -499,9 +503,8 @@ __global__ __launch_bounds__(256, 2) void bn_bwd_fused_kernel(
     float* __restrict__ partial, float* __restrict__ sums,
     float* __restrict__ db_acc, float* __restrict__ dg_acc,
     bf16* __restrict__ dx, bf16* __restrict__ dres, int* __restrict__ ws,
-    const long long M, const int C) {
+    const int nfin, const long long M, const int C) {
   __shared__ float lsum[2 * 2048];
-  __shared__ int role;
   const int c8 = C >> 3;
   const int tpr = c8;
   const int tid = blockIdx.x * blockDim.x + threadIdx.x;
@@ -571,23 +574,34 @@ __global__ __launch_bounds__(256, 2) void bn_bwd_fused_kernel(
     for (int i = threadIdx.x; i < 2 * C; i += blockDim.x) out[i] = lsum[i];
   }
 
-  // ---- rendezvous: last-arriving block finalizes ----
+  // ---- rendezvous ----
   __threadfence();
-  if (threadIdx.x == 0) {
-    const int arrive = __hip_atomic_fetch_add(&ws[0], 1, __ATOMIC_ACQ_REL,
-                                              __HIP_MEMORY_SCOPE_AGENT);
-    role = (arrive == (int)gridDim.x - 1);
-  }
-  __syncthreads();
-  if (role) {
-    const int g = gridDim.x;
+  if (threadIdx.x == 0)
+    __hip_atomic_fetch_add(&ws[0], 1, __ATOMIC_RELEASE,
+                           __HIP_MEMORY_SCOPE_AGENT);
+  const int g = gridDim.x;
+  if ((int)blockIdx.x < nfin) {
+    // finalizer: wait for all partials, then reduce a channel slice
+    if (threadIdx.x == 0) {
+      for (long long it = 0; it < 50000000LL; ++it) {
+        if (__hip_atomic_load(&ws[0], __ATOMIC_RELAXED,
+                              __HIP_MEMORY_SCOPE_AGENT) == g)
+          break;
+        __builtin_amdgcn_s_sleep(2);
+      }
+    }
+    __syncthreads();
+    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
     const long long st = 2 * C;
-    for (int i = threadIdx.x; i < 2 * C; i += blockDim.x) {
+    const int i0 = blockIdx.x * 2 * C / nfin;
+    const int i1 = (blockIdx.x + 1) * 2 * C / nfin;
+    for (int i = i0 + threadIdx.x; i < i1; i += blockDim.x) {
       float sa[4] = {0, 0, 0, 0};
       int b = 0;
       for (; b + 4 <= g; b += 4)
 #pragma unroll
-        for (int u = 0; u < 4; ++u) sa[u] += partial[(long long)(b + u) * st + i];
+        for (int u = 0; u < 4; ++u)
+          sa[u] += partial[(long long)(b + u) * st + i];
       for (; b < g; ++b) sa[0] += partial[(long long)b * st + i];
       const float v = (sa[0] + sa[1]) + (sa[2] + sa[3]);
       sums[i] = v;
@@ -600,28 +614,27 @@ __global__ __launch_bounds__(256, 2) void bn_bwd_fused_kernel(
     __syncthreads();
     __threadfence();
     if (threadIdx.x == 0)
-      __hip_atomic_store(&ws[1], 1, __ATOMIC_RELEASE,
-                         __HIP_MEMORY_SCOPE_AGENT);
-  } else if (threadIdx.x == 0) {
-    // bounded spin: ~50M sleeps is minutes — unreachable unless the
-    // finalizer died; never hangs the box
-    for (long long it = 0; it < 50000000LL; ++it) {
-      if (__hip_atomic_load(&ws[1], __ATOMIC_ACQUIRE,
-                            __HIP_MEMORY_SCOPE_AGENT))
-        break;
-      __builtin_amdgcn_s_sleep(32);
-    }
+      __hip_atomic_fetch_add(&ws[1], 1, __ATOMIC_RELEASE,
+                             __HIP_MEMORY_SCOPE_AGENT);
   }
-  __syncthreads();
+  // everyone: wait for all slices, one acquire fence on exit
   if (threadIdx.x == 0) {
+    for (long long it = 0; it < 50000000LL; ++it) {
+      if (__hip_atomic_load(&ws[1], __ATOMIC_RELAXED,
+                            __HIP_MEMORY_SCOPE_AGENT) == nfin)
+        break;
+      __builtin_amdgcn_s_sleep(2);
+    }
     const int depart = __hip_atomic_fetch_add(&ws[2], 1, __ATOMIC_ACQ_REL,
                                               __HIP_MEMORY_SCOPE_AGENT);
-    if (depart == (int)gridDim.x - 1) {  // everyone passed the spin: reset
+    if (depart == g - 1) {  // everyone passed both waits: reset
       __hip_atomic_store(&ws[0], 0, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
       __hip_atomic_store(&ws[1], 0, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
       __hip_atomic_store(&ws[2], 0, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
     }
   }
+  __syncthreads();
+  __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
 
   // ---- phase 2: dx (+dres), training form (bn_bwd_dx) ----
   const float inv_m = 1.0f / (float)M;
@@ -632,13 +645,13 @@ __global__ __launch_bounds__(256, 2) void bn_bwd_fused_kernel(
     const int oct = (int)(i2 % c8);
     const int cc0 = oct * 8;
     const long long eoff = (i2 / c8) * C + cc0;
-    F8 g = load8(dy + eoff);
+    F8 g2 = load8(dy + eoff);
     if (RELU) {
       const unsigned mb = mask[i2];
 #pragma unroll
-      for (int k = 0; k < 8; ++k) g.v[k] = (mb >> k) & 1 ? g.v[k] : 0.0f;
+      for (int k = 0; k < 8; ++k) g2.v[k] = (mb >> k) & 1 ? g2.v[k] : 0.0f;
     }
-    if (ADD) store8(dres + eoff, g);
+    if (ADD) store8(dres + eoff, g2);
     F8 xv = load8(x + eoff);
     F8 o;
 #pragma unroll
@@ -647,7 +660,7 @@ __global__ __launch_bounds__(256, 2) void bn_bwd_fused_kernel(
       const float isv = invstd[c];
       const float xhat = (xv.v[k] - mean[c]) * isv;
       o.v[k] = gamma[c] * isv *
-               (g.v[k] - sums[c] * inv_m - xhat * sums[C + c] * inv_m);
+               (g2.v[k] - sums[c] * inv_m - xhat * sums[C + c] * inv_m);
     }
     store8(dx + eoff, o);
   }
@@ -828,13 +841,13 @@ extern "C" void launch_bn_bwd_fused(const void* dy, const unsigned char* mask,
                                     const float* invstd, const float* gamma,
                                     float* partial, float* sums, float* db_acc,
                                     float* dg_acc, void* dx, void* dres,
-                                    int* ws, int grid, long long M, int C,
-                                    bool relu, bool add, hipStream_t s) {
+                                    int* ws, int grid, int nfin, long long M,
+                                    int C, bool relu, bool add, hipStream_t s) {
 #define FCASE(R, A)                                                            \
   hipLaunchKernelGGL((bn_bwd_fused_kernel<R, A>), dim3(grid), dim3(256), 0, s, \
                      (const bf16*)dy, (R) ? mask : nullptr, (const bf16*)x,    \
                      mean, invstd, gamma, partial, sums, db_acc, dg_acc,       \
-                     (bf16*)dx, (bf16*)dres, ws, M, C)
+                     (bf16*)dx, (bf16*)dres, ws, nfin, M, C)
   if (relu && add) FCASE(true, true);
   else if (relu) FCASE(true, false);
   else if (add) FCASE(false, true);

@@ -31,8 +31,8 @@ extern "C" void launch_bn_apply(const void*, const void*, void*, unsigned char*,
 extern "C" void launch_bn_bwd_fused(const void*, const unsigned char*,
                                     const void*, const float*, const float*,
                                     const float*, float*, float*, float*,
-                                    float*, void*, void*, int*, int, long long,
-                                    int, bool, bool, hipStream_t);
+                                    float*, void*, void*, int*, int, int,
+                                    long long, int, bool, bool, hipStream_t);
 extern "C" void launch_bn_bwd_reduce(const void*, const unsigned char*,
                                      const void*, const float*, const float*,
                                      float*, int, long long, int, bool,
@@ -281,17 +281,26 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, c10::optional<torch::Tensor>
     if (wit == ws_pool.end())
       wit = ws_pool.emplace(dev, torch::zeros({4},
           x.options().dtype(torch::kInt32))).first;
-    // co-residency bound: every block must be resident for the spin to
-    // resolve (256 CUs, >=1 block/CU) — clamp whatever the grid-cap env
-    // says; partial has `grid` rows, the first fgrid are used
-    const int fgrid = grid > 256 ? 256 : grid;
+    // grid: big enough for the dx phase to saturate HBM (the reduce-grid
+    // cap of 192 starved phase 2 by 3-7x in the v1 cut), bounded at 512 =
+    // the co-residency guarantee of __launch_bounds__(256, 2) on 256 CUs
+    // (the rendezvous spins cannot deadlock)
+    const long long total8 = M * (C / 8);
+    int fgrid = (int)((total8 + 1023) / 1024);
+    if (fgrid < grid) fgrid = grid;
+    if (fgrid > 512) fgrid = 512;
+    // parallel finalize: slice 2C across up to 8 blocks, >=256 ch each
+    int nfin = (int)(2 * C) / 256;
+    if (nfin < 1) nfin = 1;
+    if (nfin > 8) nfin = 8;
+    auto fpartial = torch::empty({fgrid, 2 * C}, opts);
     launch_bn_bwd_fused(dyc.data_ptr(), mp, x.data_ptr(),
                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                        gamma.data_ptr<float>(), partial.data_ptr<float>(),
+                        gamma.data_ptr<float>(), fpartial.data_ptr<float>(),
                         sums.data_ptr<float>(), db_acc, dg_acc, dx.data_ptr(),
                         add ? dres.data_ptr() : nullptr,
-                        wit->second.data_ptr<int>(), fgrid, M, (int)C, relu,
-                        add, s);
+                        wit->second.data_ptr<int>(), fgrid, nfin, M, (int)C,
+                        relu, add, s);
     return {dx, sums[1], sums[0], dres};
   }
   launch_bn_bwd_reduce(dyc.data_ptr(), mp, x.data_ptr(),
