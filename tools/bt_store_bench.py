@@ -44,7 +44,9 @@ def worker():
         torch.cuda.synchronize()
         us = st.elapsed_time(en) * 1000 / 50
         ref = (a.float() @ b.float().t()).to(torch.bfloat16)
-        ok = bool((c.float() - ref.float()).abs().max().item() < 0.5)
+        # bf16 accumulation error grows ~sqrt(K) on random data
+        tol = 0.06 * (K ** 0.5)
+        ok = bool((c.float() - ref.float()).abs().max().item() < tol)
         out.append({"shape": [M, N, K], "us": round(us, 2), "ok": ok})
     print(json.dumps({"ldsb": os.environ.get("EDL_BT_STORE_LDS", "0"),
                       "shapes": out}))
