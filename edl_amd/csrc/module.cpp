@@ -60,6 +60,7 @@ extern "C" void launch_conv3x3_small(const void*, const void*, void*, int, int,
                                      hipStream_t);
 extern "C" void launch_pad_nhwc_cpad(const void*, void*, int, int, int, int,
                                      int, int, int, hipStream_t);
+extern "C" void launch_cast_bf16_zero(float*, void*, long long, hipStream_t);
 extern "C" void launch_repack_dgrad_w3(const void*, void*, int, int, int,
                                        hipStream_t);
 extern "C" void launch_maxpool3x3s2_fwd(const void*, void*, unsigned char*,
@@ -777,11 +778,15 @@ torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w3, int64_t stride) {
   const int splitk = conv3x3_pick_splitk((int)M, Cout, Cin);
   if (splitk > 1) {
     // CU-starved deep-K shape: fp32 split-K partials + cast
-    auto part = torch::zeros({M, Cout}, x.options().dtype(torch::kFloat32));
+    // pooled pre-zeroed fold target; cast_bf16_zero returns it clean
+    auto part = part_pool_get((int)M, Cout, x.options().dtype(torch::kFloat32));
     launch_conv3x3(xp.data_ptr(), w3.data_ptr(), nullptr, (int)M, Cout, Cin,
                    Hout * Wout, Wout, Hp, Wp, (int)stride,
                    part.data_ptr<float>(), splitk, nullptr, s);
-    return part.to(torch::kBFloat16);
+    auto y = torch::empty({M, Cout}, x.options());
+    launch_cast_bf16_zero(part.data_ptr<float>(), y.data_ptr(),
+                          (long long)M * Cout, s);
+    return y;
   }
   auto y = torch::empty({M, Cout}, x.options());
   launch_conv3x3(xp.data_ptr(), w3.data_ptr(), y.data_ptr(), (int)M, Cout, Cin,
@@ -815,11 +820,14 @@ std::vector<torch::Tensor> conv3x3_fwd_stats(torch::Tensor x, torch::Tensor w3,
   launch_pad_nhwc(x.data_ptr(), xp.data_ptr(), Nimg, H, W, Hp, Wp, Cin, s);
   const int splitk = conv3x3_pick_splitk((int)M, Cout, Cin);
   if (splitk > 1) {
-    auto part = torch::zeros({M, Cout}, x.options().dtype(torch::kFloat32));
+    auto part = part_pool_get((int)M, Cout, x.options().dtype(torch::kFloat32));
     launch_conv3x3(xp.data_ptr(), w3.data_ptr(), nullptr, (int)M, Cout, Cin,
                    Hout * Wout, Wout, Hp, Wp, (int)stride,
                    part.data_ptr<float>(), splitk, nullptr, s);
-    return {part.to(torch::kBFloat16), torch::Tensor()};
+    auto y = torch::empty({M, Cout}, x.options());
+    launch_cast_bf16_zero(part.data_ptr<float>(), y.data_ptr(),
+                          (long long)M * Cout, s);
+    return {y, torch::Tensor()};
   }
   auto y = torch::empty({M, Cout}, x.options());
   const int tiles_m = conv3x3_tiles_m((int)M, Cout);

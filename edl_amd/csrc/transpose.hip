@@ -166,3 +166,40 @@ extern "C" void launch_repack_dgrad_w3(const void* in, void* out, int Co,
                      (const __hip_bfloat16*)in, (__hip_bfloat16*)out, Co, Ci,
                      mode);
 }
+
+// fp32 -> bf16 cast that also returns the source to ZERO: consumes a
+// pooled split-K partial buffer (atomic-fold target, must start zeroed)
+// and hands it back clean — replaces a separate FillFunctor zeroing
+// launch per split-K conv (6-25 MB each). One float4 read + bf16x4
+// store + float4 zero store per thread per iteration.
+extern "C" __global__ void cast_bf16_zero_kernel(
+    float* __restrict__ src, __hip_bfloat16* __restrict__ dst,
+    const long long n4) {
+  typedef __hip_bfloat16 bf16;
+  long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  const float4 z = {0.0f, 0.0f, 0.0f, 0.0f};
+  for (; i < n4; i += stride) {
+    float4* ps = reinterpret_cast<float4*>(
+        __builtin_assume_aligned(src + i * 4, 16));
+    const float4 v = *ps;
+    ulong1 packed;
+    bf16* pv = reinterpret_cast<bf16*>(&packed);
+    pv[0] = __float2bfloat16(v.x);
+    pv[1] = __float2bfloat16(v.y);
+    pv[2] = __float2bfloat16(v.z);
+    pv[3] = __float2bfloat16(v.w);
+    *reinterpret_cast<ulong1*>(__builtin_assume_aligned(dst + i * 4, 8)) =
+        packed;
+    *ps = z;
+  }
+}
+
+extern "C" void launch_cast_bf16_zero(float* src, void* dst, long long n,
+                                      hipStream_t s) {
+  const long long n4 = n / 4;  // callers guarantee n % 4 == 0 (C % 64 == 0)
+  int grid = (int)((n4 + 255) / 256);
+  if (grid > 4096) grid = 4096;
+  hipLaunchKernelGGL(cast_bf16_zero_kernel, dim3(grid), dim3(256), 0, s, src,
+                     (__hip_bfloat16*)dst, n4);
+}

@@ -402,3 +402,27 @@ def test_repack_dgrad_w3_kernel(mode, co, ci):
     assert got.shape == ref.shape
     assert torch.equal(got.float(), ref.float()), \
         (got.float() - ref.float()).abs().max().item()
+
+
+@pytest.mark.gpu
+def test_conv3x3_splitk_pooled_partials_two_steps():
+    """Split-K conv fwd folds fp32 partials into a POOLED pre-zeroed buffer
+    and cast_bf16_zero returns it clean — two successive calls on the same
+    stage-3 shape must both match torch (a dirty pool corrupts call 2)."""
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    import torch.nn.functional as F
+
+    from edl_amd.ops.conv import Conv2dFast
+
+    torch.manual_seed(41)
+    # bs32 stage-3: M=6272 tiles ~ 52-98 -> split-K route (conv3x3_pick_splitk)
+    m = Conv2dFast(1024, 256, 3, padding=1, bias=False).cuda().to(torch.bfloat16)
+    for it in range(2):
+        x = torch.randn(32, 1024, 14, 14, device="cuda").to(torch.bfloat16)
+        x = x.contiguous(memory_format=torch.channels_last)
+        with torch.no_grad():
+            y = m(x)
+            ref = F.conv2d(x.float(), m.weight.float(), padding=1)
+        assert torch.allclose(y.float(), ref, atol=0.5, rtol=5e-2), \
+            (it, (y.float() - ref).abs().max().item())
