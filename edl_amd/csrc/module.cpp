@@ -1,6 +1,7 @@
 // Python bindings for the edl_amd CDNA4 kernel layer (edl_amd._C).
 // Compiled by hipcc directly (no hipify, no CUDA path) — see build_hip.py.
 #include <unordered_map>
+#include <unordered_set>
 #include <torch/extension.h>
 
 #include <c10/hip/HIPStream.h>
@@ -203,8 +204,13 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor gamma,
     launch_bn_stats(x.data_ptr(), partial.data_ptr<float>(), grid, M, (int)C,
                     s);
   }
+  // zero-after-read ONLY for pooled buffers (capped partials): transient
+  // uncapped partials are discarded, zeroing them is pure write waste
+  const int zero_src =
+      pre_part.has_value() &&
+      pooled_ptr_set().count((const void*)partial.data_ptr());
   launch_bn_finalize(partial.data_ptr<float>(), fin_grid,
-                     pre_part.has_value() ? 1 : 0, gamma.data_ptr<float>(),
+                     zero_src, gamma.data_ptr<float>(),
                      beta.data_ptr<float>(), mean.data_ptr<float>(),
                      invstd.data_ptr<float>(), scale.data_ptr<float>(),
                      shift.data_ptr<float>(),
@@ -337,6 +343,11 @@ torch::Tensor gemm_bt(torch::Tensor a, torch::Tensor b) {
 }
 
 
+static std::unordered_set<const void*>& pooled_ptr_set() {
+  static std::unordered_set<const void*> ptrs;
+  return ptrs;
+}
+
 static torch::Tensor part_pool_get(int rows, int64_t cols,
                                    torch::TensorOptions opts) {
   // Capped BN-stats partial buffers are accumulated with atomics, so they
@@ -351,6 +362,7 @@ static torch::Tensor part_pool_get(int rows, int64_t cols,
   if (it != pool.end()) return it->second;
   auto t = torch::zeros({rows, cols}, opts);
   pool.emplace(key, t);
+  pooled_ptr_set().insert(t.data_ptr());
   return t;
 }
 
