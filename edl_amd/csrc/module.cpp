@@ -155,6 +155,29 @@ torch::Tensor kd_ce_backward(torch::Tensor s, torch::Tensor t, double gout_over_
   return ds;
 }
 
+static std::unordered_set<const void*>& pooled_ptr_set() {
+  static std::unordered_set<const void*> ptrs;
+  return ptrs;
+}
+
+static torch::Tensor part_pool_get(int rows, int64_t cols,
+                                   torch::TensorOptions opts) {
+  // Capped BN-stats partial buffers are accumulated with atomics, so they
+  // must start zeroed. Pool them per (device, shape): zeroed ONCE here,
+  // and bn_finalize(zero_src=1) stores zeros back after reading, so reuse
+  // needs no per-step fill launch. If a producer's partials are ever
+  // dropped unconsumed, the Python side zeroes them (bnrelu.forward).
+  static std::unordered_map<int64_t, torch::Tensor> pool;
+  const int64_t key = ((int64_t)opts.device().index() << 48) |
+                      ((int64_t)rows << 28) | cols;
+  auto it = pool.find(key);
+  if (it != pool.end()) return it->second;
+  auto t = torch::zeros({rows, cols}, opts);
+  pool.emplace(key, t);
+  pooled_ptr_set().insert(t.data_ptr());
+  return t;
+}
+
 // x: NHWC bf16 viewed as [M, C] contiguous (channels_last 4-D collapses to
 // this). gamma/beta/running stats: fp32 [C].
 void check_bn_inputs(const torch::Tensor& x, int64_t C) {
@@ -342,29 +365,6 @@ torch::Tensor gemm_bt(torch::Tensor a, torch::Tensor b) {
   return c;
 }
 
-
-static std::unordered_set<const void*>& pooled_ptr_set() {
-  static std::unordered_set<const void*> ptrs;
-  return ptrs;
-}
-
-static torch::Tensor part_pool_get(int rows, int64_t cols,
-                                   torch::TensorOptions opts) {
-  // Capped BN-stats partial buffers are accumulated with atomics, so they
-  // must start zeroed. Pool them per (device, shape): zeroed ONCE here,
-  // and bn_finalize(zero_src=1) stores zeros back after reading, so reuse
-  // needs no per-step fill launch. If a producer's partials are ever
-  // dropped unconsumed, the Python side zeroes them (bnrelu.forward).
-  static std::unordered_map<int64_t, torch::Tensor> pool;
-  const int64_t key = ((int64_t)opts.device().index() << 48) |
-                      ((int64_t)rows << 28) | cols;
-  auto it = pool.find(key);
-  if (it != pool.end()) return it->second;
-  auto t = torch::zeros({rows, cols}, opts);
-  pool.emplace(key, t);
-  pooled_ptr_set().insert(t.data_ptr());
-  return t;
-}
 
 std::vector<torch::Tensor> gemm_bt_stats(torch::Tensor a, torch::Tensor b) {
   // like gemm_bt, additionally returning BN stats partials [tiles_m, 2N]
