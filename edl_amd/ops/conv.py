@@ -347,6 +347,49 @@ class Conv2dFast(nn.Conv2d):
             d[key] = fn()
         return d[key]
 
+    def prefill_derived(self):
+        """Rebuild this conv's per-weight-epoch derived tensors (wt_t /
+        dgrad repacks) for the CURRENT epoch. The engine calls this on a
+        side HIP stream at step start, so the repack kernels overlap the
+        forward pass instead of sitting serially in the step (~49
+        transpose/repack launches, ~240 us/step in the r2c27 trace);
+        forward then finds them via _cached and launches nothing inline.
+        Only keys a previous forward actually used are rebuilt; anything
+        missed falls back to the inline path (correctness never depends
+        on prefill)."""
+        if not self.weight.requires_grad or not self.weight.is_cuda:
+            return
+        if not available():
+            return
+        cache = getattr(self, "_w_cache", None)
+        if cache is None or not cache[1]:
+            return
+        old_keys = list(cache[1].keys())
+        co, ci = self.out_channels, self.in_channels
+        mb = getattr(self.weight, "_edl_bf16", None)
+        cl = getattr(self.weight, "_edl_phys_shape", None) is not None
+        builders = {}
+        if mb is not None and self.kernel_size == (1, 1):
+            builders["wt_t"] = lambda: ext().transpose_pad(mb.view(co, ci))
+        if mb is not None and cl and self.kernel_size == (3, 3):
+            w3v = mb.permute(0, 2, 3, 1).reshape(co, 9 * ci)
+            builders["w3rot"] = lambda: ext().repack_dgrad_w3(w3v, ci, 0)
+            builders["w3s2d"] = lambda: ext().repack_dgrad_w3(w3v, ci, 1)
+        if (self.kernel_size == (3, 3) and self.in_channels <= 64
+                and self.out_channels <= 64
+                and (self.in_channels % 64 != 0
+                     or self.out_channels % 64 != 0)):
+            cpt = _small_cpt(ci)
+            builders["w3s"] = lambda: _repack_w3_small(
+                self.weight.detach(), cpt)
+            builders["w3srot"] = lambda: _repack_w3_small(
+                self.weight.detach().permute(1, 0, 2, 3).flip(2, 3),
+                _small_cpt(co))
+        for k in old_keys:
+            b = builders.get(k)
+            if b is not None:
+                self._cached(k, b)
+
     def pop_bn_part(self):
         """BN stats partials of the LAST forward (or None) — consumed by
         the following BatchNorm to skip its stats kernel."""

@@ -146,6 +146,10 @@ class TrainerEngine:
         self.global_step = 0
         self._graph = None
         self._static = {}
+        # side stream for the per-epoch weight-repack prefill (overlaps
+        # forward; EDL_PREFILL_DERIVED=0 reverts to inline repacks)
+        self._prefill_stream = None
+        self._prefill_mods = None
 
     # ---- setup ----
     def setup(self, env=None):
@@ -153,6 +157,9 @@ class TrainerEngine:
         torch.manual_seed(42 + self.env.global_rank)
         self.model = build_model(self.model_name, num_classes=self.num_classes)
         self.model.to(self.device)
+        if (self.device.type == "cuda"
+                and os.environ.get("EDL_PREFILL_DERIVED", "1") == "1"):
+            self._prefill_stream = torch.cuda.Stream()
         if self.channels_last and self.device.type == "cuda":
             self.model.to(memory_format=torch.channels_last)
         if self.use_hip_ops and self.device.type == "cuda":
@@ -351,12 +358,18 @@ class TrainerEngine:
 
     def train_step(self, images, labels, teacher_logits=None):
         self.reducer.zero_grad()
+        self._prefill_derived_async()
         with _mark("edl.forward"):
             with torch.autocast(device_type=self.device.type, dtype=self.dtype,
                                 enabled=self.dtype != torch.float32):
                 logits = self.model(images)
             loss = self._loss(logits, labels, teacher_logits)
         with _mark("edl.backward"):
+            if self._prefill_stream is not None:
+                # dgrad kernels read the prefilled repacks (see
+                # _prefill_derived_async) — join the side stream here,
+                # after forward has run in parallel with the repacks
+                torch.cuda.current_stream().wait_stream(self._prefill_stream)
             if self.scaler is not None:
                 (loss * self.scaler.value).backward()
             else:
@@ -401,6 +414,27 @@ class TrainerEngine:
             self._refresh_bf16_mirrors()
         self.global_step += 1
         return loss
+
+    def _prefill_derived_async(self):
+        """Launch the per-epoch weight repacks (wt_t, dgrad w3 repacks)
+        on a side stream at step start: they only depend on the bf16
+        mirrors (written at the END of the previous step), so they run in
+        parallel with forward; backward joins the stream before its first
+        dgrad. Capture-safe: fork/join via wait_stream edges."""
+        if self._prefill_stream is None:
+            return
+        mods = self._prefill_mods
+        if mods is None:
+            from ..ops.conv import Conv2dFast
+
+            mods = [m for m in self.model.modules()
+                    if isinstance(m, Conv2dFast)]
+            self._prefill_mods = mods
+        cur = torch.cuda.current_stream()
+        self._prefill_stream.wait_stream(cur)
+        with torch.cuda.stream(self._prefill_stream):
+            for m in mods:
+                m.prefill_derived()
 
     # ---- hipGraph capture ----
     def maybe_capture(self, images, labels):

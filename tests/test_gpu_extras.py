@@ -96,3 +96,35 @@ def test_dgc_two_rank_cuda(tmp_path):
     oks = [json.loads(m) for m in
            re.findall(r'\{"dgc_cuda".*?\}', r.stdout)]
     assert oks and all(v["ok"] for v in oks)
+
+
+def test_prefill_derived_matches_inline():
+    """The side-stream weight-repack prefill (EDL_PREFILL_DERIVED) must
+    not change the training trajectory: same seed, same losses/params as
+    the inline-repack path over several steps."""
+    import os
+
+    from edl_amd.data.synthetic import SyntheticImageNet
+    from edl_amd.train.engine import TrainerEngine
+
+    def run(flag):
+        os.environ["EDL_PREFILL_DERIVED"] = flag
+        try:
+            torch.manual_seed(7)
+            eng = TrainerEngine(model="resnet18_vd", per_device_batch=8,
+                                dtype="bf16", checkpoint_dir=None).setup()
+            loader = SyntheticImageNet(8, eng.device, channels_last=True,
+                                       seed=9)
+            x, y = loader.next()
+            losses = [float(eng.train_step(x, y).item()) for _ in range(4)]
+            params = [p.detach().float().clone()
+                      for p in eng.model.parameters()]
+            return losses, params
+        finally:
+            os.environ.pop("EDL_PREFILL_DERIVED", None)
+
+    l0, p0 = run("0")
+    l1, p1 = run("1")
+    assert l0 == pytest.approx(l1, rel=1e-3), (l0, l1)
+    for a, b in zip(p0, p1):
+        assert torch.allclose(a, b, atol=1e-3, rtol=1e-3)
