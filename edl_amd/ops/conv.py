@@ -379,9 +379,9 @@ class Conv2dFast(nn.Conv2d):
                 and self.out_channels <= 64
                 and (self.in_channels % 64 != 0
                      or self.out_channels % 64 != 0)):
-            cpt = _small_cpt(ci)
-            builders["w3s"] = lambda: _repack_w3_small(
-                self.weight.detach(), cpt)
+            # w3srot only: it is read by BACKWARD (after the engine joins
+            # the side stream). w3s is read by FORWARD, which does NOT
+            # wait on the prefill stream — prefilling it races.
             builders["w3srot"] = lambda: _repack_w3_small(
                 self.weight.detach().permute(1, 0, 2, 3).flip(2, 3),
                 _small_cpt(co))

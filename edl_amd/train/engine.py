@@ -146,8 +146,10 @@ class TrainerEngine:
         self.global_step = 0
         self._graph = None
         self._static = {}
-        # side stream for the per-epoch weight-repack prefill (overlaps
-        # forward; EDL_PREFILL_DERIVED=0 reverts to inline repacks)
+        # side stream for the per-epoch weight-repack prefill
+        # (EDL_PREFILL_DERIVED=1 opt-in: measured -5% under hipGraph
+        # capture on MI355X — the two-stream graph replays with less
+        # launch pipelining than the single-stream one; see NOTES)
         self._prefill_stream = None
         self._prefill_mods = None
 
@@ -158,7 +160,7 @@ class TrainerEngine:
         self.model = build_model(self.model_name, num_classes=self.num_classes)
         self.model.to(self.device)
         if (self.device.type == "cuda"
-                and os.environ.get("EDL_PREFILL_DERIVED", "1") == "1"):
+                and os.environ.get("EDL_PREFILL_DERIVED", "0") == "1"):
             self._prefill_stream = torch.cuda.Stream()
         if self.channels_last and self.device.type == "cuda":
             self.model.to(memory_format=torch.channels_last)
