@@ -99,32 +99,52 @@ def test_dgc_two_rank_cuda(tmp_path):
 
 
 def test_prefill_derived_matches_inline():
-    """The side-stream weight-repack prefill (EDL_PREFILL_DERIVED) must
-    not change the training trajectory: same seed, same losses/params as
-    the inline-repack path over several steps."""
+    """The side-stream weight-repack prefill (EDL_PREFILL_DERIVED=1,
+    opt-in) must produce byte-identical repacks to the inline builders.
+    (Trajectory comparison across runs is NOT valid here: the step uses
+    fp32 atomics — split-K folds, capped stats partials — so two runs
+    diverge in the low bits regardless of prefill.)"""
     import os
 
     from edl_amd.data.synthetic import SyntheticImageNet
+    from edl_amd.ops import ext
+    from edl_amd.ops.conv import Conv2dFast
     from edl_amd.train.engine import TrainerEngine
 
-    def run(flag):
-        os.environ["EDL_PREFILL_DERIVED"] = flag
-        try:
-            torch.manual_seed(7)
-            eng = TrainerEngine(model="resnet18_vd", per_device_batch=8,
-                                dtype="bf16", checkpoint_dir=None).setup()
-            loader = SyntheticImageNet(8, eng.device, channels_last=True,
-                                       seed=9)
-            x, y = loader.next()
-            losses = [float(eng.train_step(x, y).item()) for _ in range(4)]
-            params = [p.detach().float().clone()
-                      for p in eng.model.parameters()]
-            return losses, params
-        finally:
-            os.environ.pop("EDL_PREFILL_DERIVED", None)
+    os.environ["EDL_PREFILL_DERIVED"] = "1"
+    try:
+        torch.manual_seed(7)
+        eng = TrainerEngine(model="resnet18_vd", per_device_batch=8,
+                            dtype="bf16", checkpoint_dir=None).setup()
+        assert eng._prefill_stream is not None
+        loader = SyntheticImageNet(8, eng.device, channels_last=True, seed=9)
+        x, y = loader.next()
+        losses = [float(eng.train_step(x, y).item()) for _ in range(3)]
+        assert all(l == l for l in losses), losses  # finite
+        torch.cuda.synchronize()
 
-    l0, p0 = run("0")
-    l1, p1 = run("1")
-    assert l0 == pytest.approx(l1, rel=1e-3), (l0, l1)
-    for a, b in zip(p0, p1):
-        assert torch.allclose(a, b, atol=1e-3, rtol=1e-3)
+        checked = 0
+        for m in eng.model.modules():
+            if not isinstance(m, Conv2dFast):
+                continue
+            cache = getattr(m, "_w_cache", None)
+            mb = getattr(m.weight, "_edl_bf16", None)
+            if cache is None or mb is None:
+                continue
+            d = cache[1]
+            co, ci = m.out_channels, m.in_channels
+            cl = getattr(m.weight, "_edl_phys_shape", None) is not None
+            if "wt_t" in d and m.kernel_size == (1, 1):
+                assert torch.equal(d["wt_t"],
+                                   ext().transpose_pad(mb.view(co, ci)))
+                checked += 1
+            if cl and m.kernel_size == (3, 3):
+                w3v = mb.permute(0, 2, 3, 1).reshape(co, 9 * ci)
+                for key, mode in (("w3rot", 0), ("w3s2d", 1)):
+                    if key in d:
+                        assert torch.equal(
+                            d[key], ext().repack_dgrad_w3(w3v, ci, mode)), key
+                        checked += 1
+        assert checked > 0, "prefill never produced a repack to verify"
+    finally:
+        os.environ.pop("EDL_PREFILL_DERIVED", None)
