@@ -119,31 +119,44 @@ def test_prefill_derived_matches_inline():
         assert eng._prefill_stream is not None
         loader = SyntheticImageNet(8, eng.device, channels_last=True, seed=9)
         x, y = loader.next()
-        losses = [float(eng.train_step(x, y).item()) for _ in range(3)]
+        losses = [float(eng.train_step(x, y).item()) for _ in range(2)]
         assert all(l == l for l in losses), losses  # finite
+        torch.cuda.synchronize()
+
+        # snapshot the mirrors BEFORE the next step: the repacks that
+        # step's backward consumes must equal repacks of THESE values
+        # (comparing against post-step mirrors is wrong — the optimizer
+        # refreshes them at step end)
+        snaps = {}
+        for m in eng.model.modules():
+            if isinstance(m, Conv2dFast):
+                mb = getattr(m.weight, "_edl_bf16", None)
+                if mb is not None:
+                    snaps[id(m)] = mb.detach().clone()
+        eng.train_step(x, y)
         torch.cuda.synchronize()
 
         checked = 0
         for m in eng.model.modules():
-            if not isinstance(m, Conv2dFast):
+            if not isinstance(m, Conv2dFast) or id(m) not in snaps:
                 continue
             cache = getattr(m, "_w_cache", None)
-            mb = getattr(m.weight, "_edl_bf16", None)
-            if cache is None or mb is None:
+            if cache is None:
                 continue
             d = cache[1]
+            snap = snaps[id(m)]
             co, ci = m.out_channels, m.in_channels
             cl = getattr(m.weight, "_edl_phys_shape", None) is not None
             if "wt_t" in d and m.kernel_size == (1, 1):
-                assert torch.equal(d["wt_t"],
-                                   ext().transpose_pad(mb.view(co, ci)))
+                ref = ext().transpose_pad(snap.reshape(co, ci))
+                assert torch.equal(d["wt_t"], ref), "wt_t raced"
                 checked += 1
             if cl and m.kernel_size == (3, 3):
-                w3v = mb.permute(0, 2, 3, 1).reshape(co, 9 * ci)
+                w3v = snap.permute(0, 2, 3, 1).reshape(co, 9 * ci).contiguous()
                 for key, mode in (("w3rot", 0), ("w3s2d", 1)):
                     if key in d:
-                        assert torch.equal(
-                            d[key], ext().repack_dgrad_w3(w3v, ci, mode)), key
+                        ref = ext().repack_dgrad_w3(w3v, ci, mode)
+                        assert torch.equal(d[key], ref), key + " raced"
                         checked += 1
         assert checked > 0, "prefill never produced a repack to verify"
     finally:
