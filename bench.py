@@ -21,8 +21,8 @@ def parse_args():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=30)
-    # MIOpen's find phase on the stem convs settles around step 8 (naive
-    # kernels until then — profiles/r01_step5); keep warmup clear of it
+    # r2: the in-repo stem kernels removed MIOpen's find phase from the
+    # step (warmup 5 measures the same as 15); 15 stays as a safe default
     p.add_argument("--warmup", type=int, default=15)
     p.add_argument("--batch_size", type=int, default=32, help="per-GPU batch")
     p.add_argument("--model", default="resnet50_vd")
