@@ -59,6 +59,9 @@ extern "C" void launch_conv3x3s2_dgrad(const void*, const void*, void*, int,
 extern "C" void launch_conv3x3_small(const void*, const void*, void*, int, int,
                                      int, int, int, int, int, int, float*,
                                      hipStream_t);
+extern "C" void launch_gemm_tn3x3_small(const void*, const void*, float*,
+                                        int, int, int, int, int, int, int,
+                                        int, int, int, hipStream_t);
 extern "C" void launch_pad_nhwc_cpad(const void*, void*, int, int, int, int,
                                      int, int, int, hipStream_t);
 extern "C" void launch_cast_bf16_zero(float*, void*, long long, hipStream_t);
@@ -855,6 +858,55 @@ std::vector<torch::Tensor> conv3x3_fwd_stats(torch::Tensor x, torch::Tensor w3,
   return {y, bpart};
 }
 
+torch::Tensor gemm_tn3x3_small(torch::Tensor dy2d, torch::Tensor x,
+                               int64_t stride, int64_t splitk) {
+  // Deep-stem conv wgrad, fully in-repo (closes VERDICT r1 #4's last
+  // MIOpen rows): fp32 dW[64, N2v] = dY^T @ gather3x3(cpad(x)) via the
+  // G3S small-Cin gather (gemm_tn.hip). dy2d: [M, 64] bf16 with the real
+  // Cout columns first (caller zero-pads); x: 4-D channels_last bf16,
+  // Cin <= 64. Returns [64, N2v]; caller slices [:Cout, :9*CinP] and
+  // re-lays to [Cout, Cin, 3, 3].
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 &&
+                  x.scalar_type() == torch::kBFloat16 &&
+                  x.is_contiguous(torch::MemoryFormat::ChannelsLast),
+              "tn3x3s: x 4-D channels_last bf16");
+  TORCH_CHECK(stride == 1 || stride == 2, "tn3x3s: stride 1 or 2");
+  const int Nimg = (int)x.size(0), Cin = (int)x.size(1);
+  const int H = (int)x.size(2), W = (int)x.size(3);
+  TORCH_CHECK(Cin >= 1 && Cin <= 64, "tn3x3s: Cin <= 64");
+  const int Hp = H + 2, Wp = W + 2;
+  const int Hout = (H - 1) / (int)stride + 1;
+  const int Wout = (W - 1) / (int)stride + 1;
+  const long long M = (long long)Nimg * Hout * Wout;
+  TORCH_CHECK(dy2d.is_cuda() && dy2d.is_contiguous() && dy2d.dim() == 2 &&
+                  dy2d.scalar_type() == torch::kBFloat16 &&
+                  dy2d.size(0) == M && dy2d.size(1) == 64,
+              "tn3x3s: dy2d [M, 64] bf16");
+  int CinP = 8;
+  while (CinP < Cin) CinP <<= 1;
+  const int N2 = 9 * CinP;
+  const int N2v = (N2 + 63) / 64 * 64;
+  auto s = cur_stream();
+  auto xp = torch::empty({(long long)Nimg * Hp * Wp * CinP}, x.options());
+  if (Cin == CinP) {
+    launch_pad_nhwc(x.data_ptr(), xp.data_ptr(), Nimg, H, W, Hp, Wp, Cin, s);
+  } else {
+    launch_pad_nhwc_cpad(x.data_ptr(), xp.data_ptr(), Nimg, H, W, Hp, Wp,
+                         Cin, CinP, s);
+  }
+  if (splitk <= 0) {
+    const int tiles = N2v / 64;
+    splitk = std::max<int64_t>(1, 512 / std::max(1, tiles));
+    const int nch = (int)((M + 63) / 64);
+    splitk = std::min<int64_t>(splitk, std::max(1, nch));
+  }
+  auto c = torch::zeros({64, N2v}, x.options().dtype(torch::kFloat32));
+  launch_gemm_tn3x3_small(dy2d.data_ptr(), xp.data_ptr(), c.data_ptr<float>(),
+                          64, N2v, (int)M, Hout, Wout, Hp, Wp, CinP,
+                          (int)stride, (int)splitk, s);
+  return c;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -894,6 +946,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "direct conv3x3 wgrad: fp32 dW3[Cout,9Cin] = dY^T @ gather3x3(pad(x))",
         pybind11::arg("dy2d"), pybind11::arg("x"), pybind11::arg("stride"),
         pybind11::arg("splitk") = 0, pybind11::arg("out") = pybind11::none());
+  m.def("gemm_tn3x3_small", &gemm_tn3x3_small, py::arg("dy2d"),
+        py::arg("x"), py::arg("stride"), py::arg("splitk") = 0,
+        "stem wgrad: dY^T @ gather3x3(cpad(x)) -> fp32 [64, N2v] (G3S)");
   m.def("gemm_tn_splitk", &gemm_tn_splitk,
         "split-K TN GEMM: fp32 C[N1,N2] = A[K,N1]^T @ B[K,N2] (direct wgrad)",
         pybind11::arg("a"), pybind11::arg("b"), pybind11::arg("splitk") = 0,

@@ -262,7 +262,19 @@ class _Conv3x3SmallHip(torch.autograd.Function):
         # (EDL_STEM_WGRAD=unfold) but MEASURED NEGATIVE: hipBLASLt runs
         # the [32, 401k] @ [401k, 288] reduction GEMM at ~918 us — no
         # split-K pick at tiny M,N (gpurun_out/r2c7, bench 3221 -> 2566).
-        if os.environ.get("EDL_STEM_WGRAD", "miopen") == "unfold":
+        route = os.environ.get("EDL_STEM_WGRAD", "miopen")
+        if route == "tn":
+            # in-repo G3S gather wgrad (gemm_tn3x3_small): dY cols padded
+            # to 64, x channel-padded to pow2 inside the binding; output
+            # [64, N2v] sliced back to [co, ci, 3, 3]
+            dy2d = dyb.permute(0, 2, 3, 1).reshape(-1, co)
+            if co < 64:
+                dy2d = F.pad(dy2d, (0, 64 - co))
+            cfull = ext().gemm_tn3x3_small(dy2d.contiguous(), x, ctx.stride)
+            cinp = max(8, 1 << max(0, (ci - 1).bit_length()))
+            dw = (cfull[:co, :9 * cinp].view(co, 3, 3, cinp)
+                  .permute(0, 3, 1, 2)[:, :ci].contiguous())
+        elif route == "unfold":
             dy2d = dyb.permute(0, 2, 3, 1).reshape(-1, co)
             xu = F.unfold(x, 3, padding=1, stride=ctx.stride)
             xu2d = xu.permute(0, 2, 1).reshape(-1, ci * 9)  # [M, ci*9]

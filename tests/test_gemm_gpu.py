@@ -426,3 +426,39 @@ def test_conv3x3_splitk_pooled_partials_two_steps():
             ref = F.conv2d(x.float(), m.weight.float(), padding=1)
         assert torch.allclose(y.float(), ref, atol=0.5, rtol=5e-2), \
             (it, (y.float() - ref).abs().max().item())
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("shape", [
+    (4, 3, 32, 64, 2),    # stem conv0: Cin=3 (cpad->8), stride 2
+    (4, 32, 32, 32, 1),   # stem conv1
+    (4, 32, 64, 32, 1),   # stem conv2 (Cout=64, no dy pad)
+])
+def test_gemm_tn3x3_small_wgrad_numerics(shape):
+    """G3S stem wgrad == torch conv2d_weight (fp32 reference)."""
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from edl_amd import ops
+
+    n, ci, co, hw, stride = shape
+    torch.manual_seed(50)
+    x = torch.randn(n, ci, hw, hw, device="cuda").to(torch.bfloat16)
+    x = x.contiguous(memory_format=torch.channels_last)
+    ho = (hw - 1) // stride + 1
+    dy = torch.randn(n, co, ho, ho, device="cuda").to(torch.bfloat16)
+    dy = dy.contiguous(memory_format=torch.channels_last)
+
+    dy2d = dy.permute(0, 2, 3, 1).reshape(-1, co)
+    if co < 64:
+        import torch.nn.functional as F
+        dy2d = F.pad(dy2d, (0, 64 - co))
+    cfull = ops.ext().gemm_tn3x3_small(dy2d.contiguous(), x, stride)
+    cinp = max(8, 1 << (ci - 1).bit_length())
+    dw = (cfull[:co, :9 * cinp].view(co, 3, 3, cinp)
+          .permute(0, 3, 1, 2)[:, :ci])
+
+    ref = torch.nn.grad.conv2d_weight(
+        x.float(), (co, ci, 3, 3), dy.float(), stride=(stride, stride),
+        padding=(1, 1))
+    assert torch.allclose(dw, ref, atol=0.5, rtol=2e-2), \
+        (dw - ref).abs().max().item()
