@@ -148,9 +148,12 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_kernel(
     const int hw = Ho * Wo;
     const int n = m / hw, rem = m % hw;
     const int oy = rem / Wo, ox = rem % Wo;
-    const int n2 = t2 + ncol;
-    int tap = n2 >> cin_l2;
-    if (tap > 8) tap = 8;
+    int n2 = t2 + ncol;
+    // virtual-padding cols clamp to the LAST real slot (slot-aligned:
+    // 9*CinP is a multiple of 8) — in-bounds, 16-B aligned, discarded
+    const int n2max = 9 * Cin - 8;
+    if (n2 > n2max) n2 = n2max;
+    const int tap = n2 >> cin_l2;
     const int cc = n2 - (tap << cin_l2);
     return B + ((long long)(n * Hp + oy * stride + tap / 3) * Wp +
                 ox * stride + tap % 3) *
