@@ -462,3 +462,31 @@ def test_gemm_tn3x3_small_wgrad_numerics(shape):
         padding=(1, 1))
     assert torch.allclose(dw, ref, atol=0.5, rtol=2e-2), \
         (dw - ref).abs().max().item()
+
+
+@pytest.mark.gpu
+def test_stem_wgrad_tn_route_matches_default(monkeypatch):
+    """EDL_STEM_WGRAD=tn through the module backward (pad + slice glue)
+    must match the default route's weight grad."""
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    import copy
+
+    from edl_amd.ops.conv import Conv2dFast
+
+    torch.manual_seed(51)
+    m1 = Conv2dFast(32, 32, 3, padding=1, bias=False).cuda()
+    m2 = copy.deepcopy(m1)
+    x = torch.randn(4, 32, 32, 32, device="cuda").to(torch.bfloat16)
+    x = x.contiguous(memory_format=torch.channels_last).requires_grad_(True)
+
+    monkeypatch.setenv("EDL_STEM_WGRAD", "tn")
+    y1 = m1(x)
+    y1.float().square().mean().backward()
+    monkeypatch.delenv("EDL_STEM_WGRAD")
+    x2 = x.detach().clone().requires_grad_(True)
+    y2 = m2(x2)
+    y2.float().square().mean().backward()
+    assert torch.allclose(m1.weight.grad, m2.weight.grad, atol=1e-3,
+                          rtol=1e-2), \
+        (m1.weight.grad - m2.weight.grad).abs().max().item()
