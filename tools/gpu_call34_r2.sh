@@ -34,7 +34,7 @@ while time.monotonic() < t_end:
               flush=True)
         window = now; wsteps = 0
     if steps % 5000 == 0:
-        eng.save_checkpoint(epoch=0, step=steps)
+        eng.save_checkpoint(epoch=0, extra={"step": steps})
 torch.cuda.synchronize()
 print(f"SOAK OK: {steps} steps, peak mem "
       f"{torch.cuda.max_memory_allocated()/2**30:.2f} GiB, "
