@@ -245,12 +245,16 @@ __global__ void bn_apply_kernel(
     const float* __restrict__ scale,
     const float* __restrict__ shift, const long long M, const int C) {
   const int c8 = C >> 3;
+  // every model width is pow2: i%/c8 as mask/shift (the generic 64-bit
+  // divmod lowers to a ~100-instruction sequence per octet)
+  const bool p2 = (c8 & (c8 - 1)) == 0;
+  const int c8l = 31 - __clz((unsigned)c8);
   const long long total = M * c8;
   long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
   const long long stride = (long long)gridDim.x * blockDim.x;
   for (; i < total; i += stride) {
-    const int oct = (int)(i % c8);
-    const long long eoff = (i / c8) * C + oct * 8;
+    const int oct = p2 ? (int)(i & (c8 - 1)) : (int)(i % c8);
+    const long long eoff = (p2 ? (i >> c8l) : (i / c8)) * C + oct * 8;
     F8 v = load8(x + eoff);
     const int c0 = oct * 8;
 #pragma unroll
@@ -447,14 +451,16 @@ __global__ void bn_bwd_dx_kernel(
     const float* __restrict__ sums, bf16* __restrict__ dx,
     bf16* __restrict__ dres, const long long M, const int C) {
   const int c8 = C >> 3;
+  const bool p2 = (c8 & (c8 - 1)) == 0;  // pow2 fast path (see bn_apply)
+  const int c8l = 31 - __clz((unsigned)c8);
   const long long total = M * c8;
   const float inv_m = 1.0f / (float)M;
   long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
   const long long stride = (long long)gridDim.x * blockDim.x;
   for (; i < total; i += stride) {
-    const int oct = (int)(i % c8);
+    const int oct = p2 ? (int)(i & (c8 - 1)) : (int)(i % c8);
     const int c0 = oct * 8;
-    const long long eoff = (i / c8) * C + c0;
+    const long long eoff = (p2 ? (i >> c8l) : (i / c8)) * C + c0;
     F8 g = load8(dy + eoff);
     if (RELU) {
       const unsigned mb = mask[i];
