@@ -7,7 +7,7 @@ Configs (reference README.md:84-85):
 
 Single-GPU form (gpurun): both processes share cuda:0.
 
-    python tools/bench_distill.py --steps 20 --batch_size 32
+    python tools/bench_distill.py --steps 200 --batch_size 32
 Service form (BASELINE config 4, 4 teacher + 4 student GPUs):
     python tools/bench_distill.py --teacher_gpus 0,1,2,3 --student_gpus 4,5,6,7
 (teachers run as one process per GPU; students as one torchrun rank per
@@ -28,8 +28,12 @@ sys.path.insert(0, REPO)
 
 def main():
     ap = argparse.ArgumentParser()
-    ap.add_argument("--steps", type=int, default=20)
-    ap.add_argument("--warmup", type=int, default=5)
+    # the threaded reader/teacher pipeline ramps for O(100) steps (queues
+    # filling); short windows under-measure ~2x — window sweep in
+    # profiles/r2_distill_window_sweep.json (120/250/500 steps =
+    # 1156/1382/1445 img/s on one MI355X)
+    ap.add_argument("--steps", type=int, default=200)
+    ap.add_argument("--warmup", type=int, default=30)
     ap.add_argument("--batch_size", type=int, default=32)
     ap.add_argument("--teacher_batch_size", type=int, default=16)
     ap.add_argument("--require_num", type=int, default=1)
