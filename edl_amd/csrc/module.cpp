@@ -276,7 +276,6 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, c10::optional<torch::Tensor>
   const long long M = x.numel() / C;
   auto opts = gamma.options().dtype(torch::kFloat32);
   const int grid = bn_bwd_grid(M, (int)C);
-  auto partial = torch::empty({grid, 2 * C}, opts);
   auto sums = torch::empty({2, C}, opts);
   auto dx = torch::empty_like(x);
   auto dres = add ? torch::empty_like(x) : torch::Tensor();
@@ -336,6 +335,7 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, c10::optional<torch::Tensor>
                         relu, add, s);
     return {dx, sums[1], sums[0], dres};
   }
+  auto partial = torch::empty({grid, 2 * C}, opts);
   launch_bn_bwd_reduce(dyc.data_ptr(), mp, x.data_ptr(),
                        mean.data_ptr<float>(), invstd.data_ptr<float>(),
                        partial.data_ptr<float>(), grid, M, (int)C, relu, s);
