@@ -101,12 +101,7 @@ class BNReLU2d(nn.Module):
             return False
         if not (x.is_cuda and x.dtype == torch.bfloat16 and available()):
             return False
-        if self.num_features % 8 != 0:
-            return False
-        # the 2048 cap protects the TRAINING stats/reduce kernels' LDS
-        # (lsum[2*2048]); the eval apply kernel is elementwise and wide
-        # models (ResNeXt101_32x16d stage-4: C=4096) can use it
-        if self.training and self.num_features > 2048:
+        if self.num_features % 8 != 0 or self.num_features > 2048:
             return False
         if not x.is_contiguous(memory_format=torch.channels_last):
             return False

@@ -121,19 +121,6 @@ class TestFusedGPU:
             ref = _ref_bn(x, m, relu=True, training=False)
         assert torch.allclose(y.float(), ref, atol=3e-2, rtol=3e-2)
 
-    def test_eval_wide_channels_fused(self):
-        """C=4096 (ResNeXt101_32x16d stage-4 widths) runs the fused eval
-        apply kernel; training at that width stays on the fallback (the
-        reduce kernels' LDS caps at 2048 channels)."""
-        m, x, _ = self._mk(4096, H=5, W=5)
-        m(x)  # training step (fallback path at C>2048) moves stats
-        m.eval()
-        with torch.no_grad():
-            y = m(x)
-            ref = _ref_bn(x, m, relu=True, training=False)
-        assert torch.allclose(y.float(), ref, atol=3e-2, rtol=3e-2), \
-            (y.float() - ref).abs().max().item()
-
     def test_resnet50vd_fused_forward_close_to_fallback(self):
         """Whole-model: fused kernels vs the fp32 fallback path."""
         from edl_amd.models import resnet50_vd
