@@ -101,8 +101,15 @@ class BNReLU2d(nn.Module):
             return False
         if not (x.is_cuda and x.dtype == torch.bfloat16 and available()):
             return False
-        if self.num_features % 8 != 0 or self.num_features > 2048:
+        if self.num_features % 8 != 0:
             return False
+        if self.num_features > 2048:
+            # training reduce kernels cap at 2048 (LDS); the EVAL apply
+            # kernel has no cap — opt-in for wide teachers until the
+            # round-3 GPU validation (EDL_BN_WIDE_EVAL=1; see NOTES r2c42)
+            if (self.training or torch.is_grad_enabled()
+                    or os.environ.get("EDL_BN_WIDE_EVAL", "0") != "1"):
+                return False
         if not x.is_contiguous(memory_format=torch.channels_last):
             return False
         if res is not None and (
