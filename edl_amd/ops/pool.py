@@ -1,9 +1,14 @@
-"""AvgPool2x2 — the vd-shortcut downsample pool on CDNA4 kernels.
+"""Pooling on CDNA4 kernels: the vd-shortcut AvgPool2x2 and the stem
+MaxPool3x3s2 (parity: the reference model zoo's pool2d ops,
+example/distill/resnet/models/resnet_vd.py — avg_pool for the vd
+shortcut, 3x3/s2 max_pool after the deep stem).
 
-Drop-in for nn.AvgPool2d(2, 2, ceil_mode=True) with no padding: with
-kernel == stride every input position feeds exactly one window, so the
-backward is elementwise (torch's NHWC avg_pool2d_backward measured
-95 us/dispatch). Plain-torch fallback everywhere else."""
+AvgPool2x2 is a drop-in for nn.AvgPool2d(2, 2, ceil_mode=True) with no
+padding: with kernel == stride every input position feeds exactly one
+window, so the backward is elementwise (torch's NHWC
+avg_pool2d_backward measured 95 us/dispatch). MaxPool3x3s2 keeps u8
+tap indices from forward so backward is a <=4-window gather instead of
+an argmax re-scan. Plain-torch fallback everywhere else."""
 import torch
 import torch.nn as nn
 
